@@ -1,0 +1,56 @@
+"""DataParallel: bucketed gradient all-reduce overlapping backward.
+
+Reference parity: nn/data_parallel/data_parallel.py (per-param hook
+all-reduce; expert params reduce over EXPERT_DATA).  MI355X redesign:
+  - grads are packed into flat buckets (bucket.py) and all-reduced
+    asynchronously while backward continues — RCCL overlaps the collective
+    with compute on its own HIP stream;
+  - post-divide (sum, then /dp) instead of the reference's bf16-lossy
+    pre-divide;
+  - completion is counter-triggered: when the last grad hook fires, the tail
+    buckets flush and all works are waited on, so ``loss.backward()`` returns
+    with gradients synchronized (no explicit user call needed).
+"""
+import torch
+from torch import nn
+
+from pipegoose_amd.distributed.parallel_context import ParallelContext
+from pipegoose_amd.distributed.parallel_mode import ParallelMode
+from pipegoose_amd.nn.data_parallel.bucket import BucketManager
+from pipegoose_amd.nn.parallel import Parallel
+
+
+class DataParallel(Parallel):
+    def __init__(self, module: nn.Module, parallel_context: ParallelContext):
+        super().__init__(module, parallel_context)
+        self._bucket_manager = BucketManager(parallel_context)
+        self._hooked_params = []
+        self._fired = 0
+
+    def parallelize(self) -> nn.Module:
+        module = self.module
+        if self.parallel_context.get_world_size(ParallelMode.DATA) > 1:
+            self._register_grad_hooks(module)
+        self._save_metadata(module, self.parallel_context)
+        # expose a manual sync hook for cases where not all params get grads
+        module.finish_gradient_sync = self.finish_gradient_sync
+        return module
+
+    def _register_grad_hooks(self, module: nn.Module):
+        for p in module.parameters():
+            if p.requires_grad:
+                self._hooked_params.append(p)
+                p.register_post_accumulate_grad_hook(self._on_grad_ready)
+
+    def _on_grad_ready(self, param: torch.nn.Parameter):
+        mode = ParallelMode.EXPERT_DATA if getattr(param, "is_expert", False) \
+            else ParallelMode.DATA
+        self._bucket_manager.add_param(param, mode)
+        self._fired += 1
+        if self._fired == len(self._hooked_params):
+            self.finish_gradient_sync()
+
+    def finish_gradient_sync(self):
+        self._bucket_manager.flush()
+        self._bucket_manager.wait_all()
+        self._fired = 0

@@ -1,0 +1,31 @@
+"""MoE layer: router -> loss bookkeeping -> expert dispatch.
+
+Reference parity: nn/expert_parallel/layers.py:40-48.
+"""
+from torch import nn
+
+from pipegoose_amd.distributed.parallel_context import ParallelContext
+from pipegoose_amd.nn.expert_parallel.expert_context import ExpertContext
+
+
+class ExpertLayer(nn.Module):
+    def __init__(self, num_experts: int, expert: nn.Module, router: nn.Module,
+                 enable_tensor_parallel: bool, parallel_context: ParallelContext):
+        super().__init__()
+        from pipegoose_amd.nn.expert_parallel.experts import Experts
+        self.router = router
+        self._experts = Experts(num_experts, expert, enable_tensor_parallel, parallel_context)
+        self.parallel_context = parallel_context
+
+    @property
+    def experts(self):
+        return self._experts.experts
+
+    def forward(self, *args, **kwargs):
+        inputs = args[0]
+        router_output = self.router(inputs)
+        expert_context = ExpertContext.get_instance()
+        expert_context.push_aux_loss(router_output.aux_loss)
+        expert_context.push_z_loss(router_output.z_loss)
+        outputs = self._experts(inputs, router_output.dispatch_order, *args, **kwargs)
+        return outputs

@@ -1,0 +1,90 @@
+"""Megatron-style 1D tensor-parallel linear layers.
+
+Reference parity: nn/tensor_parallel/linear.py:40-82.  The local GEMM runs on
+hipBLASLt via torch.matmul (library GEMM); fused bias(+GeLU) epilogues use the
+hand-written CDNA4 kernels in pipegoose_amd.ops when on GPU.
+
+Fixes vs reference: ``bias=False`` is supported (reference crashed,
+linear.py:44).
+"""
+import torch
+from torch import nn
+import torch.nn.functional as TF
+
+from pipegoose_amd.distributed.parallel_context import ParallelContext
+from pipegoose_amd.distributed.parallel_mode import ParallelMode
+from pipegoose_amd.nn.tensor_parallel._functional import (
+    broadcast_to_tensor_group,
+    gather_to_tensor_group,
+    reduce_to_tensor_group,
+    scatter_to_tensor_group,
+)
+
+
+class ColumnParallelLinear(nn.Module):
+    """Y = XW^T + b with W split along output dim: each rank computes a slice
+    of the output features; optionally all-gathered along the last dim."""
+
+    def __init__(
+        self,
+        in_features: int,
+        out_features: int,
+        bias: bool = True,
+        gather_output: bool = False,
+        parallel_context: ParallelContext = None,
+    ):
+        super().__init__()
+        world = parallel_context.get_world_size(ParallelMode.TENSOR)
+        assert out_features % world == 0
+        self.in_features = in_features
+        self.out_features = out_features // world
+        self.gather_output = gather_output
+        self.parallel_context = parallel_context
+        self.weight = nn.Parameter(torch.empty(self.out_features, in_features))
+        if bias:
+            self.bias = nn.Parameter(torch.empty(self.out_features))
+        else:
+            self.register_parameter("bias", None)
+
+    def forward(self, input: torch.Tensor) -> torch.Tensor:
+        input = broadcast_to_tensor_group(input, self.parallel_context)
+        output = TF.linear(input, self.weight, self.bias)
+        if self.gather_output:
+            output = gather_to_tensor_group(output, dim=-1,
+                                            parallel_context=self.parallel_context)
+        return output
+
+
+class RowParallelLinear(nn.Module):
+    """Y = XW^T + b with W split along input dim: input is scattered along the
+    last dim, partial products are all-reduced, bias added once (unsliced)."""
+
+    def __init__(
+        self,
+        in_features: int,
+        out_features: int,
+        bias: bool = True,
+        parallel_context: ParallelContext = None,
+    ):
+        super().__init__()
+        world = parallel_context.get_world_size(ParallelMode.TENSOR)
+        assert in_features % world == 0
+        self.in_features = in_features // world
+        self.out_features = out_features
+        self.parallel_context = parallel_context
+        self.weight = nn.Parameter(torch.empty(out_features, self.in_features))
+        if bias:
+            self.bias = nn.Parameter(torch.empty(out_features))
+        else:
+            self.register_parameter("bias", None)
+
+    def forward(self, input: torch.Tensor) -> torch.Tensor:
+        tp = self.parallel_context.get_world_size(ParallelMode.TENSOR)
+        if input.size(-1) == self.in_features * tp:
+            input = scatter_to_tensor_group(input, dim=-1,
+                                            parallel_context=self.parallel_context)
+        output = TF.linear(input, self.weight)
+        output = reduce_to_tensor_group(output, self.parallel_context)
+        if self.bias is not None:
+            output = output + self.bias
+        return output

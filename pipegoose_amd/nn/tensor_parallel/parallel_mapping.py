@@ -1,0 +1,46 @@
+"""TP role mapping for HF architectures (reference:
+nn/tensor_parallel/parallel_mapping.py + nn/parallel_mapping.py).
+
+Ships bloom + albert + gpt2-style entries; extensible via
+``TensorParallelMapping.register``.
+"""
+from pipegoose_amd.nn.parallel_mapping import ParallelInfo, ParallelMapping
+
+
+class Column(ParallelInfo):
+    pass
+
+
+class Row(ParallelInfo):
+    pass
+
+
+class LMHead(ParallelInfo):
+    pass
+
+
+class TensorParallelMapping(ParallelMapping):
+    __MAPPING__ = {
+        "bloom-560m": [
+            Column("mlp.dense_h_to_4h", "self_attention.query_key_value"),
+            Row("mlp.dense_4h_to_h", "self_attention.dense"),
+            LMHead("lm_head"),
+        ],
+        "albert-base-v2": [
+            Column("attention.query", "attention.key", "attention.value", "ffn"),
+            Row("attention.dense", "ffn_output"),
+        ],
+        # pipegoose_amd native models (models/bloom.py) use the same names.
+    }
+
+    @classmethod
+    def is_column_parallel(cls, module_name: str) -> bool:
+        return isinstance(cls._search(module_name), Column)
+
+    @classmethod
+    def is_row_parallel(cls, module_name: str) -> bool:
+        return isinstance(cls._search(module_name), Row)
+
+    @classmethod
+    def is_lm_head(cls, module_name: str) -> bool:
+        return isinstance(cls._search(module_name), LMHead)

@@ -1,0 +1,3 @@
+from pipegoose_amd.optim.zero import DistributedOptimizer
+
+__all__ = ["DistributedOptimizer"]
