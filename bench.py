@@ -1,0 +1,148 @@
+#!/usr/bin/env python3
+"""Flagship training benchmark: BLOOM tokens/sec on MI355X.
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`.
+For N>1 the driver launches via torch.distributed.run, one rank per GPU over
+RCCL; this script reads RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from the env.
+Weak scaling: per-GPU microbatch fixed as N grows.
+
+Measures BASELINE.json's metric: tokens/sec (whole node), BLOOM-560M TP2xDP2
+(and BLOOM-7B1 TP2xPP2xDP2 via --model bloom-7b1 once PP lands in the bench
+path), bf16, synthetic data, random-init weights.
+"""
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--model", type=str, default="bloom-560m",
+                   choices=["bloom-560m", "bloom-1b7", "bloom-7b1", "bloom-tiny"])
+    p.add_argument("--seq-len", type=int, default=2048)
+    p.add_argument("--micro-batch", type=int, default=8,
+                   help="per-DP-rank batch size (weak scaling)")
+    p.add_argument("--tp", type=int, default=0, help="0 = auto by world size")
+    p.add_argument("--dp", type=int, default=0)
+    p.add_argument("--device", type=str, default=None)
+    return p.parse_args()
+
+
+def pick_parallelism(world_size: int, tp_arg: int, dp_arg: int):
+    """BASELINE config: TP2xDP2 at 4 GPUs; weak-scale DP beyond."""
+    if tp_arg > 0:
+        tp = tp_arg
+        dp = dp_arg if dp_arg > 0 else world_size // tp
+        return tp, dp
+    table = {1: (1, 1), 2: (1, 2), 4: (2, 2), 8: (2, 4)}
+    return table.get(world_size, (2, world_size // 2))
+
+
+def main():
+    args = parse_args()
+    world_size = int(os.environ.get("WORLD_SIZE", 1))
+    rank = int(os.environ.get("RANK", 0))
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29571")
+
+    tp, dp = pick_parallelism(world_size, args.tp, args.dp)
+
+    from pipegoose_amd import ParallelContext, ParallelMode
+    from pipegoose_amd.models.bloom import (
+        BloomForCausalLM, bloom_1b7, bloom_560m, bloom_7b1, bloom_tiny)
+    from pipegoose_amd.nn import DataParallel
+    from pipegoose_amd.optim import DistributedOptimizer
+
+    ctx = ParallelContext.from_torch(
+        tensor_parallel_size=tp, pipeline_parallel_size=1, data_parallel_size=dp)
+
+    use_gpu = torch.cuda.is_available()
+    device = torch.device(args.device) if args.device else ctx.device
+    dtype = torch.bfloat16 if use_gpu else torch.float32
+
+    cfg = {"bloom-560m": bloom_560m, "bloom-1b7": bloom_1b7,
+           "bloom-7b1": bloom_7b1, "bloom-tiny": bloom_tiny}[args.model]()
+
+    torch.manual_seed(1234)
+    model = BloomForCausalLM(cfg, ctx).to(device=device, dtype=dtype)
+    if dp > 1:
+        model = DataParallel(model, ctx).parallelize()
+
+    optim = torch.optim.AdamW(model.parameters(), lr=1e-4, betas=(0.9, 0.95))
+    if dp > 1:
+        optim = DistributedOptimizer(optim, ctx)
+
+    B, S = args.micro_batch, args.seq_len
+    # synthetic data, fixed per rank (weak scaling: per-GPU work constant)
+    g = torch.Generator().manual_seed(4242 + rank)
+    input_ids = torch.randint(0, cfg.vocab_size, (B, S), generator=g).to(device)
+
+    def one_step():
+        optim.zero_grad()
+        loss = model(input_ids, labels=input_ids)
+        loss.backward()
+        optim.step()
+        return loss
+
+    for _ in range(args.warmup):
+        one_step()
+
+    if dist.is_initialized() and world_size > 1:
+        dist.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    if use_gpu:
+        torch.cuda.synchronize()
+    if dist.is_initialized() and world_size > 1:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks (slowest rank defines the step time)
+    if dist.is_initialized() and world_size > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if dist.get_backend() == "nccl" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = t.item()
+
+    global_batch = B * dp
+    tokens_per_step = global_batch * S
+    tokens_per_sec = tokens_per_step * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1e3
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "tokens/sec",
+            "value": round(tokens_per_sec, 1),
+            "unit": "tokens/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": global_batch,
+                "seq_len": S,
+                "parallelism": f"tp{tp}dp{dp}",
+            },
+        }))
+
+    ctx.destroy()
+
+
+if __name__ == "__main__":
+    main()

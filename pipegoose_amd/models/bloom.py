@@ -1,0 +1,220 @@
+"""Native BLOOM model family, tensor-parallel by construction.
+
+This is the framework's flagship model (BASELINE.json configs: bloom-560m,
+bloom-1b7, bloom-7b1): ALiBi attention, LayerNorm, GELU MLP, tied LM head —
+behaviorally the BLOOM architecture, implemented MI355X-first:
+
+  - built directly on Column/RowParallelLinear + ParallelEmbedding, so TP
+    needs no module surgery and head counts are local by construction;
+  - LayerNorm runs the fused CDNA4 HIP kernel (ops/layer_norm);
+  - MLP uses the fused bias-GeLU kernel on GPU (ops/fused_bias_gelu);
+  - loss is the vocab-parallel streaming CE (no full-logit gather when tp>1).
+
+Module names match TensorParallelMapping ("self_attention.query_key_value",
+"mlp.dense_h_to_4h", ...) so checkpoints line up with HF-surgered models.
+"""
+import math
+from dataclasses import dataclass
+
+import torch
+from torch import nn
+import torch.nn.functional as TF
+
+from pipegoose_amd.distributed.parallel_context import ParallelContext
+from pipegoose_amd.distributed.parallel_mode import ParallelMode
+from pipegoose_amd.nn.tensor_parallel.embedding import ParallelEmbedding
+from pipegoose_amd.nn.tensor_parallel.layer_norm import LayerNorm
+from pipegoose_amd.nn.tensor_parallel.linear import ColumnParallelLinear, RowParallelLinear
+from pipegoose_amd.nn.tensor_parallel.loss import VocabParallelCrossEntropy
+from pipegoose_amd.ops.fused_bias_gelu import fused_bias_gelu
+
+
+@dataclass
+class BloomConfig:
+    vocab_size: int = 250880
+    hidden_size: int = 1024
+    n_layer: int = 24
+    n_head: int = 16
+    layer_norm_epsilon: float = 1e-5
+    initializer_range: float = 0.02
+
+    @property
+    def head_dim(self):
+        return self.hidden_size // self.n_head
+
+
+def bloom_560m():
+    return BloomConfig(hidden_size=1024, n_layer=24, n_head=16)
+
+
+def bloom_1b7():
+    return BloomConfig(hidden_size=2048, n_layer=24, n_head=16)
+
+
+def bloom_7b1():
+    return BloomConfig(hidden_size=4096, n_layer=30, n_head=32)
+
+
+def bloom_tiny():
+    """For tests."""
+    return BloomConfig(vocab_size=256, hidden_size=64, n_layer=2, n_head=4)
+
+
+def alibi_slopes(n_head: int) -> torch.Tensor:
+    """ALiBi per-head slopes (standard construction, handles non-pow2 heads)."""
+    def pow2_slopes(n):
+        start = 2.0 ** (-(2.0 ** -(math.log2(n) - 3)))
+        return [start * (start ** i) for i in range(n)]
+
+    if math.log2(n_head).is_integer():
+        return torch.tensor(pow2_slopes(n_head))
+    closest = 2 ** math.floor(math.log2(n_head))
+    slopes = pow2_slopes(closest)
+    extra = pow2_slopes(2 * closest)[0::2][: n_head - closest]
+    return torch.tensor(slopes + extra)
+
+
+class BloomAttention(nn.Module):
+    def __init__(self, config: BloomConfig, parallel_context: ParallelContext):
+        super().__init__()
+        tp = parallel_context.get_world_size(ParallelMode.TENSOR)
+        tp_rank = parallel_context.get_local_rank(ParallelMode.TENSOR)
+        assert config.n_head % tp == 0
+        self.num_heads = config.n_head // tp
+        self.head_dim = config.head_dim
+        self.hidden_size = config.hidden_size
+        self.inv_norm = 1.0 / math.sqrt(self.head_dim)
+
+        # QKV fused, grouped per head ([head][q|k|v][head_dim]) so the column
+        # slice hands whole heads to each rank.
+        self.query_key_value = ColumnParallelLinear(
+            config.hidden_size, 3 * config.hidden_size,
+            parallel_context=parallel_context)
+        self.dense = RowParallelLinear(
+            config.hidden_size, config.hidden_size,
+            parallel_context=parallel_context)
+
+        slopes = alibi_slopes(config.n_head)
+        local = slopes.chunk(tp)[tp_rank].clone()
+        self.register_buffer("alibi_slopes", local, persistent=False)
+        self._bias_cache = {}
+
+    def _alibi_bias(self, seq_len: int, device, dtype) -> torch.Tensor:
+        key = (seq_len, device, dtype)
+        cached = self._bias_cache.get(key)
+        if cached is None:
+            pos = torch.arange(seq_len, device=device)
+            # bias[h, i, j] = slope_h * (j - i) for j <= i; -inf above diagonal
+            rel = pos[None, :] - pos[:, None]
+            bias = self.alibi_slopes.to(device=device, dtype=torch.float32)[:, None, None] \
+                * rel[None, :, :].float()
+            causal = torch.full((seq_len, seq_len), float("-inf"), device=device)
+            causal = torch.triu(causal, diagonal=1)
+            bias = (bias + causal[None]).to(dtype)
+            cached = bias.unsqueeze(0)  # [1, H_local, S, S]
+            self._bias_cache = {key: cached}
+        return cached
+
+    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
+        B, S, _ = hidden.shape
+        fused = self.query_key_value(hidden)  # [B, S, local_heads * 3 * hd]
+        fused = fused.view(B, S, self.num_heads, 3, self.head_dim)
+        q = fused[..., 0, :].transpose(1, 2)  # [B, H, S, hd]
+        k = fused[..., 1, :].transpose(1, 2)
+        v = fused[..., 2, :].transpose(1, 2)
+
+        bias = self._alibi_bias(S, hidden.device, q.dtype)
+        out = TF.scaled_dot_product_attention(q, k, v, attn_mask=bias,
+                                              scale=self.inv_norm)
+        out = out.transpose(1, 2).reshape(B, S, self.num_heads * self.head_dim)
+        return self.dense(out)
+
+
+class BloomMLP(nn.Module):
+    def __init__(self, config: BloomConfig, parallel_context: ParallelContext):
+        super().__init__()
+        h = config.hidden_size
+        self.dense_h_to_4h = ColumnParallelLinear(h, 4 * h, parallel_context=parallel_context)
+        self.dense_4h_to_h = RowParallelLinear(4 * h, h, parallel_context=parallel_context)
+
+    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
+        # fused GEMM + bias + GeLU epilogue on GPU (ops/fused_bias_gelu)
+        x = fused_bias_gelu(self.dense_h_to_4h, hidden)
+        return self.dense_4h_to_h(x)
+
+
+class BloomBlock(nn.Module):
+    def __init__(self, config: BloomConfig, parallel_context: ParallelContext):
+        super().__init__()
+        eps = config.layer_norm_epsilon
+        self.input_layernorm = LayerNorm(config.hidden_size, eps=eps)
+        self.self_attention = BloomAttention(config, parallel_context)
+        self.post_attention_layernorm = LayerNorm(config.hidden_size, eps=eps)
+        self.mlp = BloomMLP(config, parallel_context)
+
+    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
+        hidden = hidden + self.self_attention(self.input_layernorm(hidden))
+        hidden = hidden + self.mlp(self.post_attention_layernorm(hidden))
+        return hidden
+
+
+class BloomModel(nn.Module):
+    def __init__(self, config: BloomConfig, parallel_context: ParallelContext):
+        super().__init__()
+        self.config = config
+        eps = config.layer_norm_epsilon
+        self.word_embeddings = ParallelEmbedding(
+            config.vocab_size, config.hidden_size, parallel_context=parallel_context)
+        self.word_embeddings_layernorm = LayerNorm(config.hidden_size, eps=eps)
+        self.h = nn.ModuleList(
+            [BloomBlock(config, parallel_context) for _ in range(config.n_layer)]
+        )
+        self.ln_f = LayerNorm(config.hidden_size, eps=eps)
+
+    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+        hidden = self.word_embeddings_layernorm(self.word_embeddings(input_ids))
+        for block in self.h:
+            hidden = block(hidden)
+        return self.ln_f(hidden)
+
+
+class BloomForCausalLM(nn.Module):
+    def __init__(self, config: BloomConfig, parallel_context: ParallelContext):
+        super().__init__()
+        self.config = config
+        self.parallel_context = parallel_context
+        self.transformer = BloomModel(config, parallel_context)
+        tp = parallel_context.get_world_size(ParallelMode.TENSOR)
+        # LM head: column-split over vocab, weights tied to the (vocab-sharded)
+        # embedding; logits stay sharded — the parallel CE consumes them.
+        self.lm_head = ColumnParallelLinear(
+            config.hidden_size, config.vocab_size, bias=False,
+            gather_output=False, parallel_context=parallel_context)
+        self.lm_head.weight = self.transformer.word_embeddings.weight
+        self.loss_fn = VocabParallelCrossEntropy(parallel_context=parallel_context) \
+            if tp > 1 else None
+        self.apply(self._init_weights)
+
+    def _init_weights(self, module):
+        std = self.config.initializer_range
+        if isinstance(module, (ColumnParallelLinear, RowParallelLinear)):
+            nn.init.normal_(module.weight, mean=0.0, std=std)
+            if module.bias is not None:
+                nn.init.zeros_(module.bias)
+        elif isinstance(module, ParallelEmbedding):
+            nn.init.normal_(module.weight, mean=0.0, std=std)
+
+    def forward(self, input_ids: torch.Tensor, labels: torch.Tensor = None):
+        hidden = self.transformer(input_ids)
+        logits = self.lm_head(hidden)
+        if labels is None:
+            return logits
+        shift_logits = logits[:, :-1].contiguous()
+        shift_labels = labels[:, 1:].contiguous()
+        if self.loss_fn is not None:
+            loss = self.loss_fn(shift_logits, shift_labels)
+        else:
+            loss = TF.cross_entropy(
+                shift_logits.float().reshape(-1, shift_logits.size(-1)),
+                shift_labels.reshape(-1))
+        return loss

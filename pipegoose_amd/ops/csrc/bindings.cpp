@@ -1,0 +1,17 @@
+// Python bindings for the pipegoose_amd CDNA4 kernel extension.
+#include <torch/extension.h>
+
+std::vector<torch::Tensor> layer_norm_fwd(torch::Tensor x, torch::Tensor w,
+                                          torch::Tensor b, double eps);
+std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
+                                          torch::Tensor w, torch::Tensor mean,
+                                          torch::Tensor rstd);
+torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias);
+torch::Tensor bias_gelu_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor bias);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
+    m.def("layer_norm_bwd", &layer_norm_bwd, "fused LayerNorm backward (gfx950)");
+    m.def("bias_gelu_fwd", &bias_gelu_fwd, "fused bias+GeLU forward (gfx950)");
+    m.def("bias_gelu_bwd", &bias_gelu_bwd, "fused bias+GeLU backward (gfx950)");
+}

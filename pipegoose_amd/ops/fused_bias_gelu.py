@@ -1,0 +1,44 @@
+"""Fused bias + GeLU epilogue.
+
+On GPU: GEMM via hipBLASLt (torch.matmul) + one fused HIP kernel applying
+bias+GeLU (tanh approx) in a single HBM pass, with a fused dgelu backward.
+CPU path: eager bias + torch GELU (the numerics oracle).
+"""
+import torch
+import torch.nn.functional as TF
+
+from pipegoose_amd.ops import get_extension
+
+
+class _BiasGelu(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, input, bias):
+        ext = get_extension(required=True)
+        out = ext.bias_gelu_fwd(input, bias)
+        ctx.save_for_backward(input, bias)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        ext = get_extension(required=True)
+        input, bias = ctx.saved_tensors
+        grad_in = ext.bias_gelu_bwd(grad_out.contiguous(), input, bias)
+        # bias grad: reduce over all leading dims
+        grad_bias = grad_in.reshape(-1, grad_in.size(-1)).sum(dim=0)
+        return grad_in, grad_bias
+
+
+def bias_gelu(input: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
+    if input.is_cuda:
+        return _BiasGelu.apply(input.contiguous(), bias)
+    return TF.gelu(input + bias, approximate="tanh")
+
+
+def fused_bias_gelu(column_linear, hidden: torch.Tensor) -> torch.Tensor:
+    """Run a ColumnParallelLinear with its bias+GeLU fused into the epilogue."""
+    from pipegoose_amd.nn.tensor_parallel._functional import broadcast_to_tensor_group
+    hidden = broadcast_to_tensor_group(hidden, column_linear.parallel_context)
+    x = TF.linear(hidden, column_linear.weight)  # hipBLASLt GEMM, no bias
+    if column_linear.bias is not None:
+        return bias_gelu(x, column_linear.bias)
+    return TF.gelu(x, approximate="tanh")

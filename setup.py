@@ -1,0 +1,37 @@
+"""Build the in-tree HIP extension for gfx950 (MI355X).
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The built .so lands at pipegoose_amd/ops/_C*.so and travels with the repo
+snapshot to GPU boxes.
+"""
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+SRC = [
+    "pipegoose_amd/ops/csrc/bindings.cpp",
+    "pipegoose_amd/ops/csrc/layer_norm.hip",
+    "pipegoose_amd/ops/csrc/bias_gelu.hip",
+]
+
+setup(
+    name="pipegoose_amd",
+    version="0.1.0",
+    packages=["pipegoose_amd"],
+    ext_modules=[
+        CUDAExtension(
+            name="pipegoose_amd.ops._C",
+            sources=SRC,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension},
+)
