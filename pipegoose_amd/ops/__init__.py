@@ -15,6 +15,8 @@ _EXT_ERR = None
 
 def _load():
     global _EXT, _EXT_ERR
+    if os.environ.get("PIPEGOOSE_DISABLE_EXT") == "1":  # debugging only
+        return None
     if _EXT is not None or _EXT_ERR is not None:
         return _EXT
     try:

@@ -3,6 +3,8 @@
 CPU path: torch.nn.functional.layer_norm (the numerics oracle the GPU kernel
 is tested against, tests/ops/test_layer_norm.py).
 """
+import os
+
 import torch
 import torch.nn.functional as TF
 
@@ -26,6 +28,6 @@ class _FusedLayerNorm(torch.autograd.Function):
 
 
 def fused_layer_norm(input, normalized_shape, weight, bias, eps=1e-5):
-    if input.is_cuda:
+    if input.is_cuda and os.environ.get("PIPEGOOSE_DISABLE_EXT") != "1":
         return _FusedLayerNorm.apply(input.contiguous(), weight, bias, eps)
     return TF.layer_norm(input, normalized_shape, weight, bias, eps)

@@ -21,8 +21,12 @@ def test_layer_norm_fwd(shape, dtype):
     b = torch.randn(H, device="cuda", dtype=dtype)
     y, mean, rstd = ext.layer_norm_fwd(x.contiguous(), w, b, 1e-5)
     ref = torch.nn.functional.layer_norm(x.float(), (H,), w.float(), b.float(), 1e-5)
-    tol = 1e-5 if dtype == torch.float32 else 3e-2
-    assert torch.allclose(y.float(), ref, atol=tol), (y.float() - ref).abs().max()
+    if dtype == torch.float32:
+        assert torch.allclose(y, ref, atol=1e-5), (y - ref).abs().max()
+    else:
+        # bf16 output rounding: half-ULP relative error on the stored value
+        assert torch.allclose(y.float(), ref, rtol=1e-2, atol=2e-2), \
+            (y.float() - ref).abs().max()
 
 
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
