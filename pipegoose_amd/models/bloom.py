@@ -213,6 +213,11 @@ class BloomForCausalLM(nn.Module):
         shift_labels = labels[:, 1:].contiguous()
         if self.loss_fn is not None:
             loss = self.loss_fn(shift_logits, shift_labels)
+        elif logits.is_cuda:
+            from pipegoose_amd.ops.cross_entropy import fused_cross_entropy
+            loss = fused_cross_entropy(
+                shift_logits.reshape(-1, shift_logits.size(-1)),
+                shift_labels.reshape(-1))
         else:
             loss = TF.cross_entropy(
                 shift_logits.float().reshape(-1, shift_logits.size(-1)),

@@ -76,6 +76,21 @@ class VocabParallelCrossEntropy(nn.Module):
         orig_shape = targets.shape
         logits = logits.reshape(-1, logits.size(-1))
         targets = targets.reshape(-1)
+        if logits.is_cuda:
+            # fused single-pass HIP kernel + 2 allreduces (ops/cross_entropy)
+            from pipegoose_amd.distributed.parallel_mode import ParallelMode
+            from pipegoose_amd.nn.tensor_parallel._utils import VocabUtility
+            from pipegoose_amd.ops.cross_entropy import fused_cross_entropy
+            rank = self.parallel_context.get_local_rank(ParallelMode.TENSOR)
+            start, end = VocabUtility.get_vocab_range_from_partition_size(
+                logits.size(-1), rank)
+            loss = fused_cross_entropy(logits, targets, start, end,
+                                       self.parallel_context, reduction="none")
+            if self.reduction == "mean":
+                return loss.mean()
+            if self.reduction == "sum":
+                return loss.sum()
+            return loss.reshape(orig_shape)
         loss = _VocabParallelCrossEntropy.apply(logits, targets, self.parallel_context)
         if self.reduction == "mean":
             return loss.mean()

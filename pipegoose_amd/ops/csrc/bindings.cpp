@@ -8,10 +8,20 @@ std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
                                           torch::Tensor rstd);
 torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias);
 torch::Tensor bias_gelu_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor bias);
+std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits, torch::Tensor targets,
+                                             int64_t vocab_start, int64_t vocab_end);
+torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor targets,
+                                torch::Tensor row_max, torch::Tensor row_sumexp,
+                                torch::Tensor gscale, int64_t vocab_start,
+                                int64_t vocab_end);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
     m.def("layer_norm_bwd", &layer_norm_bwd, "fused LayerNorm backward (gfx950)");
     m.def("bias_gelu_fwd", &bias_gelu_fwd, "fused bias+GeLU forward (gfx950)");
     m.def("bias_gelu_bwd", &bias_gelu_bwd, "fused bias+GeLU backward (gfx950)");
+    m.def("cross_entropy_fwd", &cross_entropy_fwd,
+          "fused online-softmax CE forward over a vocab shard (gfx950)");
+    m.def("cross_entropy_bwd", &cross_entropy_bwd,
+          "fused CE backward: (softmax - onehot) * g (gfx950)");
 }

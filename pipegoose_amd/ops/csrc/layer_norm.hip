@@ -131,12 +131,48 @@ __global__ void layer_norm_bwd_dx_kernel(
 }
 
 // dw[j] = sum_i dy[i,j] * xhat[i,j]; db[j] = sum_i dy[i,j]
-// Column-parallel: thread t owns column col = blockIdx.x*BLOCK + t, loops rows.
-// Consecutive threads read consecutive addresses (coalesced); partials land in
-// fp32 dw/db directly via atomics-free two-stage: each block covers all rows
-// for its columns, so no atomics needed.
+// Column-parallel, 8 columns per thread (vectorized 16-B bf16 loads — G13:
+// scalar bf16 column reads were 8x off the HBM roofline).  Thread t owns
+// columns [8*(blockIdx.x*BLOCK + t), +8); rows strided over gridDim.y.
 template <typename T, int BLOCK>
 __global__ void layer_norm_bwd_dwdb_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const float* __restrict__ mean_in, const float* __restrict__ rstd_in,
+    float* __restrict__ dw, float* __restrict__ db, int64_t N, int H) {
+    using V = typename vec8<T>::type;
+    const int colv = blockIdx.x * BLOCK + threadIdx.x;  // packet column index
+    if (colv >= H / 8) return;
+    float sw[8] = {0.f}, sb[8] = {0.f};
+    const int HV = H / 8;
+    for (int64_t i = blockIdx.y; i < N; i += gridDim.y) {
+        const float mean = mean_in[i];
+        const float rstd = rstd_in[i];
+        V dyp = reinterpret_cast<const V*>(dy + i * H)[colv];
+        V xp = reinterpret_cast<const V*>(x + i * H)[colv];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float dyv = to_float(dyp.v[j]);
+            float xhat = (to_float(xp.v[j]) - mean) * rstd;
+            sw[j] += dyv * xhat;
+            sb[j] += dyv;
+        }
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        const int col = colv * 8 + j;
+        if (gridDim.y == 1) {
+            dw[col] = sw[j];
+            db[col] = sb[j];
+        } else {
+            atomicAdd(&dw[col], sw[j]);
+            atomicAdd(&db[col], sb[j]);
+        }
+    }
+}
+
+// scalar fallback when H % 8 != 0
+template <typename T, int BLOCK>
+__global__ void layer_norm_bwd_dwdb_scalar_kernel(
     const T* __restrict__ dy, const T* __restrict__ x,
     const float* __restrict__ mean_in, const float* __restrict__ rstd_in,
     float* __restrict__ dw, float* __restrict__ db, int64_t N, int H) {
@@ -203,9 +239,12 @@ std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
 
     constexpr int BLOCK = 256;
     auto stream = at::cuda::getCurrentCUDAStream();
-    // dwdb: spread row-loop over enough blocks to fill 256 CUs x 8 XCDs.
-    int grid_y = (int)std::min<int64_t>((N + 255) / 256, 64);
-    dim3 grid_dw((H + BLOCK - 1) / BLOCK, grid_y);
+    // dwdb: fill the chip (256 CUs, >=2 blocks/CU) with row-strided blocks;
+    // fp32 atomics combine partials (512 adds/column max — low contention).
+    const bool vec = (H % 8 == 0);
+    const int64_t cols = vec ? H / 8 : H;
+    int grid_y = (int)std::min<int64_t>(N, 512);
+    dim3 grid_dw((cols + BLOCK - 1) / BLOCK, grid_y);
 
     if (x.scalar_type() == torch::kBFloat16) {
         hipLaunchKernelGGL((layer_norm_bwd_dx_kernel<__hip_bfloat16, BLOCK>), dim3(N), dim3(BLOCK), 0, stream,
@@ -214,20 +253,35 @@ std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
             reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
             mean.data_ptr<float>(), rstd.data_ptr<float>(),
             reinterpret_cast<__hip_bfloat16*>(dx.data_ptr()), (int)H);
-        hipLaunchKernelGGL((layer_norm_bwd_dwdb_kernel<__hip_bfloat16, BLOCK>), grid_dw, dim3(BLOCK), 0, stream,
-            reinterpret_cast<const __hip_bfloat16*>(dy.data_ptr()),
-            reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
-            mean.data_ptr<float>(), rstd.data_ptr<float>(),
-            dw.data_ptr<float>(), db.data_ptr<float>(), N, (int)H);
+        if (vec) {
+            hipLaunchKernelGGL((layer_norm_bwd_dwdb_kernel<__hip_bfloat16, BLOCK>), grid_dw, dim3(BLOCK), 0, stream,
+                reinterpret_cast<const __hip_bfloat16*>(dy.data_ptr()),
+                reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                dw.data_ptr<float>(), db.data_ptr<float>(), N, (int)H);
+        } else {
+            hipLaunchKernelGGL((layer_norm_bwd_dwdb_scalar_kernel<__hip_bfloat16, BLOCK>), grid_dw, dim3(BLOCK), 0, stream,
+                reinterpret_cast<const __hip_bfloat16*>(dy.data_ptr()),
+                reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                dw.data_ptr<float>(), db.data_ptr<float>(), N, (int)H);
+        }
     } else if (x.scalar_type() == torch::kFloat) {
         hipLaunchKernelGGL((layer_norm_bwd_dx_kernel<float, BLOCK>), dim3(N), dim3(BLOCK), 0, stream,
             dy.data_ptr<float>(), x.data_ptr<float>(), w.data_ptr<float>(),
             mean.data_ptr<float>(), rstd.data_ptr<float>(),
             dx.data_ptr<float>(), (int)H);
-        hipLaunchKernelGGL((layer_norm_bwd_dwdb_kernel<float, BLOCK>), grid_dw, dim3(BLOCK), 0, stream,
-            dy.data_ptr<float>(), x.data_ptr<float>(),
-            mean.data_ptr<float>(), rstd.data_ptr<float>(),
-            dw.data_ptr<float>(), db.data_ptr<float>(), N, (int)H);
+        if (vec) {
+            hipLaunchKernelGGL((layer_norm_bwd_dwdb_kernel<float, BLOCK>), grid_dw, dim3(BLOCK), 0, stream,
+                dy.data_ptr<float>(), x.data_ptr<float>(),
+                mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                dw.data_ptr<float>(), db.data_ptr<float>(), N, (int)H);
+        } else {
+            hipLaunchKernelGGL((layer_norm_bwd_dwdb_scalar_kernel<float, BLOCK>), grid_dw, dim3(BLOCK), 0, stream,
+                dy.data_ptr<float>(), x.data_ptr<float>(),
+                mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                dw.data_ptr<float>(), db.data_ptr<float>(), N, (int)H);
+        }
     } else {
         TORCH_CHECK(false, "layer_norm_bwd: unsupported dtype");
     }
