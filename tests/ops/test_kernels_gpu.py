@@ -140,3 +140,36 @@ def test_fused_cross_entropy_sharded_equivalence():
     assert torch.allclose(full_loss, sharded_loss, atol=1e-5)
     ref = torch.nn.functional.cross_entropy(logits, targets, reduction="none")
     assert torch.allclose(full_loss, ref, atol=1e-4)
+
+
+def test_alibi_fold_matches_mask():
+    """The q/k-append ALiBi fold must match explicit-mask attention."""
+    torch.manual_seed(9)
+    B, H, S, D = 2, 4, 512, 64
+    q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    k, v = torch.randn_like(q), torch.randn_like(q)
+    from pipegoose_amd.models.bloom import alibi_slopes
+    slopes = alibi_slopes(H).cuda()
+    scale = 1.0 / (D ** 0.5)
+
+    # reference: fp32 math attention with explicit alibi+causal mask
+    pos = torch.arange(S, device="cuda")
+    rel = (pos[None, :] - pos[:, None]).float()
+    bias = slopes[:, None, None] * rel[None]
+    bias = bias + torch.triu(torch.full((S, S), float("-inf"), device="cuda"), 1)[None]
+    scores = (q.float() @ k.float().transpose(-1, -2)) * scale + bias[None]
+    ref = torch.softmax(scores, dim=-1) @ v.float()
+
+    # fold: append [slope*256, slope] to q*scale and [j_hi, j_lo] to k
+    j = pos
+    j_hi = (j // 256).bfloat16()
+    j_lo = (j % 256).bfloat16()
+    k_ext = torch.stack([j_hi, j_lo], -1)[None, None].expand(B, H, S, 2)
+    q_ext = torch.stack([slopes * 256, slopes], -1).bfloat16()[None, :, None, :] \
+        .expand(B, H, S, 2)
+    qf = torch.cat([q * scale, q_ext], -1)
+    kf = torch.cat([k, k_ext], -1)
+    out = torch.nn.functional.scaled_dot_product_attention(qf, kf, v,
+                                                           is_causal=True, scale=1.0)
+    err = (out.float() - ref).abs().max()
+    assert err < 0.06, err
