@@ -116,25 +116,6 @@ class BloomAttention(nn.Module):
             self._bias_cache = {key: cached}
         return cached
 
-    def _fold_cache(self, S, device, dtype):
-        """Tensors for the ALiBi fold trick: bias slope*(j-i) is, per softmax
-        row, shift-equivalent to slope*j — which is expressible as a dot
-        product of appended q/k columns, so the fast causal flash path runs
-        with no materialized [H,S,S] mask.  j is split j_hi*256 + j_lo so both
-        parts are exact in bf16 (integers < 256)."""
-        key = (S, device, dtype)
-        cached = self._fold_cache_store.get(key)
-        if cached is None:
-            j = torch.arange(S, device=device)
-            j_hi = (j // 256).to(dtype)
-            j_lo = (j % 256).to(dtype)
-            k_ext = torch.stack([j_hi, j_lo], dim=-1)  # [S, 2]
-            slopes = self.alibi_slopes.to(device=device, dtype=torch.float32)
-            q_ext = torch.stack([slopes * 256.0, slopes], dim=-1).to(dtype)  # [H, 2]
-            self._fold_cache_store = {key: (q_ext, k_ext)}
-            cached = (q_ext, k_ext)
-        return cached
-
     def forward(self, hidden: torch.Tensor) -> torch.Tensor:
         B, S, _ = hidden.shape
         fused = self.query_key_value(hidden)  # [B, S, local_heads * 3 * hd]
@@ -143,18 +124,9 @@ class BloomAttention(nn.Module):
         k = fused[..., 1, :].transpose(1, 2)
         v = fused[..., 2, :].transpose(1, 2)
 
-        if hidden.is_cuda:
-            q_ext, k_ext = self._fold_cache(S, hidden.device, q.dtype)
-            qf = torch.cat(
-                [q * self.inv_norm,
-                 q_ext[None, :, None, :].expand(B, self.num_heads, S, 2)], dim=-1)
-            kf = torch.cat(
-                [k, k_ext[None, None, :, :].expand(B, self.num_heads, S, 2)], dim=-1)
-            out = TF.scaled_dot_product_attention(qf, kf, v, is_causal=True, scale=1.0)
-        else:
-            bias = self._alibi_bias(S, hidden.device, q.dtype)
-            out = TF.scaled_dot_product_attention(q, k, v, attn_mask=bias,
-                                                  scale=self.inv_norm)
+        from pipegoose_amd.ops.attention import alibi_attention
+        out = alibi_attention(q, k, v, self.alibi_slopes, self.inv_norm,
+                              mask_fallback=self._alibi_bias)
         out = out.transpose(1, 2).reshape(B, S, self.num_heads * self.head_dim)
         return self.dense(out)
 

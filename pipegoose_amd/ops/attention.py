@@ -1,0 +1,52 @@
+"""Causal ALiBi attention dispatch.
+
+GPU path: hand-written CDNA4 flash-attention kernel with the ALiBi bias
+slope*(j-i) computed in-kernel from the per-head slopes (no [H,S,S] mask
+tensor) — see csrc/attention.hip.  Until/unless the kernel supports a given
+shape, falls back to torch sdpa with an explicit additive mask.
+CPU path: sdpa math with the mask (numerics oracle).
+"""
+import os
+
+import torch
+import torch.nn.functional as TF
+
+from pipegoose_amd.ops import get_extension
+
+
+def _kernel_supported(q: torch.Tensor) -> bool:
+    if not q.is_cuda or os.environ.get("PIPEGOOSE_DISABLE_EXT") == "1":
+        return False
+    D = q.size(-1)
+    S = q.size(-2)
+    ext = get_extension()
+    if ext is None or not hasattr(ext, "attn_fwd"):
+        return False
+    return D in (64, 128) and S % 64 == 0 and q.dtype == torch.bfloat16
+
+
+class _AlibiFlashAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, slopes, scale):
+        ext = get_extension(required=True)
+        o, lse = ext.attn_fwd(q, k, v, slopes, scale)
+        ctx.save_for_backward(q, k, v, o, lse, slopes)
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        ext = get_extension(required=True)
+        q, k, v, o, lse, slopes = ctx.saved_tensors
+        dq, dk, dv = ext.attn_bwd(do.contiguous(), q, k, v, o, lse, slopes, ctx.scale)
+        return dq, dk, dv, None, None
+
+
+def alibi_attention(q, k, v, slopes, scale, mask_fallback=None):
+    """q,k,v: [B, H, S, D]; slopes: [H] fp32; causal + alibi bias in-kernel."""
+    if _kernel_supported(q):
+        return _AlibiFlashAttention.apply(
+            q.contiguous(), k.contiguous(), v.contiguous(),
+            slopes.to(device=q.device, dtype=torch.float32), scale)
+    bias = mask_fallback(q.size(-2), q.device, q.dtype)
+    return TF.scaled_dot_product_attention(q, k, v, attn_mask=bias, scale=scale)
