@@ -26,15 +26,30 @@ class DataParallel(Parallel):
         self._bucket_manager = BucketManager(parallel_context)
         self._hooked_params = []
         self._fired = 0
+        # False during pipeline microbatch accumulation; engine calls
+        # sync_now() after the last microbatch.
+        self.sync_enabled = True
 
     def parallelize(self) -> nn.Module:
         module = self.module
         if self.parallel_context.get_world_size(ParallelMode.DATA) > 1:
             self._register_grad_hooks(module)
         self._save_metadata(module, self.parallel_context)
-        # expose a manual sync hook for cases where not all params get grads
+        # expose hooks for manual / pipeline-driven sync
         module.finish_gradient_sync = self.finish_gradient_sync
+        module._dp_wrapper = self
         return module
+
+    def sync_now(self):
+        """Bucket + all-reduce every accumulated grad (pipeline tail sync)."""
+        if self.parallel_context.get_world_size(ParallelMode.DATA) == 1:
+            return
+        for p in self._hooked_params:
+            if p.grad is not None:
+                mode = ParallelMode.EXPERT_DATA if getattr(p, "is_expert", False) \
+                    else ParallelMode.DATA
+                self._bucket_manager.add_param(p, mode)
+        self.finish_gradient_sync()
 
     def _register_grad_hooks(self, module: nn.Module):
         for p in module.parameters():
@@ -43,6 +58,8 @@ class DataParallel(Parallel):
                 p.register_post_accumulate_grad_hook(self._on_grad_ready)
 
     def _on_grad_ready(self, param: torch.nn.Parameter):
+        if not self.sync_enabled:
+            return
         mode = ParallelMode.EXPERT_DATA if getattr(param, "is_expert", False) \
             else ParallelMode.DATA
         self._bucket_manager.add_param(param, mode)

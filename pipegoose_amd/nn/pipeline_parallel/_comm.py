@@ -1,0 +1,41 @@
+"""Pipeline P2P transport over RCCL/gloo.
+
+Replaces the reference's TensorPipe RPC data plane (nn/pipeline_parallel/
+_comm.py:9-41 pushed Packages via rpc_sync into a global queue).  Here
+activations/grads move as raw tensors over torch.distributed P2P — RCCL
+send/recv rides xGMI directly; shapes are negotiated once per engine run
+(same [B/m, S, H] every microbatch), so steady-state transfers are a single
+payload message with no metadata round trips.
+"""
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from pipegoose_amd.distributed.parallel_context import ParallelContext
+from pipegoose_amd.distributed.parallel_mode import ParallelMode
+
+
+class PipelineP2P:
+    def __init__(self, parallel_context: ParallelContext):
+        self.pc = parallel_context
+        self.group = parallel_context.get_group(ParallelMode.PIPELINE)
+        self._shape_cache = {}
+
+    def _device(self):
+        if dist.get_backend(self.group) == "nccl":
+            return self.pc.device
+        return torch.device("cpu")
+
+    # Fixed-shape fast path: caller guarantees shape/dtype via negotiate()
+    def send_activation(self, tensor: torch.Tensor, dst_global: int, tag: int = 0):
+        payload = tensor.detach().contiguous().to(self._device())
+        work = dist.isend(payload, dst=dst_global, group=self.group, tag=tag)
+        return work, payload  # keep payload alive until work completes
+
+    def recv_activation(self, shape, dtype, src_global: int, tag: int = 0,
+                        requires_grad: bool = False) -> torch.Tensor:
+        buf = torch.empty(shape, dtype=dtype, device=self._device())
+        dist.recv(buf, src=src_global, group=self.group, tag=tag)
+        buf.requires_grad_(requires_grad)
+        return buf

@@ -1,0 +1,164 @@
+"""Pipeline engine: executes a 1F1B (or GPipe) schedule with RCCL P2P.
+
+Redesign of the reference's engine (nn/pipeline_parallel/pipeline_engine.py):
+no RPC, no worker threads, no per-clock global barriers, no autograd-side-
+effect backward (the part SURVEY says not to imitate — _job/creator.py:162-277).
+Each rank walks its own action list; ordering is enforced by the blocking P2P
+matches themselves (forward acts flow down, grads flow up — distinct channel
+directions, so RCCL needs no tags).  The first microbatch of a run uses the
+typed P2P codec to negotiate shape/dtype; steady state sends raw payloads.
+
+Backward IS an explicit engine phase: saved (input, output) per microbatch,
+`torch.autograd.backward(output, grad)` then ship `input.grad` upstream.
+"""
+from typing import Callable, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from pipegoose_amd.distributed.p2p import P2P
+from pipegoose_amd.distributed.parallel_context import ParallelContext
+from pipegoose_amd.distributed.parallel_mode import ParallelMode
+from pipegoose_amd.nn.pipeline_parallel import microbatch
+from pipegoose_amd.nn.pipeline_parallel._comm import PipelineP2P
+from pipegoose_amd.nn.pipeline_parallel.scheduler import (
+    GPipeScheduler,
+    JobType,
+    OneFOneBScheduler,
+)
+
+
+class PipelineEngine:
+    def __init__(
+        self,
+        stage_module: torch.nn.Module,
+        parallel_context: ParallelContext,
+        n_microbatches: int,
+        schedule: str = "1f1b",
+        loss_fn: Optional[Callable] = None,
+    ):
+        self.stage = stage_module
+        self.pc = parallel_context
+        self.n_microbatches = n_microbatches
+        self.schedule_kind = schedule
+        self.loss_fn = loss_fn
+        self.p2p = PipelineP2P(parallel_context)
+        self.codec = P2P(parallel_context, ParallelMode.PIPELINE)
+
+        self.rank = parallel_context.get_local_rank(ParallelMode.PIPELINE)
+        self.n_stages = parallel_context.get_world_size(ParallelMode.PIPELINE)
+        self.is_first = self.rank == 0
+        self.is_last = self.rank == self.n_stages - 1
+        self.prev_rank = parallel_context.get_prev_global_rank(ParallelMode.PIPELINE)
+        self.next_rank = parallel_context.get_next_global_rank(ParallelMode.PIPELINE)
+
+        self._act_shape = None
+        self._act_dtype = None
+
+    def _actions(self):
+        if self.schedule_kind == "gpipe":
+            sched = GPipeScheduler(self.n_microbatches, self.n_stages)
+            actions = []
+            for clock in sched.get_schedule():
+                actions.extend(t for t in clock if t.partition_idx == self.rank)
+            return actions
+        return OneFOneBScheduler(self.n_microbatches, self.n_stages) \
+            .get_rank_schedule(self.rank)
+
+    # ------------------------------------------------------------------- run
+
+    def run(self, inputs: torch.Tensor, labels: Optional[torch.Tensor] = None):
+        """One training step over the pipeline.
+
+        Returns the mean loss on the LAST stage (None elsewhere) when
+        loss_fn/labels are given; otherwise the gathered outputs on the last
+        stage.
+        """
+        m = self.n_microbatches
+        input_mbs = microbatch.split(inputs, m) if self.is_first else [None] * m
+        label_mbs = microbatch.split(labels, m) \
+            if (self.is_last and labels is not None) else [None] * m
+
+        saved_in: List[Optional[torch.Tensor]] = [None] * m
+        saved_out: List[Optional[torch.Tensor]] = [None] * m
+        losses: List[torch.Tensor] = []
+        outputs: List[torch.Tensor] = []
+        pending = []  # (work, payload) keep-alives
+
+        dp = getattr(self.stage, "_dp_wrapper", None)
+        if dp is not None:
+            dp.sync_enabled = False
+
+        for task in self._actions():
+            mb = task.microbatch_idx
+            if task.job_type == JobType.FORWARD:
+                if self.is_first:
+                    x = input_mbs[mb]
+                else:
+                    x = self._recv_forward(mb)
+                    x.requires_grad_(x.is_floating_point())
+                saved_in[mb] = x
+                out = self.stage(x)
+                saved_out[mb] = out
+                if not self.is_last:
+                    pending.append(self._send_forward(out, mb))
+                elif self.loss_fn is None or labels is None:
+                    outputs.append(out.detach())
+            else:  # BACKWARD
+                out = saved_out[mb]
+                if self.is_last:
+                    if self.loss_fn is not None and labels is not None:
+                        loss = self.loss_fn(out, label_mbs[mb]) / m
+                        losses.append(loss.detach())
+                        loss.backward()
+                    # inference-only: nothing to do
+                else:
+                    grad = self._recv_backward(mb, out)
+                    torch.autograd.backward(out, grad_tensors=grad)
+                x = saved_in[mb]
+                if not self.is_first and x is not None and x.grad is not None:
+                    pending.append(self._send_backward(x.grad, mb))
+                saved_in[mb] = saved_out[mb] = None  # free activations
+
+        for work, _payload in pending:
+            work.wait()
+
+        if dp is not None:
+            dp.sync_enabled = True
+            dp.sync_now()
+
+        if self.loss_fn is not None and labels is not None:
+            if self.is_last and losses:
+                return torch.stack(losses).sum()
+            return None
+        if self.is_last:
+            return torch.cat(outputs, dim=0) if outputs else None
+        return None
+
+    # ------------------------------------------------------------- transport
+
+    def _send_forward(self, out, mb):
+        if mb == 0:
+            self.codec.send(out, self.next_rank)
+            return _NullWork(), None
+        return self.p2p.send_activation(out, self.next_rank)
+
+    def _recv_forward(self, mb):
+        if mb == 0:
+            t = self.codec.recv(self.prev_rank)
+            self._act_shape = tuple(t.shape)
+            self._act_dtype = t.dtype
+            return t
+        return self.p2p.recv_activation(self._act_shape, self._act_dtype,
+                                        self.prev_rank)
+
+    def _send_backward(self, grad, mb):
+        return self.p2p.send_activation(grad, self.prev_rank)
+
+    def _recv_backward(self, mb, out):
+        return self.p2p.recv_activation(tuple(out.shape), out.dtype, self.next_rank)
+
+
+class _NullWork:
+    def wait(self):
+        pass

@@ -1,0 +1,74 @@
+"""PipelineParallel wrapper: partition the model, keep this rank's stage,
+expose a module whose forward runs the 1F1B engine.
+
+Reference parity: nn/pipeline_parallel/pipeline_parallel.py:26-50 (same
+user-facing shape: wrap, parallelize, call the model).  The returned module's
+``forward(inputs, labels=None)`` executes one full pipelined step (forward
++ backward when labels given) and returns the mean loss on the last stage.
+"""
+from typing import Callable, Optional
+
+import torch
+from torch import nn
+
+from pipegoose_amd.distributed.parallel_context import ParallelContext
+from pipegoose_amd.distributed.parallel_mode import ParallelMode
+from pipegoose_amd.nn.parallel import Parallel
+from pipegoose_amd.nn.pipeline_parallel.engine import PipelineEngine
+from pipegoose_amd.nn.pipeline_parallel.partitioner import UniformPartitioner
+
+
+class PipelineStageModule(nn.Module):
+    def __init__(self, engine: PipelineEngine):
+        super().__init__()
+        self.stage = engine.stage  # registers stage params
+        self.engine = engine
+
+    def forward(self, inputs, labels=None):
+        return self.engine.run(inputs, labels)
+
+
+class PipelineParallel(Parallel):
+    def __init__(
+        self,
+        module: nn.Module,
+        parallel_context: ParallelContext,
+        n_microbatches: int = 1,
+        schedule: str = "1f1b",
+        loss_fn: Optional[Callable] = None,
+        partition_sizes=None,
+    ):
+        super().__init__(module, parallel_context)
+        self.n_microbatches = n_microbatches
+        self.schedule = schedule
+        self.loss_fn = loss_fn
+        self.partition_sizes = partition_sizes
+
+    def parallelize(self) -> nn.Module:
+        pp = self.parallel_context.get_world_size(ParallelMode.PIPELINE)
+        if pp == 1:
+            return self.module
+        self._untie_shared_weights()
+        stage = UniformPartitioner(
+            self.module, self.parallel_context, sizes=self.partition_sizes
+        ).get_model_partition()
+        engine = PipelineEngine(
+            stage, self.parallel_context, self.n_microbatches,
+            schedule=self.schedule, loss_fn=self.loss_fn)
+        wrapped = PipelineStageModule(engine)
+        self._save_metadata(wrapped, self.parallel_context)
+        return wrapped
+
+    def _untie_shared_weights(self):
+        """Embedding/LM-head weight tying cannot span pipeline stages; clone
+        the tied weight so each stage owns its parameter (forward-identical;
+        Megatron-style tied-grad all-reduce is a later optimization)."""
+        from pipegoose_amd.nn.tensor_parallel.embedding import ParallelEmbedding
+        emb_weights = {id(m.weight) for m in self.module.modules()
+                       if isinstance(m, (nn.Embedding, ParallelEmbedding))}
+        for m in self.module.modules():
+            if isinstance(m, nn.Embedding) or isinstance(m, ParallelEmbedding):
+                continue
+            w = getattr(m, "weight", None)
+            if w is not None and id(w) in emb_weights:
+                m.weight = nn.Parameter(w.detach().clone())

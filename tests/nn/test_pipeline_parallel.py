@@ -1,0 +1,165 @@
+"""Pipeline parallelism: schedules (pure logic) + engine parity vs a
+single-process oracle (reference: tests/nn/pipeline_parallel/)."""
+import pytest
+import torch
+from torch import nn
+
+from pipegoose_amd.nn.pipeline_parallel.microbatch import split
+from pipegoose_amd.nn.pipeline_parallel.scheduler import (
+    GPipeScheduler,
+    JobType,
+    OneFOneBScheduler,
+    Task,
+)
+from pipegoose_amd.testing import init_parallel_context, spawn
+
+
+# ----------------------------------------------------------------- scheduler
+
+def test_gpipe_forward_schedule():
+    sched = GPipeScheduler(n_microbatches=3, n_partitions=2)
+    fwd = sched.get_forward_schedule()
+    assert len(fwd) == 4  # p + m - 1
+    assert fwd[0] == [Task(JobType.FORWARD, 0, 0)]
+    assert fwd[1] == [Task(JobType.FORWARD, 1, 0), Task(JobType.FORWARD, 0, 1)]
+    assert fwd[3] == [Task(JobType.FORWARD, 2, 1)]
+    total = sum(len(c) for c in fwd)
+    assert total == 6  # m * p tasks
+
+
+def test_1f1b_rank_schedules():
+    sched = OneFOneBScheduler(n_microbatches=4, n_partitions=2)
+    r0 = sched.get_rank_schedule(0)
+    r1 = sched.get_rank_schedule(1)
+    # every rank runs m forwards + m backwards
+    for r in (r0, r1):
+        assert sum(t.job_type == JobType.FORWARD for t in r) == 4
+        assert sum(t.job_type == JobType.BACKWARD for t in r) == 4
+    # last rank strictly alternates F,B (no warmup)
+    kinds = [t.job_type for t in r1]
+    assert kinds == [JobType.FORWARD, JobType.BACKWARD] * 4
+    # rank0 warms up with exactly p-1 forwards
+    assert [t.job_type for t in r0[:2]] == [JobType.FORWARD, JobType.FORWARD]
+    # in-flight microbatches on rank0 never exceed p
+    depth, max_depth = 0, 0
+    for t in r0:
+        depth += 1 if t.job_type == JobType.FORWARD else -1
+        max_depth = max(max_depth, depth)
+    assert max_depth == 2
+
+
+def test_microbatch_split():
+    x = torch.arange(12).reshape(6, 2)
+    mbs = split(x, 3)
+    assert len(mbs) == 3 and all(mb.shape == (2, 2) for mb in mbs)
+    d = split({"a": x, "b": x + 1}, 2)
+    assert len(d) == 2 and d[0]["a"].shape == (3, 2)
+
+
+# -------------------------------------------------------------------- engine
+
+HID = 16
+
+
+def _toy_model(seed=0):
+    torch.manual_seed(seed)
+    return nn.Sequential(
+        nn.Linear(HID, HID), nn.Tanh(),
+        nn.Linear(HID, HID), nn.Tanh(),
+        nn.Linear(HID, HID),
+    )
+
+
+def _loss_fn(out, target):
+    return ((out - target) ** 2).mean()
+
+
+def run_pp_engine(rank, world_size, port, schedule):
+    from pipegoose_amd.nn.pipeline_parallel import PipelineParallel
+
+    ctx = init_parallel_context(rank, world_size, port,
+                               pipeline_parallel_size=world_size)
+    model = _toy_model()
+    ref = _toy_model()  # same seed -> same weights
+
+    torch.manual_seed(50)
+    x = torch.randn(8, HID)
+    target = torch.randn(8, HID)
+
+    pp = PipelineParallel(model, ctx, n_microbatches=4, schedule=schedule,
+                          loss_fn=_loss_fn).parallelize()
+    loss = pp(x, target)
+
+    # oracle
+    ref_loss = _loss_fn(ref(x), target)
+    ref_loss.backward()
+
+    if rank == world_size - 1:
+        assert loss is not None
+        assert torch.allclose(loss, ref_loss, atol=1e-6), (loss, ref_loss)
+    else:
+        assert loss is None
+
+    # per-stage grads must match the oracle's corresponding layers
+    ref_layers = list(ref)
+    n_per_stage = len(ref_layers) // world_size  # by param weight, see below
+    # map: stage params -> compare against same-named layers in the split
+    from pipegoose_amd.nn.pipeline_parallel.partitioner import UniformPartitioner
+    stages = UniformPartitioner(ref, ctx).split(world_size)
+    ref_stage = stages[rank]
+    own_params = dict(pp.named_parameters())
+    for (name, p_ref) in ref_stage.named_parameters():
+        p = own_params["stage." + name]
+        assert p_ref.grad is not None
+        assert torch.allclose(p.grad, p_ref.grad, atol=1e-6), \
+            f"{name}: {(p.grad - p_ref.grad).abs().max()}"
+    ctx.destroy()
+
+
+@pytest.mark.parametrize("schedule", ["1f1b", "gpipe"])
+def test_pp2_engine_matches_oracle(schedule):
+    spawn(run_pp_engine, world_size=2, schedule=schedule)
+
+
+def run_pp_bloom(rank, world_size, port):
+    from pipegoose_amd.models.bloom import BloomForCausalLM, bloom_tiny
+    from pipegoose_amd.nn.pipeline_parallel import PipelineParallel
+
+    ctx = init_parallel_context(rank, world_size, port,
+                               pipeline_parallel_size=world_size)
+    torch.manual_seed(123)
+    model = BloomForCausalLM(bloom_tiny(), ctx)
+    ref = None
+    if rank == world_size - 1:
+        torch.manual_seed(123)
+        # need a fresh single-stage copy built with the same seed for parity
+        ref_state = {k: v.clone() for k, v in model.state_dict().items()}
+
+    def lm_loss(logits, labels):
+        import torch.nn.functional as TF
+        shift_logits = logits[:, :-1].reshape(-1, logits.size(-1)).float()
+        shift_labels = labels[:, 1:].reshape(-1)
+        return TF.cross_entropy(shift_logits, shift_labels)
+
+    torch.manual_seed(7)
+    ids = torch.randint(0, 256, (4, 16))
+
+    pp = PipelineParallel(model, ctx, n_microbatches=2,
+                          loss_fn=lm_loss).parallelize()
+    loss = pp(ids, ids)
+
+    if rank == world_size - 1:
+        torch.manual_seed(123)
+        ref_model = BloomForCausalLM(bloom_tiny(), ctx2_placeholder(ctx))
+        ref_model.load_state_dict(ref_state)
+        ref_loss = lm_loss(ref_model(ids), ids)
+        assert torch.allclose(loss, ref_loss, atol=1e-5), (loss, ref_loss)
+    ctx.destroy()
+
+
+def ctx2_placeholder(ctx):
+    return ctx  # tp=1 context works for reference construction
+
+
+def test_pp2_bloom_matches_oracle():
+    spawn(run_pp_bloom, world_size=2)
