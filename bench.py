@@ -30,19 +30,25 @@ def parse_args():
     p.add_argument("--micro-batch", type=int, default=8,
                    help="per-DP-rank batch size (weak scaling)")
     p.add_argument("--tp", type=int, default=0, help="0 = auto by world size")
+    p.add_argument("--pp", type=int, default=0)
     p.add_argument("--dp", type=int, default=0)
+    p.add_argument("--microbatches", type=int, default=4, help="pipeline microbatches")
     p.add_argument("--device", type=str, default=None)
     return p.parse_args()
 
 
-def pick_parallelism(world_size: int, tp_arg: int, dp_arg: int):
-    """BASELINE config: TP2xDP2 at 4 GPUs; weak-scale DP beyond."""
-    if tp_arg > 0:
-        tp = tp_arg
-        dp = dp_arg if dp_arg > 0 else world_size // tp
-        return tp, dp
-    table = {1: (1, 1), 2: (1, 2), 4: (2, 2), 8: (2, 4)}
-    return table.get(world_size, (2, world_size // 2))
+def pick_parallelism(world_size: int, model: str, tp_arg: int, pp_arg: int, dp_arg: int):
+    """BASELINE configs: bloom-560m TP2xDP2 @4; bloom-7b1 TP2xPP2xDP2 @8."""
+    if tp_arg > 0 or pp_arg > 0:
+        tp = max(tp_arg, 1)
+        pp = max(pp_arg, 1)
+        dp = dp_arg if dp_arg > 0 else world_size // (tp * pp)
+        return tp, pp, dp
+    if model == "bloom-7b1":
+        table = {1: (1, 1, 1), 2: (2, 1, 1), 4: (2, 2, 1), 8: (2, 2, 2)}
+    else:
+        table = {1: (1, 1, 1), 2: (1, 1, 2), 4: (2, 1, 2), 8: (2, 1, 4)}
+    return table.get(world_size, (2, 1, world_size // 2))
 
 
 def main():
@@ -52,16 +58,18 @@ def main():
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29571")
 
-    tp, dp = pick_parallelism(world_size, args.tp, args.dp)
+    tp, pp, dp = pick_parallelism(world_size, args.model, args.tp, args.pp, args.dp)
 
     from pipegoose_amd import ParallelContext, ParallelMode
     from pipegoose_amd.models.bloom import (
-        BloomForCausalLM, bloom_1b7, bloom_560m, bloom_7b1, bloom_tiny)
+        BloomForCausalLM, bloom_1b7, bloom_560m, bloom_7b1, bloom_tiny,
+        make_causal_lm_loss)
     from pipegoose_amd.nn import DataParallel
+    from pipegoose_amd.nn.pipeline_parallel import PipelineParallel
     from pipegoose_amd.optim import DistributedOptimizer
 
     ctx = ParallelContext.from_torch(
-        tensor_parallel_size=tp, pipeline_parallel_size=1, data_parallel_size=dp)
+        tensor_parallel_size=tp, pipeline_parallel_size=pp, data_parallel_size=dp)
 
     use_gpu = torch.cuda.is_available()
     device = torch.device(args.device) if args.device else ctx.device
@@ -71,7 +79,12 @@ def main():
            "bloom-7b1": bloom_7b1, "bloom-tiny": bloom_tiny}[args.model]()
 
     torch.manual_seed(1234)
-    model = BloomForCausalLM(cfg, ctx).to(device=device, dtype=dtype)
+    model = BloomForCausalLM(cfg, ctx)
+    if pp > 1:
+        model = PipelineParallel(
+            model, ctx, n_microbatches=args.microbatches,
+            loss_fn=make_causal_lm_loss(ctx)).parallelize()
+    model = model.to(device=device, dtype=dtype)
     if dp > 1:
         model = DataParallel(model, ctx).parallelize()
 
@@ -86,8 +99,12 @@ def main():
 
     def one_step():
         optim.zero_grad()
-        loss = model(input_ids, labels=input_ids)
-        loss.backward()
+        if pp > 1:
+            # the engine runs forward AND backward internally (1F1B)
+            loss = model(input_ids, input_ids)
+        else:
+            loss = model(input_ids, labels=input_ids)
+            loss.backward()
         optim.step()
         return loss
 
@@ -137,7 +154,8 @@ def main():
                 "model": args.model,
                 "global_batch": global_batch,
                 "seq_len": S,
-                "parallelism": f"tp{tp}dp{dp}",
+                "parallelism": f"tp{tp}pp{pp}dp{dp}"
+                               + (f"mb{args.microbatches}" if pp > 1 else ""),
             },
         }))
 

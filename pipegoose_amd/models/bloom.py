@@ -179,6 +179,30 @@ class BloomModel(nn.Module):
         return self.ln_f(hidden)
 
 
+def make_causal_lm_loss(parallel_context: ParallelContext):
+    """Shift-CE loss fn over (possibly vocab-sharded) logits — for the
+    pipeline engine's last stage."""
+    tp = parallel_context.get_world_size(ParallelMode.TENSOR)
+    vp_ce = VocabParallelCrossEntropy(parallel_context=parallel_context) \
+        if tp > 1 else None
+
+    def loss_fn(logits, labels):
+        shift_logits = logits[:, :-1].contiguous()
+        shift_labels = labels[:, 1:].contiguous()
+        if vp_ce is not None:
+            return vp_ce(shift_logits, shift_labels)
+        if logits.is_cuda:
+            from pipegoose_amd.ops.cross_entropy import fused_cross_entropy
+            return fused_cross_entropy(
+                shift_logits.reshape(-1, shift_logits.size(-1)),
+                shift_labels.reshape(-1))
+        return TF.cross_entropy(
+            shift_logits.float().reshape(-1, shift_logits.size(-1)),
+            shift_labels.reshape(-1))
+
+    return loss_fn
+
+
 class BloomForCausalLM(nn.Module):
     def __init__(self, config: BloomConfig, parallel_context: ParallelContext):
         super().__init__()
