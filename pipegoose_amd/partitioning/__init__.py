@@ -1,0 +1,3 @@
+from pipegoose_amd.partitioning.profile import ProfileByMemory
+
+__all__ = ["ProfileByMemory"]
