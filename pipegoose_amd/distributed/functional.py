@@ -153,20 +153,92 @@ def all_to_all(
     outputs = [torch.empty_like(inputs[0]) for _ in range(world_size)]
     rank = parallel_context.get_local_rank(parallel_mode)
     if dist.get_backend(group) == "gloo":
-        # gloo has no alltoall: pairwise isend/irecv (CPU test path only).
-        outputs[rank].copy_(inputs[rank])
-        global_ranks = parallel_context.get_ranks_in_group(parallel_mode)
-        works = []
+        # gloo has no alltoall (CPU test path only).  NOTE: not pairwise
+        # isend/irecv — gloo p2p tags share slot space with collective slots,
+        # so mixing them on one group corrupts message matching; emulate via
+        # all_gather instead.
+        stacked = torch.stack(inputs)  # [world, chunk...]
+        gathered = [torch.empty_like(stacked) for _ in range(world_size)]
+        dist.all_gather(gathered, stacked, group=group)
         for peer in range(world_size):
-            if peer == rank:
-                continue
-            works.append(dist.isend(inputs[peer], dst=global_ranks[peer], group=group, tag=rank))
-            works.append(dist.irecv(outputs[peer], src=global_ranks[peer], group=group, tag=peer))
-        for w in works:
-            w.wait()
+            outputs[peer].copy_(gathered[peer][rank])
     else:
         dist.all_to_all(outputs, inputs, group=group)
     return torch.cat(outputs, dim=out_dim)
+
+
+def exchange_splits(
+    in_splits: List[int],
+    parallel_context: ParallelContext = None,
+    parallel_mode: ParallelMode = ParallelMode.GLOBAL,
+) -> List[int]:
+    """Tell every peer how many dim-0 rows it will receive from this rank;
+    returns this rank's receive counts (one small int64 all-gather)."""
+    world_size = parallel_context.get_world_size(parallel_mode)
+    if world_size == 1:
+        return list(in_splits)
+    group = _group(parallel_context, parallel_mode)
+    rank = parallel_context.get_local_rank(parallel_mode)
+    dev = "cuda" if dist.get_backend(group) == "nccl" else "cpu"
+    counts_in = torch.tensor(in_splits, dtype=torch.int64, device=dev)
+    gathered = [torch.empty_like(counts_in) for _ in range(world_size)]
+    dist.all_gather(gathered, counts_in, group=group)
+    return [int(gathered[peer][rank]) for peer in range(world_size)]
+
+
+def all_to_all_variable(
+    tensor: torch.Tensor,
+    in_splits: List[int],
+    parallel_context: ParallelContext = None,
+    parallel_mode: ParallelMode = ParallelMode.GLOBAL,
+    out_splits: Optional[List[int]] = None,
+):
+    """Variable-split all-to-all along dim 0: rank r receives ``in_splits[r]``
+    rows from every peer.  Returns ``(output, out_splits)``.
+
+    The MoE token-dispatch primitive: on RCCL this is one
+    ``all_to_all_single`` (one fused xGMI exchange); on gloo (CPU test path)
+    the split sizes are exchanged first, then pairwise isend/irecv.
+    If ``out_splits`` is already known (e.g. the combine direction reversing a
+    dispatch) the size exchange is skipped.
+    """
+    world_size = parallel_context.get_world_size(parallel_mode)
+    assert len(in_splits) == world_size
+    if world_size == 1:
+        return tensor, list(in_splits)
+    group = _group(parallel_context, parallel_mode)
+    rank = parallel_context.get_local_rank(parallel_mode)
+    use_gloo = dist.get_backend(group) == "gloo"
+
+    if out_splits is None:
+        out_splits = exchange_splits(in_splits, parallel_context, parallel_mode)
+
+    tensor = tensor.contiguous()
+    out_shape = (sum(out_splits),) + tensor.shape[1:]
+    output = torch.empty(out_shape, dtype=tensor.dtype, device=tensor.device)
+
+    if use_gloo:
+        # CPU test path: emulate variable alltoall with a padded all_gather
+        # (gloo p2p tags collide with collective slots — see all_to_all).
+        inputs = list(torch.split(tensor, in_splits, dim=0))
+        outputs = list(torch.split(output, out_splits, dim=0))
+        pad_rows = torch.tensor(max(in_splits), dtype=torch.int64)
+        dist.all_reduce(pad_rows, op=ReduceOp.MAX, group=group)
+        pad_rows = int(pad_rows)
+        padded = torch.zeros((world_size, pad_rows) + tensor.shape[1:],
+                             dtype=tensor.dtype, device=tensor.device)
+        for peer in range(world_size):
+            if in_splits[peer]:
+                padded[peer, :in_splits[peer]] = inputs[peer]
+        gathered = [torch.empty_like(padded) for _ in range(world_size)]
+        dist.all_gather(gathered, padded, group=group)
+        for peer in range(world_size):
+            if out_splits[peer]:
+                outputs[peer].copy_(gathered[peer][rank, :out_splits[peer]])
+    else:
+        dist.all_to_all_single(output, tensor, output_split_sizes=out_splits,
+                               input_split_sizes=list(in_splits), group=group)
+    return output, out_splits
 
 
 def send(

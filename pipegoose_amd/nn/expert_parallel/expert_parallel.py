@@ -23,8 +23,10 @@ class ExpertParallel(Parallel):
         noise_policy: Union[str, Callable] = "gaussian",
         enable_tensor_parallel: bool = False,
         parallel_context: ParallelContext = None,
+        dispatch: str = "mask",
     ):
         super().__init__(module, parallel_context)
+        self.dispatch = dispatch
         tp_size = parallel_context.get_world_size(ParallelMode.TENSOR)
         if enable_tensor_parallel:
             assert num_experts % tp_size == 0, (
@@ -50,6 +52,7 @@ class ExpertParallel(Parallel):
             expert_layer = ExpertLayer(
                 self.num_experts, expert, self.router,
                 self.enable_tensor_parallel, self.parallel_context,
+                dispatch=self.dispatch,
             )
             _set_submodule(self.module, name, expert_layer)
             replaced += 1

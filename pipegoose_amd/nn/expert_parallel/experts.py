@@ -21,9 +21,12 @@ from pipegoose_amd.distributed.parallel_mode import ParallelMode
 
 class Experts(nn.Module):
     def __init__(self, num_experts: int, expert: nn.Module,
-                 enable_tensor_parallel: bool, parallel_context: ParallelContext):
+                 enable_tensor_parallel: bool, parallel_context: ParallelContext,
+                 dispatch: str = "mask"):
         super().__init__()
+        assert dispatch in ("mask", "alltoall")
         self.parallel_context = parallel_context
+        self.dispatch = dispatch
         tp_size = parallel_context.get_world_size(ParallelMode.TENSOR)
         if enable_tensor_parallel and tp_size > 1:
             assert num_experts % tp_size == 0
@@ -36,9 +39,25 @@ class Experts(nn.Module):
         self.num_experts = num_experts
         self.enable_tensor_parallel = enable_tensor_parallel and tp_size > 1
 
+        # share (don't copy) the ParallelContext a prototype module may hold:
+        # ProcessGroups aren't picklable and must stay process-wide singletons
+        memo = {id(parallel_context): parallel_context}
         self.experts = nn.ModuleList(
-            [copy.deepcopy(expert) for _ in range(self.num_local_experts)]
+            [copy.deepcopy(expert, dict(memo)) for _ in range(self.num_local_experts)]
         )
+        if self.enable_tensor_parallel:
+            # An expert must be a DENSE local module: the EP axis reuses the
+            # TENSOR group, so a TP-sharded expert would fire its internal
+            # all-reduces with per-rank-different token counts → collective
+            # mismatch.  (Native models build TP layers at construction;
+            # pass a plain nn.Module via ExpertParallel(expert=...).)
+            from pipegoose_amd.nn.tensor_parallel.linear import (
+                ColumnParallelLinear, RowParallelLinear)
+            for m in self.experts.modules():
+                if isinstance(m, (ColumnParallelLinear, RowParallelLinear)):
+                    raise TypeError(
+                        "expert prototype contains tensor-parallel layers; "
+                        "pass a dense expert module to ExpertParallel(expert=...)")
         for p in self.experts.parameters():
             setattr(p, "is_expert", True)
 
@@ -46,6 +65,8 @@ class Experts(nn.Module):
         # inputs: [B, S, H]; dispatch_order: [B*S] global expert index (top-1)
         shape = inputs.shape
         flat = inputs.reshape(-1, shape[-1])
+        if self.dispatch == "alltoall" and self.enable_tensor_parallel:
+            return self._forward_alltoall(flat, dispatch_order).reshape(shape)
         outputs = torch.zeros_like(flat)
         for local_idx, expert in enumerate(self.experts):
             global_idx = self.expert_offset + local_idx
@@ -59,6 +80,43 @@ class Experts(nn.Module):
         if self.enable_tensor_parallel:
             outputs = _AllReduceCombine.apply(outputs, self.parallel_context)
         return outputs.reshape(shape)
+
+    def _forward_alltoall(self, flat: torch.Tensor, dispatch_order: torch.Tensor):
+        """EP dispatch over xGMI.  The activation is replicated across the EP
+        group, so first each rank takes its 1/ep chunk of the token set (no
+        redundant expert FLOPs — the mask path recomputes every token on every
+        rank), all-to-alls its chunk to the expert-owner ranks, computes, and
+        all-to-alls back; a final dim-0 all-gather rebuilds the replicated
+        output.  Backward mirrors: chunk the output grad, reverse exchanges,
+        all-gather input grads — every rank ends with the FULL input gradient
+        (the mask path leaves per-rank partials that are only correct after a
+        downstream TP sum)."""
+        from pipegoose_amd.nn.expert_parallel.dispatch import AllToAllDispatcher
+        from pipegoose_amd.nn.tensor_parallel._functional import (
+            _Gather, _Scatter)
+        if not hasattr(self, "_dispatcher"):
+            self._dispatcher = AllToAllDispatcher(
+                self.num_experts, self.parallel_context, ParallelMode.TENSOR)
+        ep = self.parallel_context.get_world_size(ParallelMode.TENSOR)
+        rank = self.parallel_context.get_local_rank(ParallelMode.TENSOR)
+        assert flat.size(0) % ep == 0, \
+            f"token count {flat.size(0)} not divisible by ep={ep}"
+        chunk = _Scatter.apply(flat, 0, self.parallel_context)
+        route_chunk = dispatch_order.reshape(-1).chunk(ep, dim=0)[rank]
+        recv, local_idx, state = self._dispatcher.dispatch(chunk, route_chunk)
+        counts = torch.bincount(local_idx, minlength=self.num_local_experts).tolist()
+        outs = []
+        start = 0
+        for i, expert in enumerate(self.experts):
+            seg = recv[start:start + counts[i]]
+            out = expert(seg)
+            if isinstance(out, tuple):
+                out = out[0]
+            outs.append(out.to(recv.dtype))
+            start += counts[i]
+        expert_out = torch.cat(outs, dim=0) if outs else recv
+        combined = self._dispatcher.combine(expert_out, state)
+        return _Gather.apply(combined, 0, self.parallel_context)
 
 
 class _AllReduceCombine(torch.autograd.Function):

@@ -10,11 +10,13 @@ from pipegoose_amd.nn.expert_parallel.expert_context import ExpertContext
 
 class ExpertLayer(nn.Module):
     def __init__(self, num_experts: int, expert: nn.Module, router: nn.Module,
-                 enable_tensor_parallel: bool, parallel_context: ParallelContext):
+                 enable_tensor_parallel: bool, parallel_context: ParallelContext,
+                 dispatch: str = "mask"):
         super().__init__()
         from pipegoose_amd.nn.expert_parallel.experts import Experts
         self.router = router
-        self._experts = Experts(num_experts, expert, enable_tensor_parallel, parallel_context)
+        self._experts = Experts(num_experts, expert, enable_tensor_parallel,
+                                parallel_context, dispatch=dispatch)
         self.parallel_context = parallel_context
 
     @property
