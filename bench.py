@@ -33,6 +33,9 @@ def parse_args():
     p.add_argument("--pp", type=int, default=0)
     p.add_argument("--dp", type=int, default=0)
     p.add_argument("--microbatches", type=int, default=4, help="pipeline microbatches")
+    p.add_argument("--sp", action="store_true",
+                   help="Megatron-style sequence parallelism over the TP group "
+                        "(BASELINE config 5)")
     p.add_argument("--device", type=str, default=None)
     return p.parse_args()
 
@@ -78,6 +81,8 @@ def main():
     cfg = {"bloom-560m": bloom_560m, "bloom-1b7": bloom_1b7,
            "bloom-7b1": bloom_7b1, "bloom-tiny": bloom_tiny}[args.model]()
 
+    if args.sp:
+        cfg.sequence_parallel = True
     torch.manual_seed(1234)
     model = BloomForCausalLM(cfg, ctx)
     if pp > 1:
@@ -155,7 +160,8 @@ def main():
                 "global_batch": global_batch,
                 "seq_len": S,
                 "parallelism": f"tp{tp}pp{pp}dp{dp}"
-                               + (f"mb{args.microbatches}" if pp > 1 else ""),
+                               + (f"mb{args.microbatches}" if pp > 1 else "")
+                               + ("sp" if args.sp else ""),
             },
         }))
 

@@ -37,6 +37,10 @@ class BloomConfig:
     n_head: int = 16
     layer_norm_epsilon: float = 1e-5
     initializer_range: float = 0.02
+    # Megatron-style sequence parallelism (BASELINE config 5): LayerNorms and
+    # residuals run on [B, S/tp, H] shards; TP boundaries become
+    # all-gather / reduce-scatter along S.  Requires S % tp == 0.
+    sequence_parallel: bool = False
 
     @property
     def head_dim(self):
@@ -87,12 +91,13 @@ class BloomAttention(nn.Module):
 
         # QKV fused, grouped per head ([head][q|k|v][head_dim]) so the column
         # slice hands whole heads to each rank.
+        sp = config.sequence_parallel
         self.query_key_value = ColumnParallelLinear(
             config.hidden_size, 3 * config.hidden_size,
-            parallel_context=parallel_context)
+            sequence_parallel=sp, parallel_context=parallel_context)
         self.dense = RowParallelLinear(
             config.hidden_size, config.hidden_size,
-            parallel_context=parallel_context)
+            sequence_parallel=sp, parallel_context=parallel_context)
 
         slopes = alibi_slopes(config.n_head)
         local = slopes.chunk(tp)[tp_rank].clone()
@@ -117,8 +122,9 @@ class BloomAttention(nn.Module):
         return cached
 
     def forward(self, hidden: torch.Tensor) -> torch.Tensor:
-        B, S, _ = hidden.shape
+        B = hidden.size(0)
         fused = self.query_key_value(hidden)  # [B, S, local_heads * 3 * hd]
+        S = fused.size(1)  # full sequence (SP mode all-gathered inside qkv)
         fused = fused.view(B, S, self.num_heads, 3, self.head_dim)
         q = fused[..., 0, :].transpose(1, 2)  # [B, H, S, hd]
         k = fused[..., 1, :].transpose(1, 2)
@@ -135,8 +141,11 @@ class BloomMLP(nn.Module):
     def __init__(self, config: BloomConfig, parallel_context: ParallelContext):
         super().__init__()
         h = config.hidden_size
-        self.dense_h_to_4h = ColumnParallelLinear(h, 4 * h, parallel_context=parallel_context)
-        self.dense_4h_to_h = RowParallelLinear(4 * h, h, parallel_context=parallel_context)
+        sp = config.sequence_parallel
+        self.dense_h_to_4h = ColumnParallelLinear(
+            h, 4 * h, sequence_parallel=sp, parallel_context=parallel_context)
+        self.dense_4h_to_h = RowParallelLinear(
+            4 * h, h, sequence_parallel=sp, parallel_context=parallel_context)
 
     def forward(self, hidden: torch.Tensor) -> torch.Tensor:
         # fused GEMM + bias + GeLU epilogue on GPU (ops/fused_bias_gelu)
@@ -148,9 +157,14 @@ class BloomBlock(nn.Module):
     def __init__(self, config: BloomConfig, parallel_context: ParallelContext):
         super().__init__()
         eps = config.layer_norm_epsilon
-        self.input_layernorm = LayerNorm(config.hidden_size, eps=eps)
+        sp = config.sequence_parallel
+        self.input_layernorm = LayerNorm(
+            config.hidden_size, eps=eps, sequence_parallel=sp,
+            parallel_context=parallel_context if sp else None)
         self.self_attention = BloomAttention(config, parallel_context)
-        self.post_attention_layernorm = LayerNorm(config.hidden_size, eps=eps)
+        self.post_attention_layernorm = LayerNorm(
+            config.hidden_size, eps=eps, sequence_parallel=sp,
+            parallel_context=parallel_context if sp else None)
         self.mlp = BloomMLP(config, parallel_context)
 
     def forward(self, hidden: torch.Tensor) -> torch.Tensor:
@@ -165,12 +179,19 @@ class BloomModel(nn.Module):
         self.config = config
         eps = config.layer_norm_epsilon
         self.word_embeddings = ParallelEmbedding(
-            config.vocab_size, config.hidden_size, parallel_context=parallel_context)
-        self.word_embeddings_layernorm = LayerNorm(config.hidden_size, eps=eps)
+            config.vocab_size, config.hidden_size,
+            sequence_parallel=config.sequence_parallel,
+            parallel_context=parallel_context)
+        sp = config.sequence_parallel
+        self.word_embeddings_layernorm = LayerNorm(
+            config.hidden_size, eps=eps, sequence_parallel=sp,
+            parallel_context=parallel_context if sp else None)
         self.h = nn.ModuleList(
             [BloomBlock(config, parallel_context) for _ in range(config.n_layer)]
         )
-        self.ln_f = LayerNorm(config.hidden_size, eps=eps)
+        self.ln_f = LayerNorm(
+            config.hidden_size, eps=eps, sequence_parallel=sp,
+            parallel_context=parallel_context if sp else None)
 
     def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
         hidden = self.word_embeddings_layernorm(self.word_embeddings(input_ids))
@@ -214,7 +235,8 @@ class BloomForCausalLM(nn.Module):
         # embedding; logits stay sharded — the parallel CE consumes them.
         self.lm_head = ColumnParallelLinear(
             config.hidden_size, config.vocab_size, bias=False,
-            gather_output=False, parallel_context=parallel_context)
+            gather_output=False, sequence_parallel=config.sequence_parallel,
+            parallel_context=parallel_context)
         self.lm_head.weight = self.transformer.word_embeddings.weight
         self.loss_fn = VocabParallelCrossEntropy(parallel_context=parallel_context) \
             if tp > 1 else None

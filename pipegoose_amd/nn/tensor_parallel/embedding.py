@@ -11,14 +11,19 @@ import torch.nn.functional as TF
 
 from pipegoose_amd.distributed.parallel_context import ParallelContext
 from pipegoose_amd.distributed.parallel_mode import ParallelMode
-from pipegoose_amd.nn.tensor_parallel._functional import reduce_to_tensor_group
+from pipegoose_amd.nn.tensor_parallel._functional import (
+    reduce_scatter_sequence, reduce_to_tensor_group)
 
 
 class ParallelEmbedding(nn.Module):
     def __init__(self, num_embeddings: int, embedding_dim: int,
+                 sequence_parallel: bool = False,
                  parallel_context: ParallelContext = None):
         super().__init__()
         world = parallel_context.get_world_size(ParallelMode.TENSOR)
+        # SP: the partial-sum combine is a reduce-scatter along S, so the
+        # embedding emits the [B, S/tp, H] shard directly (Megatron-SP entry).
+        self.sequence_parallel = sequence_parallel and world > 1
         assert num_embeddings % world == 0
         rank = parallel_context.get_local_rank(ParallelMode.TENSOR)
         self.num_embeddings = num_embeddings
@@ -37,4 +42,6 @@ class ParallelEmbedding(nn.Module):
         masked_input = (input - self.vocab_start_idx).masked_fill(mask, 0)
         output = TF.embedding(masked_input, self.weight)
         output = output.masked_fill(mask.unsqueeze(-1), 0.0)
+        if self.sequence_parallel:
+            return reduce_scatter_sequence(output, self.parallel_context, dim=1)
         return reduce_to_tensor_group(output, self.parallel_context)

@@ -11,16 +11,29 @@ from pipegoose_amd.ops.layer_norm import fused_layer_norm
 
 
 class LayerNorm(nn.Module):
-    def __init__(self, normalized_shape, eps: float = 1e-5, parallel_context=None):
+    def __init__(self, normalized_shape, eps: float = 1e-5,
+                 sequence_parallel: bool = False, parallel_context=None):
         super().__init__()
         if isinstance(normalized_shape, int):
             normalized_shape = (normalized_shape,)
         self.normalized_shape = tuple(normalized_shape)
         self.eps = eps
         self.parallel_context = parallel_context
+        # SP: this norm sees only the local [B, S/tp, H] shard, so its
+        # (replicated) weight/bias grads are partial sums over local tokens;
+        # routing the params through _Broadcast (identity fwd, all-reduce bwd
+        # over TENSOR) restores the full gradient per backward — correct under
+        # microbatch accumulation too.
+        self.sequence_parallel = sequence_parallel and parallel_context is not None
         self.weight = nn.Parameter(torch.ones(self.normalized_shape))
         self.bias = nn.Parameter(torch.zeros(self.normalized_shape))
 
     def forward(self, input: torch.Tensor) -> torch.Tensor:
-        return fused_layer_norm(input, self.normalized_shape, self.weight, self.bias,
+        weight, bias = self.weight, self.bias
+        if self.sequence_parallel:
+            from pipegoose_amd.nn.tensor_parallel._functional import (
+                broadcast_to_tensor_group)
+            weight = broadcast_to_tensor_group(weight, self.parallel_context)
+            bias = broadcast_to_tensor_group(bias, self.parallel_context)
+        return fused_layer_norm(input, self.normalized_shape, weight, bias,
                                 self.eps)

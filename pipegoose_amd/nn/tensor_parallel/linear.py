@@ -14,8 +14,10 @@ import torch.nn.functional as TF
 from pipegoose_amd.distributed.parallel_context import ParallelContext
 from pipegoose_amd.distributed.parallel_mode import ParallelMode
 from pipegoose_amd.nn.tensor_parallel._functional import (
+    all_gather_sequence,
     broadcast_to_tensor_group,
     gather_to_tensor_group,
+    reduce_scatter_sequence,
     reduce_to_tensor_group,
     scatter_to_tensor_group,
 )
@@ -31,6 +33,7 @@ class ColumnParallelLinear(nn.Module):
         out_features: int,
         bias: bool = True,
         gather_output: bool = False,
+        sequence_parallel: bool = False,
         parallel_context: ParallelContext = None,
     ):
         super().__init__()
@@ -39,6 +42,10 @@ class ColumnParallelLinear(nn.Module):
         self.in_features = in_features
         self.out_features = out_features // world
         self.gather_output = gather_output
+        # Megatron-SP: input arrives sequence-sharded [B, S/tp, H]; forward
+        # all-gathers S, backward reduce-scatters the input grad (replacing
+        # _Broadcast's all-reduce — same wire bytes, sharded activations).
+        self.sequence_parallel = sequence_parallel and world > 1
         self.parallel_context = parallel_context
         self.weight = nn.Parameter(torch.empty(self.out_features, in_features))
         if bias:
@@ -47,7 +54,10 @@ class ColumnParallelLinear(nn.Module):
             self.register_parameter("bias", None)
 
     def forward(self, input: torch.Tensor) -> torch.Tensor:
-        input = broadcast_to_tensor_group(input, self.parallel_context)
+        if self.sequence_parallel:
+            input = all_gather_sequence(input, self.parallel_context, dim=1)
+        else:
+            input = broadcast_to_tensor_group(input, self.parallel_context)
         output = TF.linear(input, self.weight, self.bias)
         if self.gather_output:
             output = gather_to_tensor_group(output, dim=-1,
@@ -64,6 +74,7 @@ class RowParallelLinear(nn.Module):
         in_features: int,
         out_features: int,
         bias: bool = True,
+        sequence_parallel: bool = False,
         parallel_context: ParallelContext = None,
     ):
         super().__init__()
@@ -71,6 +82,9 @@ class RowParallelLinear(nn.Module):
         assert in_features % world == 0
         self.in_features = in_features // world
         self.out_features = out_features
+        # Megatron-SP: partial outputs reduce-scatter along S instead of
+        # all-reduce; result is the complete [B, S/tp, H] shard.
+        self.sequence_parallel = sequence_parallel and world > 1
         self.parallel_context = parallel_context
         self.weight = nn.Parameter(torch.empty(out_features, self.in_features))
         if bias:
@@ -84,6 +98,14 @@ class RowParallelLinear(nn.Module):
             input = scatter_to_tensor_group(input, dim=-1,
                                             parallel_context=self.parallel_context)
         output = TF.linear(input, self.weight)
+        if self.sequence_parallel:
+            output = reduce_scatter_sequence(output, self.parallel_context, dim=1)
+            if self.bias is not None:
+                # replicated bias sees only the S-shard: restore the full
+                # grad with an identity-fwd / all-reduce-bwd on the param
+                output = output + broadcast_to_tensor_group(
+                    self.bias, self.parallel_context)
+            return output
         output = reduce_to_tensor_group(output, self.parallel_context)
         if self.bias is not None:
             output = output + self.bias

@@ -38,8 +38,12 @@ def bias_gelu(input: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
 
 def fused_bias_gelu(column_linear, hidden: torch.Tensor) -> torch.Tensor:
     """Run a ColumnParallelLinear with its bias+GeLU fused into the epilogue."""
-    from pipegoose_amd.nn.tensor_parallel._functional import broadcast_to_tensor_group
-    hidden = broadcast_to_tensor_group(hidden, column_linear.parallel_context)
+    from pipegoose_amd.nn.tensor_parallel._functional import (
+        all_gather_sequence, broadcast_to_tensor_group)
+    if getattr(column_linear, "sequence_parallel", False):
+        hidden = all_gather_sequence(hidden, column_linear.parallel_context, dim=1)
+    else:
+        hidden = broadcast_to_tensor_group(hidden, column_linear.parallel_context)
     x = TF.linear(hidden, column_linear.weight)  # hipBLASLt GEMM, no bias
     if column_linear.bias is not None:
         return bias_gelu(x, column_linear.bias)
