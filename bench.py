@@ -36,6 +36,8 @@ def parse_args():
     p.add_argument("--sp", action="store_true",
                    help="Megatron-style sequence parallelism over the TP group "
                         "(BASELINE config 5)")
+    p.add_argument("--no-graph", action="store_true",
+                   help="disable hipGraph capture of the training step")
     p.add_argument("--device", type=str, default=None)
     return p.parse_args()
 
@@ -93,7 +95,12 @@ def main():
     if dp > 1:
         model = DataParallel(model, ctx).parallelize()
 
-    optim = torch.optim.AdamW(model.parameters(), lr=1e-4, betas=(0.9, 0.95))
+    # hipGraph capture: single-rank path only (no RCCL inside the graph);
+    # kills per-kernel launch latency on the launch-bound small models.
+    use_graph = (use_gpu and world_size == 1 and pp == 1 and dp == 1
+                 and not args.no_graph)
+    optim = torch.optim.AdamW(model.parameters(), lr=1e-4, betas=(0.9, 0.95),
+                              foreach=True, capturable=use_graph)
     if dp > 1:
         optim = DistributedOptimizer(optim, ctx)
 
@@ -102,8 +109,8 @@ def main():
     g = torch.Generator().manual_seed(4242 + rank)
     input_ids = torch.randint(0, cfg.vocab_size, (B, S), generator=g).to(device)
 
-    def one_step():
-        optim.zero_grad()
+    def one_step(set_to_none: bool = True):
+        optim.zero_grad(set_to_none=set_to_none)
         if pp > 1:
             # the engine runs forward AND backward internally (1F1B)
             loss = model(input_ids, input_ids)
@@ -113,8 +120,21 @@ def main():
         optim.step()
         return loss
 
-    for _ in range(args.warmup):
-        one_step()
+    step_fn = one_step
+    if use_graph:
+        # grads must be stable buffers across replays: allocate them once,
+        # then capture with set_to_none=False
+        for _ in range(max(args.warmup, 2)):
+            one_step(set_to_none=False)
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            one_step(set_to_none=False)
+        step_fn = graph.replay
+        torch.cuda.synchronize()
+    else:
+        for _ in range(args.warmup):
+            one_step()
 
     if dist.is_initialized() and world_size > 1:
         dist.barrier()
@@ -122,7 +142,7 @@ def main():
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        one_step()
+        step_fn()
     if use_gpu:
         torch.cuda.synchronize()
     if dist.is_initialized() and world_size > 1:

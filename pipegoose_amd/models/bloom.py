@@ -168,9 +168,11 @@ class BloomBlock(nn.Module):
         self.mlp = BloomMLP(config, parallel_context)
 
     def forward(self, hidden: torch.Tensor) -> torch.Tensor:
-        hidden = hidden + self.self_attention(self.input_layernorm(hidden))
-        hidden = hidden + self.mlp(self.post_attention_layernorm(hidden))
-        return hidden
+        attn_out = self.self_attention(self.input_layernorm(hidden))
+        # residual add fused into the post-attention norm's HBM pass
+        normed, hidden = self.post_attention_layernorm.forward_with_residual(
+            attn_out, hidden)
+        return hidden + self.mlp(normed)
 
 
 class BloomModel(nn.Module):

@@ -7,7 +7,7 @@ on CPU it falls back to torch.nn.functional.layer_norm.
 import torch
 from torch import nn
 
-from pipegoose_amd.ops.layer_norm import fused_layer_norm
+from pipegoose_amd.ops.layer_norm import fused_add_layer_norm, fused_layer_norm
 
 
 class LayerNorm(nn.Module):
@@ -28,12 +28,23 @@ class LayerNorm(nn.Module):
         self.weight = nn.Parameter(torch.ones(self.normalized_shape))
         self.bias = nn.Parameter(torch.zeros(self.normalized_shape))
 
-    def forward(self, input: torch.Tensor) -> torch.Tensor:
+    def _params(self):
         weight, bias = self.weight, self.bias
         if self.sequence_parallel:
             from pipegoose_amd.nn.tensor_parallel._functional import (
                 broadcast_to_tensor_group)
             weight = broadcast_to_tensor_group(weight, self.parallel_context)
             bias = broadcast_to_tensor_group(bias, self.parallel_context)
+        return weight, bias
+
+    def forward(self, input: torch.Tensor) -> torch.Tensor:
+        weight, bias = self._params()
         return fused_layer_norm(input, self.normalized_shape, weight, bias,
                                 self.eps)
+
+    def forward_with_residual(self, input: torch.Tensor, residual):
+        """(LN(input + residual), input + residual) — residual add fused into
+        the norm's HBM pass on GPU; residual may be None."""
+        weight, bias = self._params()
+        return fused_add_layer_norm(input, residual, self.normalized_shape,
+                                    weight, bias, self.eps)

@@ -6,6 +6,9 @@ std::vector<torch::Tensor> layer_norm_fwd(torch::Tensor x, torch::Tensor w,
 std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
                                           torch::Tensor w, torch::Tensor mean,
                                           torch::Tensor rstd);
+std::vector<torch::Tensor> layer_norm_res_fwd(torch::Tensor x, torch::Tensor r,
+                                              torch::Tensor w, torch::Tensor b,
+                                              double eps);
 torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias);
 torch::Tensor bias_gelu_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor bias);
 std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits, torch::Tensor targets,
@@ -18,6 +21,8 @@ torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor targets,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
     m.def("layer_norm_bwd", &layer_norm_bwd, "fused LayerNorm backward (gfx950)");
+    m.def("layer_norm_res_fwd", &layer_norm_res_fwd,
+          "fused residual-add + LayerNorm forward (gfx950)");
     m.def("bias_gelu_fwd", &bias_gelu_fwd, "fused bias+GeLU forward (gfx950)");
     m.def("bias_gelu_bwd", &bias_gelu_bwd, "fused bias+GeLU backward (gfx950)");
     m.def("cross_entropy_fwd", &cross_entropy_fwd,

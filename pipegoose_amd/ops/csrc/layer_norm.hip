@@ -288,3 +288,119 @@ std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
     HIP_CHECK_LAUNCH();
     return {dx, dw.to(x.scalar_type()), db.to(x.scalar_type())};
 }
+
+// ------------------------------------------------------------------------
+// Residual-fused forward: s = x + r; y = LN(s).  One HBM pass instead of a
+// separate elementwise add kernel (the add showed up as ~2.6% of step time,
+// profiles/bloom560m_1gpu_kernel_stats_r01.md); backward reuses
+// layer_norm_bwd with x := s.
+namespace {
+
+template <typename T, int BLOCK>
+__global__ void layer_norm_res_fwd_kernel(
+    const T* __restrict__ x, const T* __restrict__ r,
+    const T* __restrict__ w, const T* __restrict__ b,
+    T* __restrict__ s, T* __restrict__ y,
+    float* __restrict__ mean_out, float* __restrict__ rstd_out,
+    int H, float eps) {
+    __shared__ float smem[BLOCK / WAVE_SIZE];
+    const int64_t row = blockIdx.x;
+    const T* xr = x + row * (int64_t)H;
+    const T* rr = r + row * (int64_t)H;
+    T* sr = s + row * (int64_t)H;
+    T* yr = y + row * (int64_t)H;
+
+    using V = typename vec8<T>::type;
+    const int HV = H / 8;
+
+    float sum = 0.f, sumsq = 0.f;
+    for (int i = threadIdx.x; i < HV; i += BLOCK) {
+        V px = reinterpret_cast<const V*>(xr)[i];
+        V pr = reinterpret_cast<const V*>(rr)[i];
+        V ps;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float f = to_float(px.v[j]) + to_float(pr.v[j]);
+            ps.v[j] = from_float<T>(f);
+            // recompute from the rounded store so saved s EXACTLY matches
+            // what backward will read (bit-stable for bf16)
+            float fs = to_float(ps.v[j]);
+            sum += fs;
+            sumsq += fs * fs;
+        }
+        reinterpret_cast<V*>(sr)[i] = ps;
+    }
+    for (int i = HV * 8 + threadIdx.x; i < H; i += BLOCK) {
+        T sv = from_float<T>(to_float(xr[i]) + to_float(rr[i]));
+        sr[i] = sv;
+        float f = to_float(sv);
+        sum += f;
+        sumsq += f * f;
+    }
+    sum = block_reduce_sum(sum, smem);
+    sumsq = block_reduce_sum(sumsq, smem);
+
+    const float mean = sum / H;
+    const float var = sumsq / H - mean * mean;
+    const float rstd = rsqrtf(var + eps);
+    if (threadIdx.x == 0) {
+        mean_out[row] = mean;
+        rstd_out[row] = rstd;
+    }
+
+    for (int i = threadIdx.x; i < HV; i += BLOCK) {
+        V ps = reinterpret_cast<const V*>(sr)[i];
+        V wp = reinterpret_cast<const V*>(w)[i];
+        V bp = reinterpret_cast<const V*>(b)[i];
+        V out;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float xhat = (to_float(ps.v[j]) - mean) * rstd;
+            out.v[j] = from_float<T>(xhat * to_float(wp.v[j]) + to_float(bp.v[j]));
+        }
+        reinterpret_cast<V*>(yr)[i] = out;
+    }
+    for (int i = HV * 8 + threadIdx.x; i < H; i += BLOCK) {
+        float xhat = (to_float(sr[i]) - mean) * rstd;
+        yr[i] = from_float<T>(xhat * to_float(w[i]) + to_float(b[i]));
+    }
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> layer_norm_res_fwd(torch::Tensor x, torch::Tensor r,
+                                              torch::Tensor w, torch::Tensor b,
+                                              double eps) {
+    TORCH_CHECK(x.is_cuda() && x.is_contiguous() && r.is_contiguous());
+    TORCH_CHECK(x.sizes() == r.sizes());
+    const int64_t H = x.size(-1);
+    const int64_t N = x.numel() / H;
+    auto s = torch::empty_like(x);
+    auto y = torch::empty_like(x);
+    auto mean = torch::empty({N}, x.options().dtype(torch::kFloat));
+    auto rstd = torch::empty({N}, x.options().dtype(torch::kFloat));
+
+    constexpr int BLOCK = 256;
+    dim3 grid(N);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    if (x.scalar_type() == torch::kBFloat16) {
+        hipLaunchKernelGGL((layer_norm_res_fwd_kernel<__hip_bfloat16, BLOCK>), grid, dim3(BLOCK), 0, stream,
+            reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+            reinterpret_cast<const __hip_bfloat16*>(r.data_ptr()),
+            reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+            reinterpret_cast<const __hip_bfloat16*>(b.data_ptr()),
+            reinterpret_cast<__hip_bfloat16*>(s.data_ptr()),
+            reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
+            mean.data_ptr<float>(), rstd.data_ptr<float>(), (int)H, (float)eps);
+    } else if (x.scalar_type() == torch::kFloat) {
+        hipLaunchKernelGGL((layer_norm_res_fwd_kernel<float, BLOCK>), grid, dim3(BLOCK), 0, stream,
+            x.data_ptr<float>(), r.data_ptr<float>(),
+            w.data_ptr<float>(), b.data_ptr<float>(),
+            s.data_ptr<float>(), y.data_ptr<float>(),
+            mean.data_ptr<float>(), rstd.data_ptr<float>(), (int)H, (float)eps);
+    } else {
+        TORCH_CHECK(false, "layer_norm_res_fwd: unsupported dtype");
+    }
+    HIP_CHECK_LAUNCH();
+    return {y, s, mean, rstd};
+}

@@ -173,3 +173,52 @@ def test_alibi_fold_matches_mask():
                                                            is_causal=True, scale=1.0)
     err = (out.float() - ref).abs().max()
     assert err < 0.06, err
+
+
+@pytest.mark.parametrize("shape", [(4, 128, 1024), (3, 17, 4096)])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_layer_norm_res_fwd(shape, dtype):
+    """Fused residual-add + LayerNorm vs eager add + fp32 LN."""
+    ext = _ext()
+    torch.manual_seed(3)
+    H = shape[-1]
+    x = torch.randn(shape, device="cuda", dtype=dtype)
+    r = torch.randn(shape, device="cuda", dtype=dtype)
+    w = torch.randn(H, device="cuda", dtype=dtype)
+    b = torch.randn(H, device="cuda", dtype=dtype)
+    y, s, mean, rstd = ext.layer_norm_res_fwd(x.contiguous(), r.contiguous(),
+                                              w, b, 1e-5)
+    s_ref = (x + r)
+    assert torch.allclose(s.float(), s_ref.float(), atol=1e-6)
+    ref = torch.nn.functional.layer_norm(
+        s_ref.float(), (H,), w.float(), b.float(), 1e-5)
+    if dtype == torch.float32:
+        assert torch.allclose(y, ref, atol=1e-5)
+    else:
+        assert torch.allclose(y.float(), ref, rtol=1e-2, atol=2e-2)
+
+
+def test_fused_add_layer_norm_autograd():
+    """Autograd wrapper: grads wrt both branches equal eager reference."""
+    from pipegoose_amd.ops.layer_norm import fused_add_layer_norm
+    torch.manual_seed(4)
+    H = 512
+    x = torch.randn(2, 16, H, device="cuda", requires_grad=True)
+    r = torch.randn(2, 16, H, device="cuda", requires_grad=True)
+    w = torch.randn(H, device="cuda", requires_grad=True)
+    b = torch.randn(H, device="cuda", requires_grad=True)
+    y, s = fused_add_layer_norm(x, r, (H,), w, b, 1e-5)
+    out = (y * 1.3).sum() + (s * 0.7).sum()   # both outputs used downstream
+    out.backward()
+
+    x2 = x.detach().clone().requires_grad_(True)
+    r2 = r.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    s2 = x2 + r2
+    y2 = torch.nn.functional.layer_norm(s2, (H,), w2, b2, 1e-5)
+    ((y2 * 1.3).sum() + (s2 * 0.7).sum()).backward()
+
+    for a, bb in ((x, x2), (r, r2), (w, w2), (b, b2)):
+        assert torch.allclose(a.grad, bb.grad, atol=1e-3), \
+            (a.grad - bb.grad).abs().max()
