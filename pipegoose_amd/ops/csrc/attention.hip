@@ -1,0 +1,654 @@
+// Flash attention (causal + ALiBi) for gfx950 — hand-written CDNA4 MFMA.
+//
+// Replaces the sdpa-with-bias fallback: the ALiBi bias slope*(j-i) is
+// computed IN-KERNEL from per-head slopes, so no [H,S,S] bias tensor is
+// materialized or re-read from HBM (that tensor dominated the sdpa path's
+// memory traffic).  Online-softmax (flash2) with fp32 running stats.
+//
+// Tiling: one workgroup = 4 waves = 256 threads handles a 64-row Q block;
+// each wave owns 16 q rows.  K/V tiles (64 rows) stage through LDS; V is
+// stored transposed so the P·V MFMA's B-fragment reads are contiguous 16B.
+// MFMA: v_mfma_f32_16x16x32_bf16 (A/B: 8 bf16/lane; C/D: 4 fp32/lane,
+// row=(lane>>4)*4+reg, col=lane&15 — guide §3 fragment layout).
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "common.h"
+
+namespace {
+
+using bf16 = __hip_bfloat16;
+using frag_ab = __attribute__((ext_vector_type(8))) __bf16;   // 8 bf16 = 4 VGPR
+using frag_cd = __attribute__((ext_vector_type(4))) float;    // 4 fp32
+
+__device__ __forceinline__ frag_cd MFMA_16x16x32(frag_ab a, frag_ab b, frag_cd c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+constexpr int BLOCK_M = 64;   // q rows per workgroup
+constexpr int BLOCK_N = 64;   // kv rows per LDS tile
+constexpr int NWAVES = 4;     // BLOCK_M / 16
+constexpr int PAD = 8;        // bf16 elements (16 B) of row padding in LDS
+constexpr float NEG_INF = -1e30f;
+
+// A-fragment (M=16 side): row = lane&15, k = 8*(lane>>4) + i
+// B-fragment (N=16 side): col = lane&15, k = 8*(lane>>4) + i
+// C/D:                    col = lane&15, row = 4*(lane>>4) + reg
+
+template <int D>
+__global__ __launch_bounds__(256)
+void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
+                     const bf16* __restrict__ v,
+                     const float* __restrict__ slopes, float scale,
+                     bf16* __restrict__ o, float* __restrict__ lse,
+                     int B, int H, int S) {
+    constexpr int DCH = D / 32;    // K-chunks per QK^T mfma row
+    constexpr int DSUB = D / 16;   // output d-subtiles
+    constexpr int KSTRIDE = D + PAD;
+    constexpr int VSTRIDE = BLOCK_N + PAD;
+
+    __shared__ bf16 k_lds[BLOCK_N * KSTRIDE];
+    __shared__ bf16 vt_lds[D * VSTRIDE];
+    __shared__ bf16 p_lds[NWAVES][16 * VSTRIDE];
+
+    const int qblock = blockIdx.x;
+    const int h = blockIdx.y;
+    const int b = blockIdx.z;
+    const int tid = threadIdx.x;
+    const int wave = tid / WAVE_SIZE;
+    const int lane = tid % WAVE_SIZE;
+    const int lgrp = lane >> 4;       // 0..3
+    const int lcol = lane & 15;       // 0..15
+
+    const int64_t bh_off = ((int64_t)b * H + h) * S;
+    const bf16* qp = q + (bh_off) * D;
+    const bf16* kp = k + (bh_off) * D;
+    const bf16* vp = v + (bh_off) * D;
+
+    const int qrow0 = qblock * BLOCK_M + wave * 16;
+    const float slope = slopes[h];
+
+    // ---- Q fragments: stay in registers for the whole kernel
+    frag_ab aQ[DCH];
+    {
+        const bf16* qrow = qp + (int64_t)(qrow0 + lcol) * D;
+#pragma unroll
+        for (int c = 0; c < DCH; ++c) {
+            aQ[c] = *reinterpret_cast<const frag_ab*>(qrow + c * 32 + 8 * lgrp);
+        }
+    }
+
+    frag_cd accO[DSUB];
+#pragma unroll
+    for (int s = 0; s < DSUB; ++s) accO[s] = frag_cd{0.f, 0.f, 0.f, 0.f};
+    float m_run[4] = {NEG_INF, NEG_INF, NEG_INF, NEG_INF};
+    float l_run[4] = {0.f, 0.f, 0.f, 0.f};
+
+    const int n_kv_blocks = qblock + 1;  // causal
+    for (int nb = 0; nb < n_kv_blocks; ++nb) {
+        const int kvrow0 = nb * BLOCK_N;
+        __syncthreads();
+        // ---- cooperative K tile load (row-major, 16B packets)
+        {
+            constexpr int PACKETS = BLOCK_N * D / 8;
+            for (int p = tid; p < PACKETS; p += 256) {
+                const int row = p / (D / 8);
+                const int col = (p % (D / 8)) * 8;
+                *reinterpret_cast<frag_ab*>(&k_lds[row * KSTRIDE + col]) =
+                    *reinterpret_cast<const frag_ab*>(
+                        kp + (int64_t)(kvrow0 + row) * D + col);
+            }
+        }
+        // ---- cooperative V tile load, transposed into LDS
+        {
+            constexpr int PACKETS = BLOCK_N * D / 8;
+            for (int p = tid; p < PACKETS; p += 256) {
+                const int row = p / (D / 8);        // kv row
+                const int col = (p % (D / 8)) * 8;  // d offset
+                frag_ab pkt = *reinterpret_cast<const frag_ab*>(
+                    vp + (int64_t)(kvrow0 + row) * D + col);
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    vt_lds[(col + j) * VSTRIDE + row] = (bf16)pkt[j];
+                }
+            }
+        }
+        __syncthreads();
+
+        // ---- S = Q K^T for this wave's 16 rows x 64 cols
+        float s_tile[4][4];  // [nsub][reg]
+#pragma unroll
+        for (int ns = 0; ns < 4; ++ns) {
+            frag_cd acc = frag_cd{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int c = 0; c < DCH; ++c) {
+                frag_ab bK = *reinterpret_cast<const frag_ab*>(
+                    &k_lds[(ns * 16 + lcol) * KSTRIDE + c * 32 + 8 * lgrp]);
+                acc = MFMA_16x16x32(aQ[c], bK, acc);
+            }
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int iq = qrow0 + 4 * lgrp + r;
+                const int jk = kvrow0 + ns * 16 + lcol;
+                float sv = acc[r] * scale + slope * (float)(jk - iq);
+                s_tile[ns][r] = (jk <= iq) ? sv : NEG_INF;
+            }
+        }
+
+        // ---- online softmax update (row stats live in 4 regs per lane)
+        float m_new[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            float mx = fmaxf(fmaxf(s_tile[0][r], s_tile[1][r]),
+                             fmaxf(s_tile[2][r], s_tile[3][r]));
+#pragma unroll
+            for (int off = 1; off < 16; off <<= 1) {
+                mx = fmaxf(mx, __shfl_xor(mx, off, WAVE_SIZE));
+            }
+            m_new[r] = fmaxf(m_run[r], mx);
+        }
+        float rescale[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            rescale[r] = __expf(m_run[r] - m_new[r]);
+            m_run[r] = m_new[r];
+            l_run[r] *= rescale[r];
+        }
+#pragma unroll
+        for (int s = 0; s < DSUB; ++s) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) accO[s][r] *= rescale[r];
+        }
+
+        // ---- P = exp(S - m); accumulate l; stage P (bf16) for the PV mfma
+#pragma unroll
+        for (int ns = 0; ns < 4; ++ns) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                float pv = __expf(s_tile[ns][r] - m_run[r]);
+                s_tile[ns][r] = pv;
+                p_lds[wave][(4 * lgrp + r) * VSTRIDE + ns * 16 + lcol] =
+                    (bf16)pv;
+            }
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            float ls = s_tile[0][r] + s_tile[1][r] + s_tile[2][r] + s_tile[3][r];
+#pragma unroll
+            for (int off = 1; off < 16; off <<= 1) {
+                ls += __shfl_xor(ls, off, WAVE_SIZE);
+            }
+            l_run[r] += ls;
+        }
+        // LDS fence within the workgroup: P writes (own wave) + V/K reuse
+        __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0) & vmcnt(0)
+
+        // ---- O += P V  (K = 64 in 2 chunks of 32)
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+            frag_ab aP = *reinterpret_cast<const frag_ab*>(
+                &p_lds[wave][lcol * VSTRIDE + kc * 32 + 8 * lgrp]);
+#pragma unroll
+            for (int ds = 0; ds < DSUB; ++ds) {
+                frag_ab bV = *reinterpret_cast<const frag_ab*>(
+                    &vt_lds[(ds * 16 + lcol) * VSTRIDE + kc * 32 + 8 * lgrp]);
+                accO[ds] = MFMA_16x16x32(aP, bV, accO[ds]);
+            }
+        }
+    }
+
+    // ---- epilogue: O /= l, store O and lse
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        const int iq = qrow0 + 4 * lgrp + r;
+        const float inv_l = 1.0f / l_run[r];
+        bf16* orow = o + (bh_off + iq) * D;
+#pragma unroll
+        for (int ds = 0; ds < DSUB; ++ds) {
+            orow[ds * 16 + lcol] = (bf16)(accO[ds][r] * inv_l);
+        }
+        if (lcol == 0) {
+            lse[bh_off + iq] = m_run[r] + __logf(l_run[r]);
+        }
+    }
+}
+
+// ------------------------------------------------------------------ probe
+// Verifies the assumed A/B fragment layouts: C[16][16] = A[16][32]·B[32][16]
+// with A,B read from global memory exactly the way the attention kernel
+// reads its fragments.  One wave.
+__global__ void mfma_probe_kernel(const bf16* __restrict__ A,
+                                  const bf16* __restrict__ Bt,
+                                  float* __restrict__ C) {
+    const int lane = threadIdx.x;
+    const int lgrp = lane >> 4, lcol = lane & 15;
+    // A[m][k]: m = lcol, k = 8*lgrp + i  (row-major A: 16x32)
+    frag_ab a = *reinterpret_cast<const frag_ab*>(A + lcol * 32 + 8 * lgrp);
+    // B[k][n] with Bt stored as [n][k] (row-major 16x32): n = lcol, k = 8*lgrp+i
+    frag_ab b = *reinterpret_cast<const frag_ab*>(Bt + lcol * 32 + 8 * lgrp);
+    frag_cd c = MFMA_16x16x32(a, b, frag_cd{0.f, 0.f, 0.f, 0.f});
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        C[(4 * lgrp + r) * 16 + lcol] = c[r];
+    }
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, torch::Tensor slopes,
+                                    double scale) {
+    TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
+                v.is_contiguous());
+    TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "attn_fwd: bf16 only");
+    const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+    TORCH_CHECK(S % BLOCK_M == 0, "attn_fwd: S must be a multiple of 64");
+    TORCH_CHECK(D == 64 || D == 128, "attn_fwd: head dim 64 or 128");
+    TORCH_CHECK(slopes.numel() == H && slopes.scalar_type() == torch::kFloat);
+
+    auto o = torch::empty_like(q);
+    auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
+
+    dim3 grid(S / BLOCK_M, H, B);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    if (D == 64) {
+        hipLaunchKernelGGL((attn_fwd_kernel<64>), grid, dim3(256), 0, stream,
+            reinterpret_cast<const bf16*>(q.data_ptr()),
+            reinterpret_cast<const bf16*>(k.data_ptr()),
+            reinterpret_cast<const bf16*>(v.data_ptr()),
+            slopes.data_ptr<float>(), (float)scale,
+            reinterpret_cast<bf16*>(o.data_ptr()), lse.data_ptr<float>(),
+            B, H, S);
+    } else {
+        hipLaunchKernelGGL((attn_fwd_kernel<128>), grid, dim3(256), 0, stream,
+            reinterpret_cast<const bf16*>(q.data_ptr()),
+            reinterpret_cast<const bf16*>(k.data_ptr()),
+            reinterpret_cast<const bf16*>(v.data_ptr()),
+            slopes.data_ptr<float>(), (float)scale,
+            reinterpret_cast<bf16*>(o.data_ptr()), lse.data_ptr<float>(),
+            B, H, S);
+    }
+    HIP_CHECK_LAUNCH();
+    return {o, lse};
+}
+
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor Bt) {
+    TORCH_CHECK(A.is_cuda() && A.scalar_type() == torch::kBFloat16);
+    TORCH_CHECK(A.sizes() == torch::IntArrayRef({16, 32}) &&
+                Bt.sizes() == torch::IntArrayRef({16, 32}));
+    auto C = torch::empty({16, 16}, A.options().dtype(torch::kFloat));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream,
+        reinterpret_cast<const bf16*>(A.contiguous().data_ptr()),
+        reinterpret_cast<const bf16*>(Bt.contiguous().data_ptr()),
+        C.data_ptr<float>());
+    HIP_CHECK_LAUNCH();
+    return C;
+}
+
+// ======================================================================
+// Backward — two-pass flash2 (no atomics):
+//   delta:  delta[m] = rowsum(dO * O)
+//   pass A (kv-outer): dK, dV — one workgroup per 64-row kv block,
+//                      iterating q blocks >= kv block (causal)
+//   pass B (q-outer):  dQ — one workgroup per 64-row q block,
+//                      iterating kv blocks <= q block
+// Both passes recompute S and P from (q, k, lse): MFMA FLOPs are cheap
+// next to re-reading P from HBM.
+namespace {
+
+template <int D>
+__global__ __launch_bounds__(256)
+void attn_bwd_delta_kernel(const bf16* __restrict__ dout,
+                           const bf16* __restrict__ o,
+                           float* __restrict__ delta, int64_t rows) {
+    const int64_t row = (int64_t)blockIdx.x * 4 + threadIdx.x / WAVE_SIZE;
+    if (row >= rows) return;
+    const int lane = threadIdx.x % WAVE_SIZE;
+    const bf16* dr = dout + row * D;
+    const bf16* orow = o + row * D;
+    float acc = 0.f;
+    for (int i = lane; i < D; i += WAVE_SIZE) {
+        acc += to_float(dr[i]) * to_float(orow[i]);
+    }
+    acc = wave_reduce_sum(acc);
+    if (lane == 0) delta[row] = acc;
+}
+
+// pass A: dK(n,:) = scale * sum_m dS(m,n) Q(m,:);  dV(n,:) = sum_m P(m,n) dO(m,:)
+template <int D>
+__global__ __launch_bounds__(256)
+void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
+                          const bf16* __restrict__ q, const bf16* __restrict__ k,
+                          const bf16* __restrict__ v,
+                          const float* __restrict__ lse,
+                          const float* __restrict__ delta,
+                          const float* __restrict__ slopes, float scale,
+                          bf16* __restrict__ dk, bf16* __restrict__ dv,
+                          int B, int H, int S) {
+    constexpr int DCH = D / 32;
+    constexpr int DSUB = D / 16;
+    constexpr int RSTRIDE = D + PAD;        // row-major tiles [64][D+PAD]
+    constexpr int TSTRIDE = BLOCK_M + PAD;  // transposed tiles [D][64+PAD]
+
+    __shared__ bf16 do_lds[BLOCK_M * RSTRIDE];
+    __shared__ bf16 q_lds[BLOCK_M * RSTRIDE];
+    __shared__ bf16 qt_lds[D * TSTRIDE];
+    __shared__ bf16 dot_lds[D * TSTRIDE];
+    __shared__ bf16 pt_lds[NWAVES][16 * TSTRIDE];   // P^T  (n rows, m cols)
+    __shared__ bf16 dst_lds[NWAVES][16 * TSTRIDE];  // dS^T (n rows, m cols)
+    __shared__ float lse_lds[BLOCK_M];
+    __shared__ float delta_lds[BLOCK_M];
+
+    const int nb = blockIdx.x;
+    const int h = blockIdx.y;
+    const int b = blockIdx.z;
+    const int tid = threadIdx.x;
+    const int wave = tid / WAVE_SIZE;
+    const int lane = tid % WAVE_SIZE;
+    const int lgrp = lane >> 4, lcol = lane & 15;
+
+    const int64_t bh_off = ((int64_t)b * H + h) * S;
+    const bf16* qp = q + bh_off * D;
+    const bf16* dop = dout + bh_off * D;
+    const float slope = slopes[h];
+
+    const int kv0 = nb * BLOCK_N + wave * 16;  // this wave's 16 kv rows
+
+    // this wave's K and V rows live in registers (A-fragments over d)
+    frag_ab aK[DCH], aV[DCH];
+    {
+        const bf16* krow = k + (bh_off + kv0 + lcol) * D;
+        const bf16* vrow = v + (bh_off + kv0 + lcol) * D;
+#pragma unroll
+        for (int c = 0; c < DCH; ++c) {
+            aK[c] = *reinterpret_cast<const frag_ab*>(krow + c * 32 + 8 * lgrp);
+            aV[c] = *reinterpret_cast<const frag_ab*>(vrow + c * 32 + 8 * lgrp);
+        }
+    }
+
+    frag_cd accDK[DSUB], accDV[DSUB];
+#pragma unroll
+    for (int s = 0; s < DSUB; ++s) {
+        accDK[s] = frag_cd{0.f, 0.f, 0.f, 0.f};
+        accDV[s] = frag_cd{0.f, 0.f, 0.f, 0.f};
+    }
+
+    for (int qb = nb; qb < S / BLOCK_M; ++qb) {
+        const int q0 = qb * BLOCK_M;
+        __syncthreads();
+        {   // stage Q and dO, row-major + transposed, plus lse/delta
+            constexpr int PACKETS = BLOCK_M * D / 8;
+            for (int p = tid; p < PACKETS; p += 256) {
+                const int row = p / (D / 8);
+                const int col = (p % (D / 8)) * 8;
+                frag_ab pq = *reinterpret_cast<const frag_ab*>(
+                    qp + (int64_t)(q0 + row) * D + col);
+                frag_ab pdo = *reinterpret_cast<const frag_ab*>(
+                    dop + (int64_t)(q0 + row) * D + col);
+                *reinterpret_cast<frag_ab*>(&q_lds[row * RSTRIDE + col]) = pq;
+                *reinterpret_cast<frag_ab*>(&do_lds[row * RSTRIDE + col]) = pdo;
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    qt_lds[(col + j) * TSTRIDE + row] = (bf16)pq[j];
+                    dot_lds[(col + j) * TSTRIDE + row] = (bf16)pdo[j];
+                }
+            }
+            for (int i = tid; i < BLOCK_M; i += 256) {
+                lse_lds[i] = lse[bh_off + q0 + i];
+                delta_lds[i] = delta[bh_off + q0 + i];
+            }
+        }
+        __syncthreads();
+
+        // build P^T and dS^T for ALL 64 q columns of this block
+#pragma unroll
+        for (int ms = 0; ms < 4; ++ms) {
+            frag_cd st = frag_cd{0.f, 0.f, 0.f, 0.f};
+            frag_cd dpt = frag_cd{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int c = 0; c < DCH; ++c) {
+                frag_ab bQ = *reinterpret_cast<const frag_ab*>(
+                    &q_lds[(ms * 16 + lcol) * RSTRIDE + c * 32 + 8 * lgrp]);
+                st = MFMA_16x16x32(aK[c], bQ, st);          // S^T = K·Q^T
+                frag_ab bDO = *reinterpret_cast<const frag_ab*>(
+                    &do_lds[(ms * 16 + lcol) * RSTRIDE + c * 32 + 8 * lgrp]);
+                dpt = MFMA_16x16x32(aV[c], bDO, dpt);       // dP^T = V·dO^T
+            }
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int n_glob = kv0 + 4 * lgrp + r;
+                const int m_loc = ms * 16 + lcol;
+                const int m_glob = q0 + m_loc;
+                float z = st[r] * scale + slope * (float)(n_glob - m_glob);
+                float p = (n_glob <= m_glob) ? __expf(z - lse_lds[m_loc]) : 0.f;
+                float ds = p * (dpt[r] - delta_lds[m_loc]) * scale;
+                pt_lds[wave][(4 * lgrp + r) * TSTRIDE + m_loc] = (bf16)p;
+                dst_lds[wave][(4 * lgrp + r) * TSTRIDE + m_loc] = (bf16)ds;
+            }
+        }
+        __builtin_amdgcn_s_waitcnt(0);  // own-wave LDS writes visible
+
+        // dV += P^T·dO and dK += dS^T·Q over the 64 q rows (2 K-chunks)
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+            frag_ab aPT = *reinterpret_cast<const frag_ab*>(
+                &pt_lds[wave][lcol * TSTRIDE + kc * 32 + 8 * lgrp]);
+            frag_ab aDST = *reinterpret_cast<const frag_ab*>(
+                &dst_lds[wave][lcol * TSTRIDE + kc * 32 + 8 * lgrp]);
+#pragma unroll
+            for (int ds = 0; ds < DSUB; ++ds) {
+                frag_ab bDOT = *reinterpret_cast<const frag_ab*>(
+                    &dot_lds[(ds * 16 + lcol) * TSTRIDE + kc * 32 + 8 * lgrp]);
+                accDV[ds] = MFMA_16x16x32(aPT, bDOT, accDV[ds]);
+                frag_ab bQT = *reinterpret_cast<const frag_ab*>(
+                    &qt_lds[(ds * 16 + lcol) * TSTRIDE + kc * 32 + 8 * lgrp]);
+                accDK[ds] = MFMA_16x16x32(aDST, bQT, accDK[ds]);
+            }
+        }
+    }
+
+    // epilogue: C row = this wave's kv row (4*lgrp+reg), col = d
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        const int n_glob = kv0 + 4 * lgrp + r;
+        bf16* dkrow = dk + (bh_off + n_glob) * D;
+        bf16* dvrow = dv + (bh_off + n_glob) * D;
+#pragma unroll
+        for (int ds = 0; ds < DSUB; ++ds) {
+            dkrow[ds * 16 + lcol] = (bf16)accDK[ds][r];
+            dvrow[ds * 16 + lcol] = (bf16)accDV[ds][r];
+        }
+    }
+}
+
+// pass B: dQ(m,:) = scale * sum_n dS(m,n) K(n,:)
+template <int D>
+__global__ __launch_bounds__(256)
+void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
+                        const bf16* __restrict__ q, const bf16* __restrict__ k,
+                        const bf16* __restrict__ v,
+                        const float* __restrict__ lse,
+                        const float* __restrict__ delta,
+                        const float* __restrict__ slopes, float scale,
+                        bf16* __restrict__ dq,
+                        int B, int H, int S) {
+    constexpr int DCH = D / 32;
+    constexpr int DSUB = D / 16;
+    constexpr int RSTRIDE = D + PAD;
+    constexpr int TSTRIDE = BLOCK_N + PAD;
+
+    __shared__ bf16 k_lds[BLOCK_N * RSTRIDE];   // row-major K (QK^T B-frags)
+    __shared__ bf16 v_lds[BLOCK_N * RSTRIDE];   // row-major V (dP B-frags)
+    __shared__ bf16 kt_lds[D * TSTRIDE];        // transposed K (dQ B-frags)
+    __shared__ bf16 ds_lds[NWAVES][16 * TSTRIDE];  // dS (m rows, n cols)
+
+    const int qb = blockIdx.x;
+    const int h = blockIdx.y;
+    const int b = blockIdx.z;
+    const int tid = threadIdx.x;
+    const int wave = tid / WAVE_SIZE;
+    const int lane = tid % WAVE_SIZE;
+    const int lgrp = lane >> 4, lcol = lane & 15;
+
+    const int64_t bh_off = ((int64_t)b * H + h) * S;
+    const bf16* kp = k + bh_off * D;
+    const bf16* vp = v + bh_off * D;
+    const float slope = slopes[h];
+
+    const int qrow0 = qb * BLOCK_M + wave * 16;
+
+    // Q and dO rows of this wave in registers; lse/delta per owned row
+    frag_ab aQ[DCH], aDO[DCH];
+    {
+        const bf16* qrow = q + (bh_off + qrow0 + lcol) * D;
+        const bf16* dorow = dout + (bh_off + qrow0 + lcol) * D;
+#pragma unroll
+        for (int c = 0; c < DCH; ++c) {
+            aQ[c] = *reinterpret_cast<const frag_ab*>(qrow + c * 32 + 8 * lgrp);
+            aDO[c] = *reinterpret_cast<const frag_ab*>(dorow + c * 32 + 8 * lgrp);
+        }
+    }
+    float lse_r[4], delta_r[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        const int iq = qrow0 + 4 * lgrp + r;
+        lse_r[r] = lse[bh_off + iq];
+        delta_r[r] = delta[bh_off + iq];
+    }
+
+    frag_cd accDQ[DSUB];
+#pragma unroll
+    for (int s = 0; s < DSUB; ++s) accDQ[s] = frag_cd{0.f, 0.f, 0.f, 0.f};
+
+    for (int nb = 0; nb <= qb; ++nb) {
+        const int kv0 = nb * BLOCK_N;
+        __syncthreads();
+        {   // stage K (row-major + transposed) and V (row-major)
+            constexpr int PACKETS = BLOCK_N * D / 8;
+            for (int p = tid; p < PACKETS; p += 256) {
+                const int row = p / (D / 8);
+                const int col = (p % (D / 8)) * 8;
+                frag_ab pk = *reinterpret_cast<const frag_ab*>(
+                    kp + (int64_t)(kv0 + row) * D + col);
+                frag_ab pv = *reinterpret_cast<const frag_ab*>(
+                    vp + (int64_t)(kv0 + row) * D + col);
+                *reinterpret_cast<frag_ab*>(&k_lds[row * RSTRIDE + col]) = pk;
+                *reinterpret_cast<frag_ab*>(&v_lds[row * RSTRIDE + col]) = pv;
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    kt_lds[(col + j) * TSTRIDE + row] = (bf16)pk[j];
+                }
+            }
+        }
+        __syncthreads();
+
+        // dS for all 64 kv cols of this block
+#pragma unroll
+        for (int ns = 0; ns < 4; ++ns) {
+            frag_cd sacc = frag_cd{0.f, 0.f, 0.f, 0.f};
+            frag_cd dpacc = frag_cd{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int c = 0; c < DCH; ++c) {
+                frag_ab bK = *reinterpret_cast<const frag_ab*>(
+                    &k_lds[(ns * 16 + lcol) * RSTRIDE + c * 32 + 8 * lgrp]);
+                sacc = MFMA_16x16x32(aQ[c], bK, sacc);      // S = Q·K^T
+                frag_ab bV = *reinterpret_cast<const frag_ab*>(
+                    &v_lds[(ns * 16 + lcol) * RSTRIDE + c * 32 + 8 * lgrp]);
+                dpacc = MFMA_16x16x32(aDO[c], bV, dpacc);   // dP = dO·V^T
+            }
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int iq = qrow0 + 4 * lgrp + r;
+                const int jk = kv0 + ns * 16 + lcol;
+                float z = sacc[r] * scale + slope * (float)(jk - iq);
+                float p = (jk <= iq) ? __expf(z - lse_r[r]) : 0.f;
+                float ds = p * (dpacc[r] - delta_r[r]) * scale;
+                // store transposed to [m][n] so the dQ A-frag read is linear
+                ds_lds[wave][(4 * lgrp + r) * TSTRIDE + ns * 16 + lcol] =
+                    (bf16)ds;
+            }
+        }
+        __builtin_amdgcn_s_waitcnt(0);
+
+        // dQ += dS·K over the 64 kv rows (2 K-chunks)
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+            frag_ab aDS = *reinterpret_cast<const frag_ab*>(
+                &ds_lds[wave][lcol * TSTRIDE + kc * 32 + 8 * lgrp]);
+#pragma unroll
+            for (int ds = 0; ds < DSUB; ++ds) {
+                frag_ab bKT = *reinterpret_cast<const frag_ab*>(
+                    &kt_lds[(ds * 16 + lcol) * TSTRIDE + kc * 32 + 8 * lgrp]);
+                accDQ[ds] = MFMA_16x16x32(aDS, bKT, accDQ[ds]);
+            }
+        }
+    }
+
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        const int iq = qrow0 + 4 * lgrp + r;
+        bf16* dqrow = dq + (bh_off + iq) * D;
+#pragma unroll
+        for (int ds = 0; ds < DSUB; ++ds) {
+            dqrow[ds * 16 + lcol] = (bf16)accDQ[ds][r];
+        }
+    }
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
+                                    torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor lse,
+                                    torch::Tensor slopes, double scale) {
+    TORCH_CHECK(dout.is_cuda() && dout.is_contiguous());
+    const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+    auto dq = torch::empty_like(q);
+    auto dk = torch::empty_like(k);
+    auto dv = torch::empty_like(v);
+    auto delta = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
+
+    const int64_t rows = (int64_t)B * H * S;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    dim3 grid_a(S / BLOCK_N, H, B);
+    dim3 grid_b(S / BLOCK_M, H, B);
+
+#define LAUNCH_BWD(DV)                                                        \
+    do {                                                                      \
+        hipLaunchKernelGGL((attn_bwd_delta_kernel<DV>),                       \
+            dim3((rows + 3) / 4), dim3(256), 0, stream,                       \
+            reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
+            reinterpret_cast<const bf16*>(o.data_ptr()),                      \
+            delta.data_ptr<float>(), rows);                                   \
+        hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DV>), grid_a, dim3(256), 0,  \
+            stream,                                                           \
+            reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
+            reinterpret_cast<const bf16*>(q.data_ptr()),                      \
+            reinterpret_cast<const bf16*>(k.data_ptr()),                      \
+            reinterpret_cast<const bf16*>(v.data_ptr()),                      \
+            lse.data_ptr<float>(), delta.data_ptr<float>(),                   \
+            slopes.data_ptr<float>(), (float)scale,                           \
+            reinterpret_cast<bf16*>(dk.data_ptr()),                           \
+            reinterpret_cast<bf16*>(dv.data_ptr()), B, H, S);                 \
+        hipLaunchKernelGGL((attn_bwd_dq_kernel<DV>), grid_b, dim3(256), 0,    \
+            stream,                                                           \
+            reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
+            reinterpret_cast<const bf16*>(q.data_ptr()),                      \
+            reinterpret_cast<const bf16*>(k.data_ptr()),                      \
+            reinterpret_cast<const bf16*>(v.data_ptr()),                      \
+            lse.data_ptr<float>(), delta.data_ptr<float>(),                   \
+            slopes.data_ptr<float>(), (float)scale,                           \
+            reinterpret_cast<bf16*>(dq.data_ptr()), B, H, S);                 \
+    } while (0)
+
+    if (D == 64) {
+        LAUNCH_BWD(64);
+    } else {
+        TORCH_CHECK(D == 128, "attn_bwd: head dim 64 or 128");
+        LAUNCH_BWD(128);
+    }
+#undef LAUNCH_BWD
+    HIP_CHECK_LAUNCH();
+    return {dq, dk, dv};
+}

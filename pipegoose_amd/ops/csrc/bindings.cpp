@@ -17,6 +17,14 @@ torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor targets,
                                 torch::Tensor row_max, torch::Tensor row_sumexp,
                                 torch::Tensor gscale, int64_t vocab_start,
                                 int64_t vocab_end);
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, torch::Tensor slopes,
+                                    double scale);
+std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
+                                    torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor lse,
+                                    torch::Tensor slopes, double scale);
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor Bt);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
@@ -29,4 +37,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused online-softmax CE forward over a vocab shard (gfx950)");
     m.def("cross_entropy_bwd", &cross_entropy_bwd,
           "fused CE backward: (softmax - onehot) * g (gfx950)");
+    m.def("attn_fwd", &attn_fwd,
+          "flash attention fwd, causal + in-kernel ALiBi (gfx950 MFMA)");
+    m.def("attn_bwd", &attn_bwd,
+          "flash attention bwd (two-pass, no atomics) (gfx950 MFMA)");
+    m.def("mfma_probe", &mfma_probe,
+          "16x16x32 bf16 MFMA fragment-layout probe");
 }

@@ -18,6 +18,7 @@ SRC = [
     "pipegoose_amd/ops/csrc/layer_norm.hip",
     "pipegoose_amd/ops/csrc/bias_gelu.hip",
     "pipegoose_amd/ops/csrc/cross_entropy.hip",
+    "pipegoose_amd/ops/csrc/attention.hip",
 ]
 
 setup(
