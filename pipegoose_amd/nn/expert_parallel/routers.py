@@ -71,6 +71,11 @@ class _TopKRouter(nn.Module):
         logits = TF.linear(hidden, self.gate.weight.float(), gate_bias)
         if self.training:
             logits = logits * self.noise_policy.sample_like(logits)
+        elif (logits.is_cuda and not torch.is_grad_enabled()
+              and self.top_k == 1 and self.num_experts <= 64):
+            out = self._forward_fused(logits, orig_dtype)
+            if out is not None:
+                return out
 
         probs = TF.softmax(logits, dim=-1)
         topk_val, topk_idx = probs.topk(self.top_k, dim=-1)
@@ -96,6 +101,32 @@ class _TopKRouter(nn.Module):
             aux_loss=aux_loss,
             z_loss=z_loss,
         )
+
+
+    def _forward_fused(self, logits: torch.Tensor, orig_dtype):
+        """Serving path: one fused HIP pass (softmax+top1+colsum+lse,
+        ops/csrc/router.hip) instead of five elementwise/reduce kernels.
+        Training keeps the autograd torch ops (loss grads)."""
+        from pipegoose_amd.ops import get_extension
+        ext = get_extension()
+        if ext is None or not hasattr(ext, "router_topk"):
+            return None
+        N = logits.size(0)
+        idx, val, colsum, count, lse = ext.router_topk(logits.contiguous(), 1)
+        idx = idx.squeeze(-1).long()
+        weight = torch.zeros_like(logits)
+        weight.scatter_(-1, idx.unsqueeze(-1), val)
+        aux = self.num_experts * torch.sum(
+            (count.float() / N) * (colsum / N))
+        z = lse.square().mean()
+        if self.expert_capacity is not None:
+            capacity = self._expert_capacity(N)
+            mask = torch.zeros_like(logits)
+            mask.scatter_(-1, idx.unsqueeze(-1), 1.0)
+            position = torch.cumsum(mask, dim=0) * mask
+            weight = weight * (position <= capacity).float()
+        return RouterOutput(dispatch_order=idx, weight=weight.to(orig_dtype),
+                            aux_loss=aux, z_loss=z)
 
 
 class Top1Router(_TopKRouter):

@@ -35,21 +35,25 @@ constexpr float NEG_INF = -1e30f;
 // B-fragment (N=16 side): col = lane&15, k = 8*(lane>>4) + i
 // C/D:                    col = lane&15, row = 4*(lane>>4) + reg
 
-template <int D>
+template <int D, int MT>
 __global__ __launch_bounds__(256)
 void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
                      const bf16* __restrict__ v,
                      const float* __restrict__ slopes, float scale,
                      bf16* __restrict__ o, float* __restrict__ lse,
                      int B, int H, int S) {
+    // MT = q row-tiles per wave (16 rows each).  MT=2 doubles the MFMA work
+    // per LDS B-fragment read — the K/V tiles are read once per wave and
+    // used for 32 q rows.
     constexpr int DCH = D / 32;    // K-chunks per QK^T mfma row
     constexpr int DSUB = D / 16;   // output d-subtiles
     constexpr int KSTRIDE = D + PAD;
     constexpr int VSTRIDE = BLOCK_N + PAD;
+    constexpr int ROWS_PER_WG = NWAVES * MT * 16;
 
     __shared__ bf16 k_lds[BLOCK_N * KSTRIDE];
     __shared__ bf16 vt_lds[D * VSTRIDE];
-    __shared__ bf16 p_lds[NWAVES][16 * VSTRIDE];
+    __shared__ bf16 p_lds[NWAVES][MT * 16 * VSTRIDE];
 
     const int qblock = blockIdx.x;
     const int h = blockIdx.y;
@@ -65,31 +69,35 @@ void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* kp = k + (bh_off) * D;
     const bf16* vp = v + (bh_off) * D;
 
-    const int qrow0 = qblock * BLOCK_M + wave * 16;
+    // wave owns rows [qrow0, qrow0 + MT*16)
+    const int qrow0 = qblock * ROWS_PER_WG + wave * (MT * 16);
     const float slope = slopes[h];
 
-    // ---- Q fragments: stay in registers for the whole kernel
-    frag_ab aQ[DCH];
-    {
-        const bf16* qrow = qp + (int64_t)(qrow0 + lcol) * D;
+    frag_ab aQ[MT][DCH];
+#pragma unroll
+    for (int mt = 0; mt < MT; ++mt) {
+        const bf16* qrow = qp + (int64_t)(qrow0 + mt * 16 + lcol) * D;
 #pragma unroll
         for (int c = 0; c < DCH; ++c) {
-            aQ[c] = *reinterpret_cast<const frag_ab*>(qrow + c * 32 + 8 * lgrp);
+            aQ[mt][c] = *reinterpret_cast<const frag_ab*>(qrow + c * 32 + 8 * lgrp);
         }
     }
 
-    frag_cd accO[DSUB];
+    frag_cd accO[MT][DSUB];
+    float m_run[MT][4], l_run[MT][4];
 #pragma unroll
-    for (int s = 0; s < DSUB; ++s) accO[s] = frag_cd{0.f, 0.f, 0.f, 0.f};
-    float m_run[4] = {NEG_INF, NEG_INF, NEG_INF, NEG_INF};
-    float l_run[4] = {0.f, 0.f, 0.f, 0.f};
+    for (int mt = 0; mt < MT; ++mt) {
+#pragma unroll
+        for (int s = 0; s < DSUB; ++s) accO[mt][s] = frag_cd{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int r = 0; r < 4; ++r) { m_run[mt][r] = NEG_INF; l_run[mt][r] = 0.f; }
+    }
 
-    const int n_kv_blocks = qblock + 1;  // causal
+    const int n_kv_blocks = (qblock * ROWS_PER_WG + ROWS_PER_WG) / BLOCK_N;  // causal
     for (int nb = 0; nb < n_kv_blocks; ++nb) {
         const int kvrow0 = nb * BLOCK_N;
         __syncthreads();
-        // ---- cooperative K tile load (row-major, 16B packets)
-        {
+        {   // cooperative K tile load (row-major, 16B packets)
             constexpr int PACKETS = BLOCK_N * D / 8;
             for (int p = tid; p < PACKETS; p += 256) {
                 const int row = p / (D / 8);
@@ -99,116 +107,132 @@ void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
                         kp + (int64_t)(kvrow0 + row) * D + col);
             }
         }
-        // ---- cooperative V tile load, transposed into LDS
-        {
-            constexpr int PACKETS = BLOCK_N * D / 8;
+        {   // V tile load, transposed; two rows at once -> 4 B column stores
+            constexpr int PACKETS = (BLOCK_N / 2) * (D / 8);
             for (int p = tid; p < PACKETS; p += 256) {
-                const int row = p / (D / 8);        // kv row
-                const int col = (p % (D / 8)) * 8;  // d offset
-                frag_ab pkt = *reinterpret_cast<const frag_ab*>(
+                const int row = (p / (D / 8)) * 2;
+                const int col = (p % (D / 8)) * 8;
+                frag_ab p0 = *reinterpret_cast<const frag_ab*>(
                     vp + (int64_t)(kvrow0 + row) * D + col);
+                frag_ab p1 = *reinterpret_cast<const frag_ab*>(
+                    vp + (int64_t)(kvrow0 + row + 1) * D + col);
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
-                    vt_lds[(col + j) * VSTRIDE + row] = (bf16)pkt[j];
+                    union { __bf16 h[2]; uint32_t u; } pair;
+                    pair.h[0] = p0[j];
+                    pair.h[1] = p1[j];
+                    *reinterpret_cast<uint32_t*>(&vt_lds[(col + j) * VSTRIDE + row]) = pair.u;
                 }
             }
         }
         __syncthreads();
 
-        // ---- S = Q K^T for this wave's 16 rows x 64 cols
-        float s_tile[4][4];  // [nsub][reg]
 #pragma unroll
-        for (int ns = 0; ns < 4; ++ns) {
-            frag_cd acc = frag_cd{0.f, 0.f, 0.f, 0.f};
+        for (int mt = 0; mt < MT; ++mt) {
+            // S = Q K^T for this row-tile (16 x 64)
+            float s_tile[4][4];  // [nsub][reg]
 #pragma unroll
-            for (int c = 0; c < DCH; ++c) {
-                frag_ab bK = *reinterpret_cast<const frag_ab*>(
-                    &k_lds[(ns * 16 + lcol) * KSTRIDE + c * 32 + 8 * lgrp]);
-                acc = MFMA_16x16x32(aQ[c], bK, acc);
+            for (int ns = 0; ns < 4; ++ns) {
+                frag_cd acc = frag_cd{0.f, 0.f, 0.f, 0.f};
+                __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                for (int c = 0; c < DCH; ++c) {
+                    frag_ab bK = *reinterpret_cast<const frag_ab*>(
+                        &k_lds[(ns * 16 + lcol) * KSTRIDE + c * 32 + 8 * lgrp]);
+                    acc = MFMA_16x16x32(aQ[mt][c], bK, acc);
+                }
+                __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int iq = qrow0 + mt * 16 + 4 * lgrp + r;
+                    const int jk = kvrow0 + ns * 16 + lcol;
+                    float sv = acc[r] * scale + slope * (float)(jk - iq);
+                    s_tile[ns][r] = (jk <= iq) ? sv : NEG_INF;
+                }
+            }
+
+            // online softmax update
+            float m_new[4];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                float mx = fmaxf(fmaxf(s_tile[0][r], s_tile[1][r]),
+                                 fmaxf(s_tile[2][r], s_tile[3][r]));
+#pragma unroll
+                for (int off = 1; off < 16; off <<= 1) {
+                    mx = fmaxf(mx, __shfl_xor(mx, off, WAVE_SIZE));
+                }
+                m_new[r] = fmaxf(m_run[mt][r], mx);
+            }
+            float rescale[4];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                rescale[r] = __expf(m_run[mt][r] - m_new[r]);
+                m_run[mt][r] = m_new[r];
+                l_run[mt][r] *= rescale[r];
+            }
+#pragma unroll
+            for (int s = 0; s < DSUB; ++s) {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) accO[mt][s][r] *= rescale[r];
+            }
+
+            // P = exp(S - m); stage bf16 for the PV mfma; accumulate l
+#pragma unroll
+            for (int ns = 0; ns < 4; ++ns) {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    float pv = __expf(s_tile[ns][r] - m_run[mt][r]);
+                    s_tile[ns][r] = pv;
+                    p_lds[wave][(mt * 16 + 4 * lgrp + r) * VSTRIDE + ns * 16 + lcol] =
+                        (bf16)pv;
+                }
             }
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                const int iq = qrow0 + 4 * lgrp + r;
-                const int jk = kvrow0 + ns * 16 + lcol;
-                float sv = acc[r] * scale + slope * (float)(jk - iq);
-                s_tile[ns][r] = (jk <= iq) ? sv : NEG_INF;
+                float ls = s_tile[0][r] + s_tile[1][r] + s_tile[2][r] + s_tile[3][r];
+#pragma unroll
+                for (int off = 1; off < 16; off <<= 1) {
+                    ls += __shfl_xor(ls, off, WAVE_SIZE);
+                }
+                l_run[mt][r] += ls;
             }
         }
+        __builtin_amdgcn_s_waitcnt(0);  // own-wave P writes visible
 
-        // ---- online softmax update (row stats live in 4 regs per lane)
-        float m_new[4];
+        // O += P V  (K = 64 in 2 chunks of 32)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            float mx = fmaxf(fmaxf(s_tile[0][r], s_tile[1][r]),
-                             fmaxf(s_tile[2][r], s_tile[3][r]));
+        for (int mt = 0; mt < MT; ++mt) {
 #pragma unroll
-            for (int off = 1; off < 16; off <<= 1) {
-                mx = fmaxf(mx, __shfl_xor(mx, off, WAVE_SIZE));
-            }
-            m_new[r] = fmaxf(m_run[r], mx);
-        }
-        float rescale[4];
+            for (int kc = 0; kc < 2; ++kc) {
+                frag_ab aP = *reinterpret_cast<const frag_ab*>(
+                    &p_lds[wave][(mt * 16 + lcol) * VSTRIDE + kc * 32 + 8 * lgrp]);
+                __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            rescale[r] = __expf(m_run[r] - m_new[r]);
-            m_run[r] = m_new[r];
-            l_run[r] *= rescale[r];
-        }
-#pragma unroll
-        for (int s = 0; s < DSUB; ++s) {
-#pragma unroll
-            for (int r = 0; r < 4; ++r) accO[s][r] *= rescale[r];
-        }
-
-        // ---- P = exp(S - m); accumulate l; stage P (bf16) for the PV mfma
-#pragma unroll
-        for (int ns = 0; ns < 4; ++ns) {
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                float pv = __expf(s_tile[ns][r] - m_run[r]);
-                s_tile[ns][r] = pv;
-                p_lds[wave][(4 * lgrp + r) * VSTRIDE + ns * 16 + lcol] =
-                    (bf16)pv;
-            }
-        }
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            float ls = s_tile[0][r] + s_tile[1][r] + s_tile[2][r] + s_tile[3][r];
-#pragma unroll
-            for (int off = 1; off < 16; off <<= 1) {
-                ls += __shfl_xor(ls, off, WAVE_SIZE);
-            }
-            l_run[r] += ls;
-        }
-        // LDS fence within the workgroup: P writes (own wave) + V/K reuse
-        __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0) & vmcnt(0)
-
-        // ---- O += P V  (K = 64 in 2 chunks of 32)
-#pragma unroll
-        for (int kc = 0; kc < 2; ++kc) {
-            frag_ab aP = *reinterpret_cast<const frag_ab*>(
-                &p_lds[wave][lcol * VSTRIDE + kc * 32 + 8 * lgrp]);
-#pragma unroll
-            for (int ds = 0; ds < DSUB; ++ds) {
-                frag_ab bV = *reinterpret_cast<const frag_ab*>(
-                    &vt_lds[(ds * 16 + lcol) * VSTRIDE + kc * 32 + 8 * lgrp]);
-                accO[ds] = MFMA_16x16x32(aP, bV, accO[ds]);
+                for (int ds = 0; ds < DSUB; ++ds) {
+                    frag_ab bV = *reinterpret_cast<const frag_ab*>(
+                        &vt_lds[(ds * 16 + lcol) * VSTRIDE + kc * 32 + 8 * lgrp]);
+                    accO[mt][ds] = MFMA_16x16x32(aP, bV, accO[mt][ds]);
+                }
+                __builtin_amdgcn_s_setprio(0);
             }
         }
     }
 
-    // ---- epilogue: O /= l, store O and lse
+    // epilogue
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        const int iq = qrow0 + 4 * lgrp + r;
-        const float inv_l = 1.0f / l_run[r];
-        bf16* orow = o + (bh_off + iq) * D;
+    for (int mt = 0; mt < MT; ++mt) {
 #pragma unroll
-        for (int ds = 0; ds < DSUB; ++ds) {
-            orow[ds * 16 + lcol] = (bf16)(accO[ds][r] * inv_l);
-        }
-        if (lcol == 0) {
-            lse[bh_off + iq] = m_run[r] + __logf(l_run[r]);
+        for (int r = 0; r < 4; ++r) {
+            const int iq = qrow0 + mt * 16 + 4 * lgrp + r;
+            const float inv_l = 1.0f / l_run[mt][r];
+            bf16* orow = o + (bh_off + iq) * D;
+#pragma unroll
+            for (int ds = 0; ds < DSUB; ++ds) {
+                orow[ds * 16 + lcol] = (bf16)(accO[mt][ds][r] * inv_l);
+            }
+            if (lcol == 0) {
+                lse[bh_off + iq] = m_run[mt][r] + __logf(l_run[mt][r]);
+            }
         }
     }
 }
@@ -249,25 +273,24 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
     auto o = torch::empty_like(q);
     auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
 
-    dim3 grid(S / BLOCK_M, H, B);
     auto stream = at::cuda::getCurrentCUDAStream();
+    // MT=2 (128 q rows / workgroup) when S allows: 2x the MFMA work per LDS
+    // fragment read
+    const int mt = (S % 128 == 0) ? 2 : 1;
+    dim3 grid(S / (BLOCK_M * mt), H, B);
+#define LAUNCH_FWD(DV, MTV)                                                   \
+    hipLaunchKernelGGL((attn_fwd_kernel<DV, MTV>), grid, dim3(256), 0, stream,\
+        reinterpret_cast<const bf16*>(q.data_ptr()),                          \
+        reinterpret_cast<const bf16*>(k.data_ptr()),                          \
+        reinterpret_cast<const bf16*>(v.data_ptr()),                          \
+        slopes.data_ptr<float>(), (float)scale,                               \
+        reinterpret_cast<bf16*>(o.data_ptr()), lse.data_ptr<float>(), B, H, S)
     if (D == 64) {
-        hipLaunchKernelGGL((attn_fwd_kernel<64>), grid, dim3(256), 0, stream,
-            reinterpret_cast<const bf16*>(q.data_ptr()),
-            reinterpret_cast<const bf16*>(k.data_ptr()),
-            reinterpret_cast<const bf16*>(v.data_ptr()),
-            slopes.data_ptr<float>(), (float)scale,
-            reinterpret_cast<bf16*>(o.data_ptr()), lse.data_ptr<float>(),
-            B, H, S);
+        if (mt == 2) LAUNCH_FWD(64, 2); else LAUNCH_FWD(64, 1);
     } else {
-        hipLaunchKernelGGL((attn_fwd_kernel<128>), grid, dim3(256), 0, stream,
-            reinterpret_cast<const bf16*>(q.data_ptr()),
-            reinterpret_cast<const bf16*>(k.data_ptr()),
-            reinterpret_cast<const bf16*>(v.data_ptr()),
-            slopes.data_ptr<float>(), (float)scale,
-            reinterpret_cast<bf16*>(o.data_ptr()), lse.data_ptr<float>(),
-            B, H, S);
+        if (mt == 2) LAUNCH_FWD(128, 2); else LAUNCH_FWD(128, 1);
     }
+#undef LAUNCH_FWD
     HIP_CHECK_LAUNCH();
     return {o, lse};
 }

@@ -25,6 +25,12 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor o, torch::Tensor lse,
                                     torch::Tensor slopes, double scale);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor Bt);
+std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w,
+                                        double eps);
+std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
+                                        torch::Tensor w, torch::Tensor rstd);
+torch::Tensor rope_apply(torch::Tensor x, double theta_base, bool backward);
+std::vector<torch::Tensor> router_topk(torch::Tensor logits, int64_t k);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
@@ -43,4 +49,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "flash attention bwd (two-pass, no atomics) (gfx950 MFMA)");
     m.def("mfma_probe", &mfma_probe,
           "16x16x32 bf16 MFMA fragment-layout probe");
+    m.def("rms_norm_fwd", &rms_norm_fwd, "fused RMSNorm forward (gfx950)");
+    m.def("rms_norm_bwd", &rms_norm_bwd, "fused RMSNorm backward (gfx950)");
+    m.def("rope_apply", &rope_apply,
+          "rotary embedding with in-kernel cos/sin (gfx950)");
+    m.def("router_topk", &router_topk,
+          "fused Switch router: softmax+topk+colsum+lse in one pass (gfx950)");
 }

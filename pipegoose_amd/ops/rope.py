@@ -1,0 +1,43 @@
+"""Rotary position embedding, half-split (NeoX/LLaMA) convention.
+
+GPU: in-kernel cos/sin from the base frequency (no [S, D] tables in HBM);
+backward is the inverse rotation.  CPU oracle: explicit fp32 rotation.
+"""
+import os
+
+import torch
+
+from pipegoose_amd.ops import get_extension
+
+
+class _Rope(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, theta_base):
+        ext = get_extension(required=True)
+        ctx.theta_base = theta_base
+        return ext.rope_apply(x, theta_base, False)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = get_extension(required=True)
+        return ext.rope_apply(dy.contiguous(), ctx.theta_base, True), None
+
+
+def _rope_ref(x: torch.Tensor, theta_base: float) -> torch.Tensor:
+    B, H, S, D = x.shape
+    half = D // 2
+    d = torch.arange(half, device=x.device, dtype=torch.float32)
+    freqs = theta_base ** (-2.0 * d / D)
+    ang = torch.arange(S, device=x.device, dtype=torch.float32)[:, None] * freqs
+    cos, sin = ang.cos(), ang.sin()          # [S, D/2]
+    x1, x2 = x[..., :half].float(), x[..., half:].float()
+    y1 = x1 * cos - x2 * sin
+    y2 = x2 * cos + x1 * sin
+    return torch.cat([y1, y2], dim=-1).to(x.dtype)
+
+
+def apply_rope(x: torch.Tensor, theta_base: float = 10000.0) -> torch.Tensor:
+    """x: [B, H, S, D] — returns the rotated tensor."""
+    if x.is_cuda and os.environ.get("PIPEGOOSE_DISABLE_EXT") != "1":
+        return _Rope.apply(x.contiguous(), theta_base)
+    return _rope_ref(x, theta_base)

@@ -19,6 +19,9 @@ SRC = [
     "pipegoose_amd/ops/csrc/bias_gelu.hip",
     "pipegoose_amd/ops/csrc/cross_entropy.hip",
     "pipegoose_amd/ops/csrc/attention.hip",
+    "pipegoose_amd/ops/csrc/rms_norm.hip",
+    "pipegoose_amd/ops/csrc/rope.hip",
+    "pipegoose_amd/ops/csrc/router.hip",
 ]
 
 setup(

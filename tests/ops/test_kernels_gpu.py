@@ -303,3 +303,80 @@ def test_alibi_attention_dispatch_uses_kernel():
     out = alibi_attention(q, k, v, slopes, 0.125)
     out.sum().backward()
     assert q.grad is not None and torch.isfinite(q.grad.float()).all()
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_rms_norm_fwd_bwd(dtype):
+    from pipegoose_amd.ops.rms_norm import fused_rms_norm
+    torch.manual_seed(9)
+    H = 1024
+    x = torch.randn(4, 64, H, device="cuda", dtype=dtype, requires_grad=True)
+    w = torch.randn(H, device="cuda", dtype=dtype, requires_grad=True)
+    y = fused_rms_norm(x, w, 1e-6)
+    g = torch.randn_like(y)
+    y.backward(g)
+
+    x2 = x.detach().float().requires_grad_(True)
+    w2 = w.detach().float().requires_grad_(True)
+    rstd = torch.rsqrt(x2.pow(2).mean(-1, keepdim=True) + 1e-6)
+    y2 = x2 * rstd * w2
+    y2.backward(g.float())
+    atol = 1e-4 if dtype == torch.float32 else 5e-2
+    assert torch.allclose(y.float(), y2, atol=atol), (y.float() - y2).abs().max()
+    assert torch.allclose(x.grad.float(), x2.grad, atol=atol * 4), \
+        (x.grad.float() - x2.grad).abs().max()
+    assert torch.allclose(w.grad.float(), w2.grad, rtol=1e-2, atol=atol * 4)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_rope_fwd_bwd(dtype):
+    from pipegoose_amd.ops.rope import apply_rope, _rope_ref
+    torch.manual_seed(10)
+    x = torch.randn(2, 4, 96, 64, device="cuda", dtype=dtype, requires_grad=True)
+    y = apply_rope(x, 10000.0)
+    ref = _rope_ref(x.detach(), 10000.0)
+    atol = 1e-4 if dtype == torch.float32 else 3e-2
+    assert torch.allclose(y.float(), ref.float(), atol=atol), \
+        (y.float() - ref.float()).abs().max()
+    # rotation is orthogonal: grad of sum(y*g) wrt x is inverse-rotated g
+    g = torch.randn_like(y)
+    y.backward(g)
+    x2 = x.detach().float().requires_grad_(True)
+    _rope_ref(x2, 10000.0).backward(g.float())
+    assert torch.allclose(x.grad.float(), x2.grad, atol=atol * 2)
+
+
+def test_router_topk_kernel():
+    ext = _ext()
+    torch.manual_seed(11)
+    N, E = 4096, 8
+    logits = torch.randn(N, E, device="cuda")
+    idx, val, colsum, count, lse = ext.router_topk(logits, 1)
+    probs = torch.softmax(logits, dim=-1)
+    assert torch.equal(idx.squeeze(-1).long(), probs.argmax(-1))
+    assert torch.allclose(val.squeeze(-1), probs.max(-1).values, atol=1e-5)
+    assert torch.allclose(colsum, probs.sum(0), rtol=1e-3)
+    assert torch.allclose(lse, torch.logsumexp(logits, -1), atol=1e-4)
+    assert count.sum().item() == N
+
+    idx2, val2, _, _, _ = ext.router_topk(logits, 2)
+    ref_val, ref_idx = probs.topk(2, dim=-1)
+    assert torch.equal(idx2.long(), ref_idx)
+    assert torch.allclose(val2, ref_val, atol=1e-5)
+
+
+def test_router_fused_matches_eager():
+    """_TopKRouter eval path (fused kernel) vs the training torch path."""
+    from pipegoose_amd.nn.expert_parallel import SwitchNoisePolicy, Top1Router
+    torch.manual_seed(12)
+    router = Top1Router(SwitchNoisePolicy(), 8, 64).cuda()
+    router.eval()
+    x = torch.randn(2, 32, 64, device="cuda")
+    with torch.no_grad():
+        out_fused = router(x)
+    with torch.enable_grad():
+        out_eager = router(x)
+    assert torch.equal(out_fused.dispatch_order, out_eager.dispatch_order)
+    assert torch.allclose(out_fused.weight, out_eager.weight, atol=1e-5)
+    assert torch.allclose(out_fused.aux_loss, out_eager.aux_loss, rtol=1e-3)
+    assert torch.allclose(out_fused.z_loss, out_eager.z_loss, rtol=1e-3)
