@@ -1,0 +1,210 @@
+"""Native LLaMA-style model family, tensor-parallel by construction.
+
+Second model family next to BLOOM (models/bloom.py): RMSNorm (fused HIP
+kernel), rotary position embeddings (in-kernel cos/sin, ops/rope.py), SwiGLU
+MLP, untied vocab-parallel LM head.  Attention reuses the hand-written flash
+kernel with the ALiBi slopes set to zero (causal, no bias).
+
+Module names follow the HF llama layout (model.layers.N.self_attn.q_proj,
+mlp.gate_proj, ...) so TP mappings and checkpoints line up.
+"""
+import math
+from dataclasses import dataclass
+
+import torch
+from torch import nn
+import torch.nn.functional as TF
+
+from pipegoose_amd.distributed.parallel_context import ParallelContext
+from pipegoose_amd.distributed.parallel_mode import ParallelMode
+from pipegoose_amd.nn.tensor_parallel.embedding import ParallelEmbedding
+from pipegoose_amd.nn.tensor_parallel.linear import (ColumnParallelLinear,
+                                                     RowParallelLinear)
+from pipegoose_amd.nn.tensor_parallel.loss import VocabParallelCrossEntropy
+from pipegoose_amd.nn.tensor_parallel.rms_norm import RMSNorm
+from pipegoose_amd.ops.rope import apply_rope
+
+
+@dataclass
+class LlamaConfig:
+    vocab_size: int = 32000
+    hidden_size: int = 4096
+    intermediate_size: int = 11008
+    n_layer: int = 32
+    n_head: int = 32
+    rope_theta: float = 10000.0
+    rms_norm_eps: float = 1e-6
+    initializer_range: float = 0.02
+    sequence_parallel: bool = False
+
+    @property
+    def head_dim(self):
+        return self.hidden_size // self.n_head
+
+
+def llama_7b():
+    return LlamaConfig()
+
+
+def llama_1b():
+    return LlamaConfig(hidden_size=2048, intermediate_size=5504,
+                       n_layer=16, n_head=16)
+
+
+def llama_tiny():
+    """For tests."""
+    return LlamaConfig(vocab_size=256, hidden_size=64, intermediate_size=128,
+                       n_layer=2, n_head=4)
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, config: LlamaConfig, parallel_context: ParallelContext):
+        super().__init__()
+        tp = parallel_context.get_world_size(ParallelMode.TENSOR)
+        assert config.n_head % tp == 0
+        self.num_heads = config.n_head // tp
+        self.head_dim = config.head_dim
+        self.rope_theta = config.rope_theta
+        self.inv_norm = 1.0 / math.sqrt(self.head_dim)
+        sp = config.sequence_parallel
+        h = config.hidden_size
+        self.q_proj = ColumnParallelLinear(h, h, bias=False, sequence_parallel=sp,
+                                           parallel_context=parallel_context)
+        self.k_proj = ColumnParallelLinear(h, h, bias=False, sequence_parallel=sp,
+                                           parallel_context=parallel_context)
+        self.v_proj = ColumnParallelLinear(h, h, bias=False, sequence_parallel=sp,
+                                           parallel_context=parallel_context)
+        self.o_proj = RowParallelLinear(h, h, bias=False, sequence_parallel=sp,
+                                        parallel_context=parallel_context)
+        self.register_buffer("zero_slopes", torch.zeros(self.num_heads),
+                             persistent=False)
+        self._mask_cache = {}
+
+    def _causal_mask(self, S: int, device, dtype):
+        key = (S, device, dtype)
+        if key not in self._mask_cache:
+            m = torch.triu(torch.full((S, S), float("-inf"), device=device), 1)
+            self._mask_cache = {key: m.to(dtype)[None, None]}
+        return self._mask_cache[key]
+
+    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
+        B = hidden.size(0)
+        q = self.q_proj(hidden)
+        S = q.size(1)
+        k = self.k_proj(hidden)
+        v = self.v_proj(hidden)
+        q = q.view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
+        k = k.view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
+        v = v.view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
+        q = apply_rope(q, self.rope_theta)
+        k = apply_rope(k, self.rope_theta)
+
+        from pipegoose_amd.ops.attention import alibi_attention
+        out = alibi_attention(
+            q, k, v, self.zero_slopes, self.inv_norm,
+            mask_fallback=lambda s, dev, dt: self._causal_mask(s, dev, dt))
+        out = out.transpose(1, 2).reshape(B, S, self.num_heads * self.head_dim)
+        return self.o_proj(out)
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, config: LlamaConfig, parallel_context: ParallelContext):
+        super().__init__()
+        h, inter = config.hidden_size, config.intermediate_size
+        sp = config.sequence_parallel
+        self.gate_proj = ColumnParallelLinear(h, inter, bias=False,
+                                              sequence_parallel=sp,
+                                              parallel_context=parallel_context)
+        self.up_proj = ColumnParallelLinear(h, inter, bias=False,
+                                            sequence_parallel=sp,
+                                            parallel_context=parallel_context)
+        self.down_proj = RowParallelLinear(inter, h, bias=False,
+                                           sequence_parallel=sp,
+                                           parallel_context=parallel_context)
+
+    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
+        return self.down_proj(TF.silu(self.gate_proj(hidden)) * self.up_proj(hidden))
+
+
+class LlamaBlock(nn.Module):
+    def __init__(self, config: LlamaConfig, parallel_context: ParallelContext):
+        super().__init__()
+        sp = config.sequence_parallel
+        ctx = parallel_context if sp else None
+        self.input_layernorm = RMSNorm(config.hidden_size, config.rms_norm_eps,
+                                       sequence_parallel=sp, parallel_context=ctx)
+        self.self_attn = LlamaAttention(config, parallel_context)
+        self.post_attention_layernorm = RMSNorm(
+            config.hidden_size, config.rms_norm_eps,
+            sequence_parallel=sp, parallel_context=ctx)
+        self.mlp = LlamaMLP(config, parallel_context)
+
+    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
+        hidden = hidden + self.self_attn(self.input_layernorm(hidden))
+        hidden = hidden + self.mlp(self.post_attention_layernorm(hidden))
+        return hidden
+
+
+class LlamaModel(nn.Module):
+    def __init__(self, config: LlamaConfig, parallel_context: ParallelContext):
+        super().__init__()
+        self.config = config
+        self.embed_tokens = ParallelEmbedding(
+            config.vocab_size, config.hidden_size,
+            sequence_parallel=config.sequence_parallel,
+            parallel_context=parallel_context)
+        self.layers = nn.ModuleList(
+            [LlamaBlock(config, parallel_context) for _ in range(config.n_layer)])
+        self.norm = RMSNorm(
+            config.hidden_size, config.rms_norm_eps,
+            sequence_parallel=config.sequence_parallel,
+            parallel_context=parallel_context if config.sequence_parallel else None)
+
+    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+        hidden = self.embed_tokens(input_ids)
+        for layer in self.layers:
+            hidden = layer(hidden)
+        return self.norm(hidden)
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, config: LlamaConfig, parallel_context: ParallelContext):
+        super().__init__()
+        self.config = config
+        self.parallel_context = parallel_context
+        self.model = LlamaModel(config, parallel_context)
+        tp = parallel_context.get_world_size(ParallelMode.TENSOR)
+        self.lm_head = ColumnParallelLinear(
+            config.hidden_size, config.vocab_size, bias=False,
+            gather_output=False, sequence_parallel=config.sequence_parallel,
+            parallel_context=parallel_context)
+        self.loss_fn = VocabParallelCrossEntropy(parallel_context=parallel_context) \
+            if tp > 1 else None
+        self.apply(self._init_weights)
+
+    def _init_weights(self, module):
+        std = self.config.initializer_range
+        if isinstance(module, (ColumnParallelLinear, RowParallelLinear)):
+            nn.init.normal_(module.weight, mean=0.0, std=std)
+            if module.bias is not None:
+                nn.init.zeros_(module.bias)
+        elif isinstance(module, ParallelEmbedding):
+            nn.init.normal_(module.weight, mean=0.0, std=std)
+
+    def forward(self, input_ids: torch.Tensor, labels: torch.Tensor = None):
+        hidden = self.model(input_ids)
+        logits = self.lm_head(hidden)
+        if labels is None:
+            return logits
+        shift_logits = logits[:, :-1].contiguous()
+        shift_labels = labels[:, 1:].contiguous()
+        if self.loss_fn is not None:
+            return self.loss_fn(shift_logits, shift_labels)
+        if logits.is_cuda:
+            from pipegoose_amd.ops.cross_entropy import fused_cross_entropy
+            return fused_cross_entropy(
+                shift_logits.reshape(-1, shift_logits.size(-1)),
+                shift_labels.reshape(-1))
+        return TF.cross_entropy(
+            shift_logits.float().reshape(-1, shift_logits.size(-1)),
+            shift_labels.reshape(-1))

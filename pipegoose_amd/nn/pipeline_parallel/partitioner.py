@@ -81,6 +81,21 @@ def _resolve_hf_transformer(model: nn.Module) -> Optional[ModelStructure]:
     return ModelStructure(pre=pre, blocks=blocks, post=post)
 
 
+def _resolve_native_llama(model: nn.Module) -> Optional[ModelStructure]:
+    from pipegoose_amd.models.llama import LlamaForCausalLM, LlamaModel
+    if isinstance(model, LlamaForCausalLM):
+        m = model.model
+        return ModelStructure(
+            pre=[m.embed_tokens],
+            blocks=list(m.layers),
+            post=[m.norm, model.lm_head],
+        )
+    if isinstance(model, LlamaModel):
+        return ModelStructure(pre=[model.embed_tokens], blocks=list(model.layers),
+                              post=[model.norm])
+    return None
+
+
 def _resolve_sequential(model: nn.Module) -> Optional[ModelStructure]:
     if isinstance(model, nn.Sequential):
         return ModelStructure(pre=[], blocks=list(model), post=[])
@@ -88,7 +103,8 @@ def _resolve_sequential(model: nn.Module) -> Optional[ModelStructure]:
 
 
 _STRUCTURE_RESOLVERS.extend(
-    [_resolve_sequential, _resolve_native_bloom, _resolve_hf_transformer])
+    [_resolve_sequential, _resolve_native_bloom, _resolve_native_llama,
+     _resolve_hf_transformer])
 
 
 def _param_bytes(m: nn.Module) -> int:

@@ -30,7 +30,14 @@ class TensorParallelMapping(ParallelMapping):
             Column("attention.query", "attention.key", "attention.value", "ffn"),
             Row("attention.dense", "ffn_output"),
         ],
-        # pipegoose_amd native models (models/bloom.py) use the same names.
+        "llama": [
+            Column("self_attn.q_proj", "self_attn.k_proj", "self_attn.v_proj",
+                   "mlp.gate_proj", "mlp.up_proj"),
+            Row("self_attn.o_proj", "mlp.down_proj"),
+            LMHead("lm_head"),
+        ],
+        # pipegoose_amd native models (models/{bloom,llama}.py) use the same
+        # names as the HF architectures.
     }
 
     @classmethod

@@ -1,0 +1,105 @@
+"""Native LLaMA family: TP oracle parity, SP parity, PP partitioning.
+
+Oracle pattern (reference tests/nn/tensor_parallel/test_tensor_parallel.py):
+run the tp=1 model in one process, assert the tp=2 sharded run matches.
+"""
+import torch
+
+from pipegoose_amd.models.llama import (LlamaForCausalLM, llama_tiny)
+from pipegoose_amd.testing.utils import init_parallel_context, spawn
+
+
+def _tiny(sp=False):
+    cfg = llama_tiny()
+    cfg.sequence_parallel = sp
+    return cfg
+
+
+def test_llama_single_rank_forward_backward():
+    spawn(_run_llama_smoke, world_size=1)
+
+
+def _run_llama_smoke(rank, world_size, port):
+    ctx = init_parallel_context(rank, world_size, port)
+    torch.manual_seed(21)
+    model = LlamaForCausalLM(_tiny(), ctx)
+    ids = torch.randint(0, 256, (2, 12))
+    logits = model(ids)
+    assert logits.shape == (2, 12, 256)
+    loss = model(ids, labels=ids)
+    loss.backward()
+    assert torch.isfinite(loss)
+    assert model.model.embed_tokens.weight.grad is not None
+    ctx.destroy()
+
+
+def _run_tp2_oracle(rank, world_size, port):
+    import torch.nn.functional as TF
+    ctx = init_parallel_context(rank, world_size, port, tensor_parallel_size=2)
+    # oracle: same init with a WORLD-1-equivalent full model is impossible
+    # in-process under tp=2; instead verify sharded logits gather to the
+    # reference computed manually from gathered weights.
+    torch.manual_seed(23)
+    model = LlamaForCausalLM(_tiny(), ctx)
+    torch.manual_seed(24)
+    ids = torch.randint(0, 256, (2, 8))
+    logits_sharded = model(ids)  # [B, S, V/2]
+    assert logits_sharded.shape == (2, 8, 128)
+    loss = model(ids, labels=ids)
+    loss.backward()
+    assert torch.isfinite(loss)
+    # vocab-parallel CE must equal CE over gathered logits
+    import torch.distributed as dist
+    from pipegoose_amd.distributed.parallel_mode import ParallelMode
+    full = [torch.empty_like(logits_sharded) for _ in range(2)]
+    dist.all_gather(full, logits_sharded.contiguous(),
+                    group=ctx.get_group(ParallelMode.TENSOR))
+    full = torch.cat(full, dim=-1)
+    ref = TF.cross_entropy(
+        full[:, :-1].reshape(-1, 256).float(), ids[:, 1:].reshape(-1))
+    assert torch.allclose(loss, ref, atol=1e-5), (loss - ref).abs()
+    ctx.destroy()
+
+
+def test_llama_tp2_vocab_parallel_loss():
+    spawn(_run_tp2_oracle, world_size=2)
+
+
+def _run_sp_parity(rank, world_size, port):
+    ctx = init_parallel_context(rank, world_size, port, tensor_parallel_size=2)
+    torch.manual_seed(25)
+    ref = LlamaForCausalLM(_tiny(sp=False), ctx)
+    torch.manual_seed(25)
+    sp = LlamaForCausalLM(_tiny(sp=True), ctx)
+    torch.manual_seed(26)
+    ids = torch.randint(0, 256, (2, 8))
+    l_ref = ref(ids, labels=ids)
+    l_sp = sp(ids, labels=ids)
+    assert torch.allclose(l_ref, l_sp, atol=1e-5)
+    l_ref.backward()
+    l_sp.backward()
+    for (n, p1), p2 in zip(ref.named_parameters(), sp.parameters()):
+        if p1.grad is None:
+            continue
+        assert torch.allclose(p1.grad, p2.grad, atol=1e-4), n
+    ctx.destroy()
+
+
+def test_llama_sp_matches_tp_tp2():
+    spawn(_run_sp_parity, world_size=2)
+
+
+def _run_pp_partition(rank, world_size, port):
+    from pipegoose_amd.nn.pipeline_parallel.partitioner import UniformPartitioner
+    ctx = init_parallel_context(rank, world_size, port)
+    torch.manual_seed(27)
+    model = LlamaForCausalLM(_tiny(), ctx)
+    stages = UniformPartitioner(model, ctx).split(n_partitions=2)
+    assert len(stages) == 2
+    n_params = sum(p.numel() for s in stages for p in s.parameters())
+    assert n_params == sum(p.numel() for p in model.parameters())
+    ctx.destroy()
+
+
+def test_llama_pp_partition():
+    spawn(_run_pp_partition, world_size=1)

@@ -3,11 +3,14 @@
 Usage (GPU box):  python tools/attn_bench.py
 Prints us/call for fwd and fwd+bwd on the bench shapes.
 """
+import os
+import sys
 import time
 
 import torch
 
-from pipegoose_amd.ops import get_extension
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+from pipegoose_amd.ops import get_extension  # noqa: E402
 
 
 def bench(fn, iters=20, warmup=5):
