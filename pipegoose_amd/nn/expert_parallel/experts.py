@@ -73,7 +73,9 @@ class Experts(nn.Module):
             token_mask = dispatch_order.reshape(-1) == global_idx
             if token_mask.any():
                 selected = flat[token_mask]
-                expert_out = expert(selected, *args[1:], **kwargs)
+                # experts receive ONLY their tokens; block-level extras such
+                # as HF Bloom's residual are handled by ExpertLayer
+                expert_out = expert(selected)
                 if isinstance(expert_out, tuple):
                     expert_out = expert_out[0]
                 outputs[token_mask] = expert_out.to(outputs.dtype)

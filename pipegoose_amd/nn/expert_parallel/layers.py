@@ -29,5 +29,12 @@ class ExpertLayer(nn.Module):
         expert_context = ExpertContext.get_instance()
         expert_context.push_aux_loss(router_output.aux_loss)
         expert_context.push_z_loss(router_output.z_loss)
-        outputs = self._experts(inputs, router_output.dispatch_order, *args, **kwargs)
+        outputs = self._experts(inputs, router_output.dispatch_order)
+        # HF Bloom's MLP signature is (hidden, residual) with the residual
+        # added inside the block; when this layer replaces such an MLP the
+        # residual arrives as the 2nd positional arg — add it here.
+        import torch as _torch
+        if len(args) >= 2 and _torch.is_tensor(args[1]) \
+                and args[1].shape == outputs.shape:
+            outputs = outputs + args[1]
         return outputs

@@ -42,6 +42,6 @@ class ParallelEmbedding(nn.Module):
         masked_input = (input - self.vocab_start_idx).masked_fill(mask, 0)
         output = TF.embedding(masked_input, self.weight)
         output = output.masked_fill(mask.unsqueeze(-1), 0.0)
-        if self.sequence_parallel:
+        if getattr(self, "sequence_parallel", False):
             return reduce_scatter_sequence(output, self.parallel_context, dim=1)
         return reduce_to_tensor_group(output, self.parallel_context)

@@ -30,7 +30,7 @@ class LayerNorm(nn.Module):
 
     def _params(self):
         weight, bias = self.weight, self.bias
-        if self.sequence_parallel:
+        if getattr(self, "sequence_parallel", False):
             from pipegoose_amd.nn.tensor_parallel._functional import (
                 broadcast_to_tensor_group)
             weight = broadcast_to_tensor_group(weight, self.parallel_context)

@@ -54,7 +54,7 @@ class ColumnParallelLinear(nn.Module):
             self.register_parameter("bias", None)
 
     def forward(self, input: torch.Tensor) -> torch.Tensor:
-        if self.sequence_parallel:
+        if getattr(self, "sequence_parallel", False):
             input = all_gather_sequence(input, self.parallel_context, dim=1)
         else:
             input = broadcast_to_tensor_group(input, self.parallel_context)
@@ -98,7 +98,7 @@ class RowParallelLinear(nn.Module):
             input = scatter_to_tensor_group(input, dim=-1,
                                             parallel_context=self.parallel_context)
         output = TF.linear(input, self.weight)
-        if self.sequence_parallel:
+        if getattr(self, "sequence_parallel", False):
             output = reduce_scatter_sequence(output, self.parallel_context, dim=1)
             if self.bias is not None:
                 # replicated bias sees only the S-shard: restore the full

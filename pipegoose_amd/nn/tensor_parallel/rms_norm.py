@@ -21,7 +21,7 @@ class RMSNorm(nn.Module):
 
     def forward(self, input: torch.Tensor) -> torch.Tensor:
         weight = self.weight
-        if self.sequence_parallel:
+        if getattr(self, "sequence_parallel", False):
             from pipegoose_amd.nn.tensor_parallel._functional import (
                 broadcast_to_tensor_group)
             weight = broadcast_to_tensor_group(weight, self.parallel_context)

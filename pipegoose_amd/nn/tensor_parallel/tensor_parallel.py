@@ -5,6 +5,7 @@ Walks leaf modules (skipping ExpertLayer subtrees, which belong to
 ExpertParallel), matches each against the parallelizer registry, mutates the
 model in place.
 """
+import torch
 from torch import nn
 
 from pipegoose_amd.distributed.parallel_context import ParallelContext
@@ -34,8 +35,31 @@ class TensorParallel(Parallel):
                 parallelizer = self._find_parallelizer(name, leaf)
                 if parallelizer is not None:
                     parallelizer(name, leaf, module, self.parallel_context).parallelize()
+            self._adapt_attention_modules(module)
         self._save_metadata(module, self.parallel_context)
         return module
+
+    def _adapt_attention_modules(self, model: nn.Module):
+        """HF attention modules keep full head counts after the weight split
+        (the reference never fixed this — its HF-TP path crashed on the
+        fused-qkv reshape).  Divide per-module head attributes by tp when the
+        module owns a column-split child, and apply per-architecture quirks
+        (BloomAttention: the model-level alibi tensor arrives with ALL heads;
+        slice this rank's head block)."""
+        from pipegoose_amd.nn.tensor_parallel.linear import ColumnParallelLinear
+        tp = self.parallel_context.get_world_size(ParallelMode.TENSOR)
+        rank = self.parallel_context.get_local_rank(ParallelMode.TENSOR)
+        for mod in model.modules():
+            has_cpl = any(isinstance(c, ColumnParallelLinear)
+                          for c in mod.children())
+            if not has_cpl:
+                continue
+            for attr in ("num_heads", "num_attention_heads", "num_key_value_heads"):
+                val = getattr(mod, attr, None)
+                if isinstance(val, int) and val % tp == 0 and val >= tp:
+                    setattr(mod, attr, val // tp)
+            if type(mod).__name__ == "BloomAttention":
+                _patch_bloom_attention_alibi(mod, rank, tp)
 
     @staticmethod
     def _get_leaf_modules(model: nn.Module):
@@ -57,3 +81,32 @@ class TensorParallel(Parallel):
             if p.is_parallelizable(module_name, module):
                 return p
         return None
+
+
+def _patch_bloom_attention_alibi(attn, tp_rank: int, tp_size: int):
+    """HF BloomModel builds one alibi tensor [B * n_head_full, 1, S] and hands
+    it to every (now head-sharded) attention; wrap forward to slice out this
+    rank's contiguous head block."""
+    orig_forward = attn.forward
+    local_heads = attn.num_heads  # already divided
+
+    def forward(*args, **kwargs):
+        def slice_alibi(alibi, batch_hint):
+            full = alibi.size(0) // max(batch_hint, 1)
+            if full == local_heads:
+                return alibi
+            B = alibi.size(0) // (local_heads * tp_size)
+            a = alibi.view(B, local_heads * tp_size, *alibi.shape[1:])
+            a = a[:, tp_rank * local_heads:(tp_rank + 1) * local_heads]
+            return a.reshape(B * local_heads, *alibi.shape[1:])
+
+        hidden = kwargs.get("hidden_states", args[0] if args else None)
+        bsz = hidden.size(0) if hidden is not None else 1
+        if "alibi" in kwargs and torch.is_tensor(kwargs["alibi"]):
+            kwargs["alibi"] = slice_alibi(kwargs["alibi"], bsz)
+        elif len(args) >= 3 and torch.is_tensor(args[2]):
+            args = list(args)
+            args[2] = slice_alibi(args[2], bsz)
+        return orig_forward(*args, **kwargs)
+
+    attn.forward = forward

@@ -1,0 +1,101 @@
+"""Drop-in parallelize() on real HuggingFace models — the reference's
+headline use case (README.md:21-69): TensorParallel / DataParallel /
+ExpertParallel surgery on a random-init HF BloomForCausalLM, verified with
+the unparallelized model as oracle (reference
+tests/nn/tensor_parallel/test_tensor_parallel.py idiom).
+"""
+import pytest
+import torch
+
+transformers = pytest.importorskip("transformers")
+
+from pipegoose_amd.nn import DataParallel, ExpertParallel, TensorParallel
+from pipegoose_amd.testing.utils import init_parallel_context, spawn
+
+
+def _hf_bloom():
+    from transformers import BloomConfig, BloomForCausalLM
+    cfg = BloomConfig(vocab_size=512, hidden_size=64, n_layer=2, n_head=4)
+    torch.manual_seed(30)
+    return BloomForCausalLM(cfg)
+
+
+def _run_tp2_hf_bloom(rank, world_size, port):
+    ctx = init_parallel_context(rank, world_size, port, tensor_parallel_size=2)
+    model = _hf_bloom()
+    ref = _hf_bloom()  # same seed -> identical weights
+    torch.manual_seed(31)
+    ids = torch.randint(0, 512, (2, 10))
+
+    with torch.no_grad():
+        ref_logits = ref(ids).logits
+
+    model = TensorParallel(model, ctx).parallelize()
+    with torch.no_grad():
+        out = model(ids).logits
+    # lm_head is column-split without gather: sharded vocab
+    if out.size(-1) == ref_logits.size(-1) // 2:
+        import torch.distributed as dist
+        from pipegoose_amd.distributed.parallel_mode import ParallelMode
+        shards = [torch.empty_like(out) for _ in range(2)]
+        dist.all_gather(shards, out.contiguous(),
+                        group=ctx.get_group(ParallelMode.TENSOR))
+        out = torch.cat(shards, dim=-1)
+    assert torch.allclose(out, ref_logits, atol=1e-4), \
+        (out - ref_logits).abs().max()
+    ctx.destroy()
+
+
+def test_tensor_parallel_hf_bloom_tp2():
+    spawn(_run_tp2_hf_bloom, world_size=2)
+
+
+def _run_dp2_hf_bloom(rank, world_size, port):
+    import torch.distributed as dist
+    ctx = init_parallel_context(rank, world_size, port, data_parallel_size=2)
+    model = _hf_bloom()
+    model = DataParallel(model, ctx).parallelize()
+    torch.manual_seed(40 + rank)  # different batch per DP rank
+    ids = torch.randint(0, 512, (2, 10))
+    loss = model(ids, labels=ids).loss
+    loss.backward()
+    # after the hook sync, grads identical across DP ranks
+    g = model.transformer.word_embeddings.weight.grad.flatten()[:64].clone()
+    peers = [torch.empty_like(g) for _ in range(2)]
+    dist.all_gather(peers, g)
+    assert torch.allclose(peers[0], peers[1], atol=1e-6)
+    ctx.destroy()
+
+
+def test_data_parallel_hf_bloom_dp2():
+    spawn(_run_dp2_hf_bloom, world_size=2)
+
+
+def _run_ep_hf_bloom(rank, world_size, port):
+    from pipegoose_amd.nn.expert_parallel import (ExpertLoss, SwitchNoisePolicy,
+                                                  Top1Router)
+    from pipegoose_amd.nn.expert_parallel.layers import ExpertLayer
+    ctx = init_parallel_context(rank, world_size, port)
+    model = _hf_bloom()
+    n_layers = model.config.n_layer
+    # experts are single-input modules; HF BloomMLP's (hidden, residual)
+    # signature is handled by ExpertLayer's residual add
+    expert = torch.nn.Sequential(
+        torch.nn.Linear(64, 256), torch.nn.GELU(), torch.nn.Linear(256, 64))
+    model = ExpertParallel(
+        model, 4, expert=expert,
+        router=Top1Router(SwitchNoisePolicy(), 4, 64),
+        parallel_context=ctx).parallelize()
+    assert sum(isinstance(m, ExpertLayer) for m in model.modules()) == n_layers
+    ids = torch.randint(0, 512, (2, 10))
+    loss_fn = ExpertLoss(torch.nn.functional.cross_entropy)
+    logits = model(ids).logits
+    loss = loss_fn(logits[:, :-1].reshape(-1, 512).float(),
+                   ids[:, 1:].reshape(-1))
+    loss.backward()
+    assert torch.isfinite(loss)
+    ctx.destroy()
+
+
+def test_expert_parallel_hf_bloom():
+    spawn(_run_ep_hf_bloom, world_size=1)
