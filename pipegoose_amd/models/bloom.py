@@ -272,3 +272,13 @@ class BloomForCausalLM(nn.Module):
                 shift_logits.float().reshape(-1, shift_logits.size(-1)),
                 shift_labels.reshape(-1))
         return loss
+
+    @torch.no_grad()
+    def generate(self, input_ids, max_new_tokens: int = 20,
+                 temperature: float = 0.0, top_k: int = 0,
+                 eos_token_id=None):
+        from pipegoose_amd.models.generation import generate as _generate
+        return _generate(self, input_ids, max_new_tokens=max_new_tokens,
+                         temperature=temperature, top_k=top_k,
+                         parallel_context=self.parallel_context,
+                         eos_token_id=eos_token_id)

@@ -60,6 +60,10 @@ class Experts(nn.Module):
                         "pass a dense expert module to ExpertParallel(expert=...)")
         for p in self.experts.parameters():
             setattr(p, "is_expert", True)
+        # grouped-GEMM fast path: identical Sequential(Linear, act, Linear)
+        # experts run as one batched MFMA GEMM over the expert-sorted tokens
+        from pipegoose_amd.nn.expert_parallel.grouped import match_grouped_mlp
+        self._grouped = match_grouped_mlp(self.experts)
 
     def forward(self, inputs: torch.Tensor, dispatch_order: torch.Tensor, *args, **kwargs):
         # inputs: [B, S, H]; dispatch_order: [B*S] global expert index (top-1)
@@ -107,16 +111,22 @@ class Experts(nn.Module):
         route_chunk = dispatch_order.reshape(-1).chunk(ep, dim=0)[rank]
         recv, local_idx, state = self._dispatcher.dispatch(chunk, route_chunk)
         counts = torch.bincount(local_idx, minlength=self.num_local_experts).tolist()
-        outs = []
-        start = 0
-        for i, expert in enumerate(self.experts):
-            seg = recv[start:start + counts[i]]
-            out = expert(seg)
-            if isinstance(out, tuple):
-                out = out[0]
-            outs.append(out.to(recv.dtype))
-            start += counts[i]
-        expert_out = torch.cat(outs, dim=0) if outs else recv
+        if self._grouped is not None and recv.size(0) > 0:
+            from pipegoose_amd.nn.expert_parallel.grouped import (
+                grouped_mlp_forward)
+            w1, b1, w2, b2, act = self._grouped
+            expert_out = grouped_mlp_forward(recv, counts, w1, b1, w2, b2, act)
+        else:
+            outs = []
+            start = 0
+            for i, expert in enumerate(self.experts):
+                seg = recv[start:start + counts[i]]
+                out = expert(seg)
+                if isinstance(out, tuple):
+                    out = out[0]
+                outs.append(out.to(recv.dtype))
+                start += counts[i]
+            expert_out = torch.cat(outs, dim=0) if outs else recv
         combined = self._dispatcher.combine(expert_out, state)
         return _Gather.apply(combined, 0, self.parallel_context)
 

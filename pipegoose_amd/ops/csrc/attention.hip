@@ -276,7 +276,11 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
     auto stream = at::cuda::getCurrentCUDAStream();
     // MT=2 (128 q rows / workgroup) when S allows: 2x the MFMA work per LDS
     // fragment read
-    const int mt = (S % 128 == 0) ? 2 : 1;
+    static const int mt128 = [] {
+        const char* e = getenv("PG_ATTN_MT128");
+        return e ? atoi(e) : 2;
+    }();
+    const int mt = (S % 128 != 0) ? 1 : (D == 64 ? 2 : mt128);
     dim3 grid(S / (BLOCK_M * mt), H, B);
 #define LAUNCH_FWD(DV, MTV)                                                   \
     hipLaunchKernelGGL((attn_fwd_kernel<DV, MTV>), grid, dim3(256), 0, stream,\
@@ -339,7 +343,10 @@ void attn_bwd_delta_kernel(const bf16* __restrict__ dout,
 }
 
 // pass A: dK(n,:) = scale * sum_m dS(m,n) Q(m,:);  dV(n,:) = sum_m P(m,n) dO(m,:)
-template <int D>
+// QR = q rows staged per iteration: 32 at D=128 keeps LDS under 1/3 of the
+// 160 KB CU budget (2-3 blocks/CU — at QR=64 the kernel was LDS-capped to
+// ONE wave/SIMD and ran with zero latency hiding).
+template <int D, int QR>
 __global__ __launch_bounds__(256)
 void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
                           const bf16* __restrict__ q, const bf16* __restrict__ k,
@@ -351,17 +358,17 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
                           int B, int H, int S) {
     constexpr int DCH = D / 32;
     constexpr int DSUB = D / 16;
-    constexpr int RSTRIDE = D + PAD;        // row-major tiles [64][D+PAD]
-    constexpr int TSTRIDE = BLOCK_M + PAD;  // transposed tiles [D][64+PAD]
+    constexpr int RSTRIDE = D + PAD;   // row-major tiles [QR][D+PAD]
+    constexpr int TSTRIDE = QR + PAD;  // transposed tiles [D][QR+PAD]
 
-    __shared__ bf16 do_lds[BLOCK_M * RSTRIDE];
-    __shared__ bf16 q_lds[BLOCK_M * RSTRIDE];
+    __shared__ bf16 do_lds[QR * RSTRIDE];
+    __shared__ bf16 q_lds[QR * RSTRIDE];
     __shared__ bf16 qt_lds[D * TSTRIDE];
     __shared__ bf16 dot_lds[D * TSTRIDE];
     __shared__ bf16 pt_lds[NWAVES][16 * TSTRIDE];   // P^T  (n rows, m cols)
     __shared__ bf16 dst_lds[NWAVES][16 * TSTRIDE];  // dS^T (n rows, m cols)
-    __shared__ float lse_lds[BLOCK_M];
-    __shared__ float delta_lds[BLOCK_M];
+    __shared__ float lse_lds[QR];
+    __shared__ float delta_lds[QR];
 
     const int nb = blockIdx.x;
     const int h = blockIdx.y;
@@ -397,36 +404,44 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
         accDV[s] = frag_cd{0.f, 0.f, 0.f, 0.f};
     }
 
-    for (int qb = nb; qb < S / BLOCK_M; ++qb) {
-        const int q0 = qb * BLOCK_M;
+    for (int q0 = nb * BLOCK_N; q0 < S; q0 += QR) {
         __syncthreads();
         {   // stage Q and dO, row-major + transposed, plus lse/delta
-            constexpr int PACKETS = BLOCK_M * D / 8;
+            constexpr int PACKETS = (QR / 2) * (D / 8);
             for (int p = tid; p < PACKETS; p += 256) {
-                const int row = p / (D / 8);
+                const int row = (p / (D / 8)) * 2;
                 const int col = (p % (D / 8)) * 8;
-                frag_ab pq = *reinterpret_cast<const frag_ab*>(
+                frag_ab pq0 = *reinterpret_cast<const frag_ab*>(
                     qp + (int64_t)(q0 + row) * D + col);
-                frag_ab pdo = *reinterpret_cast<const frag_ab*>(
+                frag_ab pq1 = *reinterpret_cast<const frag_ab*>(
+                    qp + (int64_t)(q0 + row + 1) * D + col);
+                frag_ab pd0 = *reinterpret_cast<const frag_ab*>(
                     dop + (int64_t)(q0 + row) * D + col);
-                *reinterpret_cast<frag_ab*>(&q_lds[row * RSTRIDE + col]) = pq;
-                *reinterpret_cast<frag_ab*>(&do_lds[row * RSTRIDE + col]) = pdo;
+                frag_ab pd1 = *reinterpret_cast<const frag_ab*>(
+                    dop + (int64_t)(q0 + row + 1) * D + col);
+                *reinterpret_cast<frag_ab*>(&q_lds[row * RSTRIDE + col]) = pq0;
+                *reinterpret_cast<frag_ab*>(&q_lds[(row + 1) * RSTRIDE + col]) = pq1;
+                *reinterpret_cast<frag_ab*>(&do_lds[row * RSTRIDE + col]) = pd0;
+                *reinterpret_cast<frag_ab*>(&do_lds[(row + 1) * RSTRIDE + col]) = pd1;
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
-                    qt_lds[(col + j) * TSTRIDE + row] = (bf16)pq[j];
-                    dot_lds[(col + j) * TSTRIDE + row] = (bf16)pdo[j];
+                    union { __bf16 h[2]; uint32_t u; } a, b2;
+                    a.h[0] = pq0[j]; a.h[1] = pq1[j];
+                    b2.h[0] = pd0[j]; b2.h[1] = pd1[j];
+                    *reinterpret_cast<uint32_t*>(&qt_lds[(col + j) * TSTRIDE + row]) = a.u;
+                    *reinterpret_cast<uint32_t*>(&dot_lds[(col + j) * TSTRIDE + row]) = b2.u;
                 }
             }
-            for (int i = tid; i < BLOCK_M; i += 256) {
+            for (int i = tid; i < QR; i += 256) {
                 lse_lds[i] = lse[bh_off + q0 + i];
                 delta_lds[i] = delta[bh_off + q0 + i];
             }
         }
         __syncthreads();
 
-        // build P^T and dS^T for ALL 64 q columns of this block
+        // build P^T and dS^T for ALL QR q columns of this chunk
 #pragma unroll
-        for (int ms = 0; ms < 4; ++ms) {
+        for (int ms = 0; ms < QR / 16; ++ms) {
             frag_cd st = frag_cd{0.f, 0.f, 0.f, 0.f};
             frag_cd dpt = frag_cd{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
@@ -452,9 +467,9 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
         }
         __builtin_amdgcn_s_waitcnt(0);  // own-wave LDS writes visible
 
-        // dV += P^T·dO and dK += dS^T·Q over the 64 q rows (2 K-chunks)
+        // dV += P^T·dO and dK += dS^T·Q over the QR q rows
 #pragma unroll
-        for (int kc = 0; kc < 2; ++kc) {
+        for (int kc = 0; kc < QR / 32; ++kc) {
             frag_ab aPT = *reinterpret_cast<const frag_ab*>(
                 &pt_lds[wave][lcol * TSTRIDE + kc * 32 + 8 * lgrp]);
             frag_ab aDST = *reinterpret_cast<const frag_ab*>(
@@ -486,7 +501,8 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
 }
 
 // pass B: dQ(m,:) = scale * sum_n dS(m,n) K(n,:)
-template <int D>
+// KVR = kv rows staged per iteration (same LDS-occupancy reasoning as QR).
+template <int D, int KVR>
 __global__ __launch_bounds__(256)
 void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
                         const bf16* __restrict__ q, const bf16* __restrict__ k,
@@ -499,11 +515,11 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
     constexpr int DCH = D / 32;
     constexpr int DSUB = D / 16;
     constexpr int RSTRIDE = D + PAD;
-    constexpr int TSTRIDE = BLOCK_N + PAD;
+    constexpr int TSTRIDE = KVR + PAD;
 
-    __shared__ bf16 k_lds[BLOCK_N * RSTRIDE];   // row-major K (QK^T B-frags)
-    __shared__ bf16 v_lds[BLOCK_N * RSTRIDE];   // row-major V (dP B-frags)
-    __shared__ bf16 kt_lds[D * TSTRIDE];        // transposed K (dQ B-frags)
+    __shared__ bf16 k_lds[KVR * RSTRIDE];   // row-major K (QK^T B-frags)
+    __shared__ bf16 v_lds[KVR * RSTRIDE];   // row-major V (dP B-frags)
+    __shared__ bf16 kt_lds[D * TSTRIDE];    // transposed K (dQ B-frags)
     __shared__ bf16 ds_lds[NWAVES][16 * TSTRIDE];  // dS (m rows, n cols)
 
     const int qb = blockIdx.x;
@@ -544,31 +560,39 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
 #pragma unroll
     for (int s = 0; s < DSUB; ++s) accDQ[s] = frag_cd{0.f, 0.f, 0.f, 0.f};
 
-    for (int nb = 0; nb <= qb; ++nb) {
-        const int kv0 = nb * BLOCK_N;
+    const int kv_end = qb * BLOCK_M + BLOCK_M;  // causal upper bound
+    for (int kv0 = 0; kv0 < kv_end; kv0 += KVR) {
         __syncthreads();
         {   // stage K (row-major + transposed) and V (row-major)
-            constexpr int PACKETS = BLOCK_N * D / 8;
+            constexpr int PACKETS = (KVR / 2) * (D / 8);
             for (int p = tid; p < PACKETS; p += 256) {
-                const int row = p / (D / 8);
+                const int row = (p / (D / 8)) * 2;
                 const int col = (p % (D / 8)) * 8;
-                frag_ab pk = *reinterpret_cast<const frag_ab*>(
+                frag_ab pk0 = *reinterpret_cast<const frag_ab*>(
                     kp + (int64_t)(kv0 + row) * D + col);
-                frag_ab pv = *reinterpret_cast<const frag_ab*>(
+                frag_ab pk1 = *reinterpret_cast<const frag_ab*>(
+                    kp + (int64_t)(kv0 + row + 1) * D + col);
+                frag_ab pv0 = *reinterpret_cast<const frag_ab*>(
                     vp + (int64_t)(kv0 + row) * D + col);
-                *reinterpret_cast<frag_ab*>(&k_lds[row * RSTRIDE + col]) = pk;
-                *reinterpret_cast<frag_ab*>(&v_lds[row * RSTRIDE + col]) = pv;
+                frag_ab pv1 = *reinterpret_cast<const frag_ab*>(
+                    vp + (int64_t)(kv0 + row + 1) * D + col);
+                *reinterpret_cast<frag_ab*>(&k_lds[row * RSTRIDE + col]) = pk0;
+                *reinterpret_cast<frag_ab*>(&k_lds[(row + 1) * RSTRIDE + col]) = pk1;
+                *reinterpret_cast<frag_ab*>(&v_lds[row * RSTRIDE + col]) = pv0;
+                *reinterpret_cast<frag_ab*>(&v_lds[(row + 1) * RSTRIDE + col]) = pv1;
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
-                    kt_lds[(col + j) * TSTRIDE + row] = (bf16)pk[j];
+                    union { __bf16 h[2]; uint32_t u; } a;
+                    a.h[0] = pk0[j]; a.h[1] = pk1[j];
+                    *reinterpret_cast<uint32_t*>(&kt_lds[(col + j) * TSTRIDE + row]) = a.u;
                 }
             }
         }
         __syncthreads();
 
-        // dS for all 64 kv cols of this block
+        // dS for all KVR kv cols of this chunk
 #pragma unroll
-        for (int ns = 0; ns < 4; ++ns) {
+        for (int ns = 0; ns < KVR / 16; ++ns) {
             frag_cd sacc = frag_cd{0.f, 0.f, 0.f, 0.f};
             frag_cd dpacc = frag_cd{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
@@ -594,9 +618,9 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
         }
         __builtin_amdgcn_s_waitcnt(0);
 
-        // dQ += dS·K over the 64 kv rows (2 K-chunks)
+        // dQ += dS·K over the KVR kv rows
 #pragma unroll
-        for (int kc = 0; kc < 2; ++kc) {
+        for (int kc = 0; kc < KVR / 32; ++kc) {
             frag_ab aDS = *reinterpret_cast<const frag_ab*>(
                 &ds_lds[wave][lcol * TSTRIDE + kc * 32 + 8 * lgrp]);
 #pragma unroll
@@ -644,7 +668,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
             reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
             reinterpret_cast<const bf16*>(o.data_ptr()),                      \
             delta.data_ptr<float>(), rows);                                   \
-        hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DV>), grid_a, dim3(256), 0,  \
+        hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DV, 32>),\
+            grid_a, dim3(256), 0,                                             \
             stream,                                                           \
             reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
             reinterpret_cast<const bf16*>(q.data_ptr()),                      \
@@ -654,8 +679,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
             slopes.data_ptr<float>(), (float)scale,                           \
             reinterpret_cast<bf16*>(dk.data_ptr()),                           \
             reinterpret_cast<bf16*>(dv.data_ptr()), B, H, S);                 \
-        hipLaunchKernelGGL((attn_bwd_dq_kernel<DV>), grid_b, dim3(256), 0,    \
-            stream,                                                           \
+        hipLaunchKernelGGL((attn_bwd_dq_kernel<DV, 32>),  \
+            grid_b, dim3(256), 0, stream,                                                           \
             reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
             reinterpret_cast<const bf16*>(q.data_ptr()),                      \
             reinterpret_cast<const bf16*>(k.data_ptr()),                      \
