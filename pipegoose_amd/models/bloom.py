@@ -174,6 +174,21 @@ class BloomBlock(nn.Module):
             attn_out, hidden)
         return hidden + self.mlp(normed)
 
+    def forward_chained(self, hidden: torch.Tensor, pending):
+        """Residual-chained form: the PREVIOUS block's mlp output arrives as
+        ``pending`` and its residual add fuses into this block's input norm
+        (BloomModel.forward drives this; plain ``forward`` keeps the
+        single-tensor contract for pipeline stages)."""
+        if pending is not None:
+            normed, hidden = self.input_layernorm.forward_with_residual(
+                pending, hidden)
+        else:
+            normed = self.input_layernorm(hidden)
+        attn_out = self.self_attention(normed)
+        normed2, hidden = self.post_attention_layernorm.forward_with_residual(
+            attn_out, hidden)
+        return hidden, self.mlp(normed2)
+
 
 class BloomModel(nn.Module):
     def __init__(self, config: BloomConfig, parallel_context: ParallelContext):
@@ -197,9 +212,14 @@ class BloomModel(nn.Module):
 
     def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
         hidden = self.word_embeddings_layernorm(self.word_embeddings(input_ids))
+        # residual-chained blocks: every residual add fuses into a norm's
+        # HBM pass (2 fused adds per block + the final one into ln_f)
+        pending = None
         for block in self.h:
-            hidden = block(hidden)
-        return self.ln_f(hidden)
+            hidden, pending = block.forward_chained(hidden, pending)
+        normed, _ = self.ln_f.forward_with_residual(pending, hidden) \
+            if pending is not None else (self.ln_f(hidden), hidden)
+        return normed
 
 
 def make_causal_lm_loss(parallel_context: ParallelContext):

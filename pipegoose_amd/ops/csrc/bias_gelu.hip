@@ -65,6 +65,44 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy, const T* __restri
     }
 }
 
+
+// backward with the bias-grad reduction FUSED: dx written and db (fp32
+// atomics) accumulated in the same HBM pass — eager needed a second full
+// read of [*, C] for the column sum (showed as reduce_kernel ~2% of step).
+template <typename T, int BLOCK>
+__global__ void bias_gelu_bwd_fused_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const T* __restrict__ bias, T* __restrict__ dx, float* __restrict__ db,
+    int64_t N, int C) {
+    using V = typename vec8<T>::type;
+    const int CV = C / 8;
+    const int cp = blockIdx.x * BLOCK + threadIdx.x;
+    if (cp >= CV) return;
+    V bv = reinterpret_cast<const V*>(bias)[cp];
+    float bf[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) bf[j] = to_float(bv.v[j]);
+    float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int64_t r = blockIdx.y; r < N; r += gridDim.y) {
+        const int64_t i = r * CV + cp;
+        V dyv = reinterpret_cast<const V*>(dy)[i];
+        V xv = reinterpret_cast<const V*>(x)[i];
+        V out;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float u = to_float(xv.v[j]) + bf[j];
+            float g = to_float(dyv.v[j]) * dgelu_tanh(u);
+            out.v[j] = from_float<T>(g);
+            acc[j] += g;
+        }
+        reinterpret_cast<V*>(dx)[i] = out;
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        atomicAdd(&db[cp * 8 + j], acc[j]);
+    }
+}
+
 int grid_for(int64_t n_items, int block) {
     // memory-bound: cap at ~2048 blocks, grid-stride the rest (G11)
     int64_t blocks = (n_items + block - 1) / block;
@@ -97,27 +135,34 @@ torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias) {
     return y;
 }
 
-torch::Tensor bias_gelu_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor bias) {
+std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor bias) {
     TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && x.is_contiguous());
     const int C = (int)bias.numel();
+    TORCH_CHECK(C % 8 == 0);
     auto dx = torch::empty_like(x);
-    const int64_t n = x.numel();
+    auto db = torch::zeros({C}, x.options().dtype(torch::kFloat));
+    const int64_t N = x.numel() / C;
     constexpr int BLOCK = 256;
-    int grid = grid_for(n / 8, BLOCK);
+    const int CV = C / 8;
+    dim3 grid((CV + BLOCK - 1) / BLOCK, (int)std::min<int64_t>(N, 512));
     auto stream = at::cuda::getCurrentCUDAStream();
     if (x.scalar_type() == torch::kBFloat16) {
-        hipLaunchKernelGGL((bias_gelu_bwd_kernel<__hip_bfloat16>), dim3(grid), dim3(BLOCK), 0, stream,
+        hipLaunchKernelGGL((bias_gelu_bwd_fused_kernel<__hip_bfloat16, BLOCK>),
+            grid, dim3(BLOCK), 0, stream,
             reinterpret_cast<const __hip_bfloat16*>(dy.data_ptr()),
             reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
             reinterpret_cast<const __hip_bfloat16*>(bias.data_ptr()),
-            reinterpret_cast<__hip_bfloat16*>(dx.data_ptr()), n, C);
+            reinterpret_cast<__hip_bfloat16*>(dx.data_ptr()),
+            db.data_ptr<float>(), N, C);
     } else if (x.scalar_type() == torch::kFloat) {
-        hipLaunchKernelGGL((bias_gelu_bwd_kernel<float>), dim3(grid), dim3(BLOCK), 0, stream,
+        hipLaunchKernelGGL((bias_gelu_bwd_fused_kernel<float, BLOCK>),
+            grid, dim3(BLOCK), 0, stream,
             dy.data_ptr<float>(), x.data_ptr<float>(), bias.data_ptr<float>(),
-            dx.data_ptr<float>(), n, C);
+            dx.data_ptr<float>(), db.data_ptr<float>(), N, C);
     } else {
         TORCH_CHECK(false, "bias_gelu_bwd: unsupported dtype");
     }
     HIP_CHECK_LAUNCH();
-    return dx;
+    return {dx, db.to(x.scalar_type())};
 }

@@ -10,7 +10,8 @@ std::vector<torch::Tensor> layer_norm_res_fwd(torch::Tensor x, torch::Tensor r,
                                               torch::Tensor w, torch::Tensor b,
                                               double eps);
 torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias);
-torch::Tensor bias_gelu_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor bias);
+std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor bias);
 std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits, torch::Tensor targets,
                                              int64_t vocab_start, int64_t vocab_end);
 torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor targets,

@@ -24,9 +24,8 @@ class _BiasGelu(torch.autograd.Function):
     def backward(ctx, grad_out):
         ext = get_extension(required=True)
         input, bias = ctx.saved_tensors
-        grad_in = ext.bias_gelu_bwd(grad_out.contiguous(), input, bias)
-        # bias grad: reduce over all leading dims
-        grad_bias = grad_in.reshape(-1, grad_in.size(-1)).sum(dim=0)
+        # bias grad accumulated in the same HBM pass as dx (fp32 atomics)
+        grad_in, grad_bias = ext.bias_gelu_bwd(grad_out.contiguous(), input, bias)
         return grad_in, grad_bias
 
 

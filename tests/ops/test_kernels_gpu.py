@@ -67,10 +67,15 @@ def test_bias_gelu(dtype):
     assert torch.allclose(y.float(), ref, atol=tol), (y.float() - ref).abs().max()
 
     dy = torch.randn_like(x)
-    dx = ext.bias_gelu_bwd(dy, x, bias)
+    dx, db = ext.bias_gelu_bwd(dy, x, bias)
     xf = (x.float() + bias.float()).detach().requires_grad_(True)
     torch.nn.functional.gelu(xf, approximate="tanh").backward(dy.float())
     assert torch.allclose(dx.float(), xf.grad, atol=tol * 3), (dx.float() - xf.grad).abs().max()
+    # fused bias grad == column sum of dx
+    ref_db = xf.grad.sum(dim=0)
+    rtol = 1e-3 if dtype == torch.float32 else 3e-2
+    assert torch.allclose(db.float(), ref_db, rtol=rtol, atol=tol * N), \
+        (db.float() - ref_db).abs().max()
 
 
 def test_extension_is_native():
