@@ -94,6 +94,9 @@ class RowParallelLinear(nn.Module):
 
     def forward(self, input: torch.Tensor) -> torch.Tensor:
         tp = self.parallel_context.get_world_size(ParallelMode.TENSOR)
+        if tp == 1:
+            # no combine needed: let hipBLASLt fuse the bias epilogue
+            return TF.linear(input, self.weight, self.bias)
         if input.size(-1) == self.in_features * tp:
             input = scatter_to_tensor_group(input, dim=-1,
                                             parallel_context=self.parallel_context)

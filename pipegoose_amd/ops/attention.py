@@ -25,6 +25,12 @@ def _kernel_supported(q: torch.Tensor) -> bool:
     return D in (64, 128) and S % 64 == 0 and q.dtype == torch.bfloat16
 
 
+def _d_contig(t: torch.Tensor) -> torch.Tensor:
+    # kernels take arbitrary [B,H,S,D] strides as long as d is contiguous —
+    # the model's transposed views qualify, so no copies happen here
+    return t if t.stride(-1) == 1 else t.contiguous()
+
+
 class _AlibiFlashAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, slopes, scale):
@@ -38,15 +44,17 @@ class _AlibiFlashAttention(torch.autograd.Function):
     def backward(ctx, do):
         ext = get_extension(required=True)
         q, k, v, o, lse, slopes = ctx.saved_tensors
-        dq, dk, dv = ext.attn_bwd(do.contiguous(), q, k, v, o, lse, slopes, ctx.scale)
+        dq, dk, dv = ext.attn_bwd(_d_contig(do), q, k, v, o, lse, slopes,
+                                  ctx.scale)
         return dq, dk, dv, None, None
 
 
 def alibi_attention(q, k, v, slopes, scale, mask_fallback=None):
-    """q,k,v: [B, H, S, D]; slopes: [H] fp32; causal + alibi bias in-kernel."""
+    """q,k,v: [B, H, S, D] (any strides, d contiguous); slopes: [H] fp32;
+    causal + alibi bias computed in-kernel."""
     if _kernel_supported(q):
         return _AlibiFlashAttention.apply(
-            q.contiguous(), k.contiguous(), v.contiguous(),
+            _d_contig(q), _d_contig(k), _d_contig(v),
             slopes.to(device=q.device, dtype=torch.float32), scale)
     bias = mask_fallback(q.size(-2), q.device, q.dtype)
     return TF.scaled_dot_product_attention(q, k, v, attn_mask=bias, scale=scale)

@@ -35,13 +35,20 @@ constexpr float NEG_INF = -1e30f;
 // B-fragment (N=16 side): col = lane&15, k = 8*(lane>>4) + i
 // C/D:                    col = lane&15, row = 4*(lane>>4) + reg
 
-template <int D, int MT>
-__global__ __launch_bounds__(256)
+template <int D, int MT, int WAVES = NWAVES>
+__global__ __launch_bounds__(WAVES * WAVE_SIZE)
 void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
                      const bf16* __restrict__ v,
                      const float* __restrict__ slopes, float scale,
                      bf16* __restrict__ o, float* __restrict__ lse,
-                     int B, int H, int S) {
+                     int B, int H, int S,
+                     int64_t qb, int64_t qh, int64_t qs,
+                     int64_t kb, int64_t kh, int64_t ks,
+                     int64_t vb, int64_t vh, int64_t vs,
+                     int64_t ob, int64_t oh, int64_t os) {
+    // Tensors are [B, H, S, D] LOGICAL with arbitrary strides (d contiguous):
+    // the model's qkv views come straight from the fused projection with no
+    // .contiguous() copies (those copies were ~6% of the training step).
     // MT = q row-tiles per wave (16 rows each).  MT=2 doubles the MFMA work
     // per LDS B-fragment read — the K/V tiles are read once per wave and
     // used for 32 q rows.
@@ -49,11 +56,12 @@ void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
     constexpr int DSUB = D / 16;   // output d-subtiles
     constexpr int KSTRIDE = D + PAD;
     constexpr int VSTRIDE = BLOCK_N + PAD;
-    constexpr int ROWS_PER_WG = NWAVES * MT * 16;
+    constexpr int ROWS_PER_WG = WAVES * MT * 16;
+    constexpr int NT = WAVES * WAVE_SIZE;
 
     __shared__ bf16 k_lds[BLOCK_N * KSTRIDE];
     __shared__ bf16 vt_lds[D * VSTRIDE];
-    __shared__ bf16 p_lds[NWAVES][MT * 16 * VSTRIDE];
+    __shared__ bf16 p_lds[WAVES][MT * 16 * VSTRIDE];
 
     const int qblock = blockIdx.x;
     const int h = blockIdx.y;
@@ -64,10 +72,11 @@ void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
     const int lgrp = lane >> 4;       // 0..3
     const int lcol = lane & 15;       // 0..15
 
-    const int64_t bh_off = ((int64_t)b * H + h) * S;
-    const bf16* qp = q + (bh_off) * D;
-    const bf16* kp = k + (bh_off) * D;
-    const bf16* vp = v + (bh_off) * D;
+    const int64_t bh_off = ((int64_t)b * H + h) * S;  // lse layout
+    const bf16* qp = q + b * qb + h * qh;
+    const bf16* kp = k + b * kb + h * kh;
+    const bf16* vp = v + b * vb + h * vh;
+    bf16* op = o + b * ob + h * oh;
 
     // wave owns rows [qrow0, qrow0 + MT*16)
     const int qrow0 = qblock * ROWS_PER_WG + wave * (MT * 16);
@@ -76,7 +85,7 @@ void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
     frag_ab aQ[MT][DCH];
 #pragma unroll
     for (int mt = 0; mt < MT; ++mt) {
-        const bf16* qrow = qp + (int64_t)(qrow0 + mt * 16 + lcol) * D;
+        const bf16* qrow = qp + (int64_t)(qrow0 + mt * 16 + lcol) * qs;
 #pragma unroll
         for (int c = 0; c < DCH; ++c) {
             aQ[mt][c] = *reinterpret_cast<const frag_ab*>(qrow + c * 32 + 8 * lgrp);
@@ -99,23 +108,23 @@ void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
         __syncthreads();
         {   // cooperative K tile load (row-major, 16B packets)
             constexpr int PACKETS = BLOCK_N * D / 8;
-            for (int p = tid; p < PACKETS; p += 256) {
+            for (int p = tid; p < PACKETS; p += NT) {
                 const int row = p / (D / 8);
                 const int col = (p % (D / 8)) * 8;
                 *reinterpret_cast<frag_ab*>(&k_lds[row * KSTRIDE + col]) =
                     *reinterpret_cast<const frag_ab*>(
-                        kp + (int64_t)(kvrow0 + row) * D + col);
+                        kp + (int64_t)(kvrow0 + row) * ks + col);
             }
         }
         {   // V tile load, transposed; two rows at once -> 4 B column stores
             constexpr int PACKETS = (BLOCK_N / 2) * (D / 8);
-            for (int p = tid; p < PACKETS; p += 256) {
+            for (int p = tid; p < PACKETS; p += NT) {
                 const int row = (p / (D / 8)) * 2;
                 const int col = (p % (D / 8)) * 8;
                 frag_ab p0 = *reinterpret_cast<const frag_ab*>(
-                    vp + (int64_t)(kvrow0 + row) * D + col);
+                    vp + (int64_t)(kvrow0 + row) * vs + col);
                 frag_ab p1 = *reinterpret_cast<const frag_ab*>(
-                    vp + (int64_t)(kvrow0 + row + 1) * D + col);
+                    vp + (int64_t)(kvrow0 + row + 1) * vs + col);
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
                     union { __bf16 h[2]; uint32_t u; } pair;
@@ -225,7 +234,7 @@ void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
         for (int r = 0; r < 4; ++r) {
             const int iq = qrow0 + mt * 16 + 4 * lgrp + r;
             const float inv_l = 1.0f / l_run[mt][r];
-            bf16* orow = o + (bh_off + iq) * D;
+            bf16* orow = op + (int64_t)iq * os;
 #pragma unroll
             for (int ds = 0; ds < DSUB; ++ds) {
                 orow[ds * 16 + lcol] = (bf16)(accO[mt][ds][r] * inv_l);
@@ -262,37 +271,56 @@ __global__ void mfma_probe_kernel(const bf16* __restrict__ A,
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor slopes,
                                     double scale) {
-    TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
-                v.is_contiguous());
+    TORCH_CHECK(q.is_cuda() && q.stride(3) == 1 && k.stride(3) == 1 &&
+                v.stride(3) == 1, "attn_fwd: last dim must be contiguous");
     TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "attn_fwd: bf16 only");
     const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
     TORCH_CHECK(S % BLOCK_M == 0, "attn_fwd: S must be a multiple of 64");
     TORCH_CHECK(D == 64 || D == 128, "attn_fwd: head dim 64 or 128");
     TORCH_CHECK(slopes.numel() == H && slopes.scalar_type() == torch::kFloat);
 
-    auto o = torch::empty_like(q);
+    // O physically [B, S, H, D]: the model reshapes attention output back to
+    // [B, S, H*D] for the dense projection — this layout makes that free.
+    auto o_phys = torch::empty({B, S, H, D}, q.options());
+    auto o = o_phys.permute({0, 2, 1, 3});
     auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
 
     auto stream = at::cuda::getCurrentCUDAStream();
     // MT=2 (128 q rows / workgroup) when S allows: 2x the MFMA work per LDS
     // fragment read
-    static const int mt128 = [] {
-        const char* e = getenv("PG_ATTN_MT128");
+    // D=64: 4 waves x 2 row-tiles.  D=128: default 8 waves x 1 tile
+    // (same 128 q rows, lower per-wave registers); PG_ATTN_V128 picks the
+    // variant (1: 4w/MT1, 2: 4w/MT2, 3: 8w/MT1).
+    static const int v128 = [] {
+        const char* e = getenv("PG_ATTN_V128");
         return e ? atoi(e) : 2;
     }();
-    const int mt = (S % 128 != 0) ? 1 : (D == 64 ? 2 : mt128);
-    dim3 grid(S / (BLOCK_M * mt), H, B);
-#define LAUNCH_FWD(DV, MTV)                                                   \
-    hipLaunchKernelGGL((attn_fwd_kernel<DV, MTV>), grid, dim3(256), 0, stream,\
-        reinterpret_cast<const bf16*>(q.data_ptr()),                          \
-        reinterpret_cast<const bf16*>(k.data_ptr()),                          \
-        reinterpret_cast<const bf16*>(v.data_ptr()),                          \
-        slopes.data_ptr<float>(), (float)scale,                               \
-        reinterpret_cast<bf16*>(o.data_ptr()), lse.data_ptr<float>(), B, H, S)
+#define LAUNCH_FWD(DV, MTV, WV)                                               \
+    do {                                                                      \
+        dim3 grid(S / (16 * MTV * WV), H, B);                                 \
+        hipLaunchKernelGGL((attn_fwd_kernel<DV, MTV, WV>), grid,              \
+            dim3(WV * WAVE_SIZE), 0, stream,                                  \
+            reinterpret_cast<const bf16*>(q.data_ptr()),                      \
+            reinterpret_cast<const bf16*>(k.data_ptr()),                      \
+            reinterpret_cast<const bf16*>(v.data_ptr()),                      \
+            slopes.data_ptr<float>(), (float)scale,                           \
+            reinterpret_cast<bf16*>(o_phys.data_ptr()), lse.data_ptr<float>(),\
+            B, H, S,                                                          \
+            q.stride(0), q.stride(1), q.stride(2),                            \
+            k.stride(0), k.stride(1), k.stride(2),                            \
+            v.stride(0), v.stride(1), v.stride(2),                            \
+            o.stride(0), o.stride(1), o.stride(2));                           \
+    } while (0)
     if (D == 64) {
-        if (mt == 2) LAUNCH_FWD(64, 2); else LAUNCH_FWD(64, 1);
+        if (S % 128 == 0) LAUNCH_FWD(64, 2, 4); else LAUNCH_FWD(64, 1, 4);
+    } else if (S % 128 != 0) {
+        LAUNCH_FWD(128, 1, 4);
+    } else if (v128 == 3) {
+        LAUNCH_FWD(128, 1, 8);
+    } else if (v128 == 2) {
+        LAUNCH_FWD(128, 2, 4);
     } else {
-        if (mt == 2) LAUNCH_FWD(128, 2); else LAUNCH_FWD(128, 1);
+        LAUNCH_FWD(128, 1, 4);
     }
 #undef LAUNCH_FWD
     HIP_CHECK_LAUNCH();
@@ -328,12 +356,18 @@ template <int D>
 __global__ __launch_bounds__(256)
 void attn_bwd_delta_kernel(const bf16* __restrict__ dout,
                            const bf16* __restrict__ o,
-                           float* __restrict__ delta, int64_t rows) {
+                           float* __restrict__ delta, int64_t rows,
+                           int H, int S,
+                           int64_t db, int64_t dh, int64_t ds_,
+                           int64_t ob2, int64_t oh2, int64_t os2) {
     const int64_t row = (int64_t)blockIdx.x * 4 + threadIdx.x / WAVE_SIZE;
     if (row >= rows) return;
     const int lane = threadIdx.x % WAVE_SIZE;
-    const bf16* dr = dout + row * D;
-    const bf16* orow = o + row * D;
+    const int64_t b = row / ((int64_t)H * S);
+    const int64_t h = (row / S) % H;
+    const int64_t s = row % S;
+    const bf16* dr = dout + b * db + h * dh + s * ds_;
+    const bf16* orow = o + b * ob2 + h * oh2 + s * os2;
     float acc = 0.f;
     for (int i = lane; i < D; i += WAVE_SIZE) {
         acc += to_float(dr[i]) * to_float(orow[i]);
@@ -355,7 +389,12 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
                           const float* __restrict__ delta,
                           const float* __restrict__ slopes, float scale,
                           bf16* __restrict__ dk, bf16* __restrict__ dv,
-                          int B, int H, int S) {
+                          int B, int H, int S,
+                          int64_t gb, int64_t gh, int64_t gs,
+                          int64_t qb2, int64_t qh2, int64_t qs2,
+                          int64_t kb2, int64_t kh2, int64_t ks2,
+                          int64_t vb2, int64_t vh2, int64_t vs2,
+                          int64_t wb, int64_t wh, int64_t ws) {
     constexpr int DCH = D / 32;
     constexpr int DSUB = D / 16;
     constexpr int RSTRIDE = D + PAD;   // row-major tiles [QR][D+PAD]
@@ -378,9 +417,9 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
     const int lane = tid % WAVE_SIZE;
     const int lgrp = lane >> 4, lcol = lane & 15;
 
-    const int64_t bh_off = ((int64_t)b * H + h) * S;
-    const bf16* qp = q + bh_off * D;
-    const bf16* dop = dout + bh_off * D;
+    const int64_t bh_off = ((int64_t)b * H + h) * S;  // lse/delta layout
+    const bf16* qp = q + b * qb2 + h * qh2;
+    const bf16* dop = dout + b * gb + h * gh;
     const float slope = slopes[h];
 
     const int kv0 = nb * BLOCK_N + wave * 16;  // this wave's 16 kv rows
@@ -388,8 +427,8 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
     // this wave's K and V rows live in registers (A-fragments over d)
     frag_ab aK[DCH], aV[DCH];
     {
-        const bf16* krow = k + (bh_off + kv0 + lcol) * D;
-        const bf16* vrow = v + (bh_off + kv0 + lcol) * D;
+        const bf16* krow = k + b * kb2 + h * kh2 + (int64_t)(kv0 + lcol) * ks2;
+        const bf16* vrow = v + b * vb2 + h * vh2 + (int64_t)(kv0 + lcol) * vs2;
 #pragma unroll
         for (int c = 0; c < DCH; ++c) {
             aK[c] = *reinterpret_cast<const frag_ab*>(krow + c * 32 + 8 * lgrp);
@@ -412,13 +451,13 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
                 const int row = (p / (D / 8)) * 2;
                 const int col = (p % (D / 8)) * 8;
                 frag_ab pq0 = *reinterpret_cast<const frag_ab*>(
-                    qp + (int64_t)(q0 + row) * D + col);
+                    qp + (int64_t)(q0 + row) * qs2 + col);
                 frag_ab pq1 = *reinterpret_cast<const frag_ab*>(
-                    qp + (int64_t)(q0 + row + 1) * D + col);
+                    qp + (int64_t)(q0 + row + 1) * qs2 + col);
                 frag_ab pd0 = *reinterpret_cast<const frag_ab*>(
-                    dop + (int64_t)(q0 + row) * D + col);
+                    dop + (int64_t)(q0 + row) * gs + col);
                 frag_ab pd1 = *reinterpret_cast<const frag_ab*>(
-                    dop + (int64_t)(q0 + row + 1) * D + col);
+                    dop + (int64_t)(q0 + row + 1) * gs + col);
                 *reinterpret_cast<frag_ab*>(&q_lds[row * RSTRIDE + col]) = pq0;
                 *reinterpret_cast<frag_ab*>(&q_lds[(row + 1) * RSTRIDE + col]) = pq1;
                 *reinterpret_cast<frag_ab*>(&do_lds[row * RSTRIDE + col]) = pd0;
@@ -490,8 +529,8 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
         const int n_glob = kv0 + 4 * lgrp + r;
-        bf16* dkrow = dk + (bh_off + n_glob) * D;
-        bf16* dvrow = dv + (bh_off + n_glob) * D;
+        bf16* dkrow = dk + b * wb + h * wh + (int64_t)n_glob * ws;
+        bf16* dvrow = dv + b * wb + h * wh + (int64_t)n_glob * ws;
 #pragma unroll
         for (int ds = 0; ds < DSUB; ++ds) {
             dkrow[ds * 16 + lcol] = (bf16)accDK[ds][r];
@@ -511,7 +550,12 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
                         const float* __restrict__ delta,
                         const float* __restrict__ slopes, float scale,
                         bf16* __restrict__ dq,
-                        int B, int H, int S) {
+                        int B, int H, int S,
+                        int64_t gb, int64_t gh, int64_t gs,
+                        int64_t qb2, int64_t qh2, int64_t qs2,
+                        int64_t kb2, int64_t kh2, int64_t ks2,
+                        int64_t vb2, int64_t vh2, int64_t vs2,
+                        int64_t wb, int64_t wh, int64_t ws) {
     constexpr int DCH = D / 32;
     constexpr int DSUB = D / 16;
     constexpr int RSTRIDE = D + PAD;
@@ -530,9 +574,9 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
     const int lane = tid % WAVE_SIZE;
     const int lgrp = lane >> 4, lcol = lane & 15;
 
-    const int64_t bh_off = ((int64_t)b * H + h) * S;
-    const bf16* kp = k + bh_off * D;
-    const bf16* vp = v + bh_off * D;
+    const int64_t bh_off = ((int64_t)b * H + h) * S;  // lse/delta layout
+    const bf16* kp = k + b * kb2 + h * kh2;
+    const bf16* vp = v + b * vb2 + h * vh2;
     const float slope = slopes[h];
 
     const int qrow0 = qb * BLOCK_M + wave * 16;
@@ -540,8 +584,8 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
     // Q and dO rows of this wave in registers; lse/delta per owned row
     frag_ab aQ[DCH], aDO[DCH];
     {
-        const bf16* qrow = q + (bh_off + qrow0 + lcol) * D;
-        const bf16* dorow = dout + (bh_off + qrow0 + lcol) * D;
+        const bf16* qrow = q + b * qb2 + h * qh2 + (int64_t)(qrow0 + lcol) * qs2;
+        const bf16* dorow = dout + b * gb + h * gh + (int64_t)(qrow0 + lcol) * gs;
 #pragma unroll
         for (int c = 0; c < DCH; ++c) {
             aQ[c] = *reinterpret_cast<const frag_ab*>(qrow + c * 32 + 8 * lgrp);
@@ -569,13 +613,13 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
                 const int row = (p / (D / 8)) * 2;
                 const int col = (p % (D / 8)) * 8;
                 frag_ab pk0 = *reinterpret_cast<const frag_ab*>(
-                    kp + (int64_t)(kv0 + row) * D + col);
+                    kp + (int64_t)(kv0 + row) * ks2 + col);
                 frag_ab pk1 = *reinterpret_cast<const frag_ab*>(
-                    kp + (int64_t)(kv0 + row + 1) * D + col);
+                    kp + (int64_t)(kv0 + row + 1) * ks2 + col);
                 frag_ab pv0 = *reinterpret_cast<const frag_ab*>(
-                    vp + (int64_t)(kv0 + row) * D + col);
+                    vp + (int64_t)(kv0 + row) * vs2 + col);
                 frag_ab pv1 = *reinterpret_cast<const frag_ab*>(
-                    vp + (int64_t)(kv0 + row + 1) * D + col);
+                    vp + (int64_t)(kv0 + row + 1) * vs2 + col);
                 *reinterpret_cast<frag_ab*>(&k_lds[row * RSTRIDE + col]) = pk0;
                 *reinterpret_cast<frag_ab*>(&k_lds[(row + 1) * RSTRIDE + col]) = pk1;
                 *reinterpret_cast<frag_ab*>(&v_lds[row * RSTRIDE + col]) = pv0;
@@ -635,7 +679,7 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
         const int iq = qrow0 + 4 * lgrp + r;
-        bf16* dqrow = dq + (bh_off + iq) * D;
+        bf16* dqrow = dq + b * wb + h * wh + (int64_t)iq * ws;
 #pragma unroll
         for (int ds = 0; ds < DSUB; ++ds) {
             dqrow[ds * 16 + lcol] = (bf16)accDQ[ds][r];
@@ -649,11 +693,17 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
                                     torch::Tensor slopes, double scale) {
-    TORCH_CHECK(dout.is_cuda() && dout.is_contiguous());
+    TORCH_CHECK(dout.is_cuda() && dout.stride(3) == 1 && q.stride(3) == 1 &&
+                k.stride(3) == 1 && v.stride(3) == 1 && o.stride(3) == 1,
+                "attn_bwd: last dim must be contiguous");
     const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
-    auto dq = torch::empty_like(q);
-    auto dk = torch::empty_like(k);
-    auto dv = torch::empty_like(v);
+    // grads physically [B, S, H, D] (the layout the fused-qkv backward wants)
+    auto dq_phys = torch::empty({B, S, H, D}, q.options());
+    auto dk_phys = torch::empty({B, S, H, D}, q.options());
+    auto dv_phys = torch::empty({B, S, H, D}, q.options());
+    auto dq = dq_phys.permute({0, 2, 1, 3});
+    auto dk = dk_phys.permute({0, 2, 1, 3});
+    auto dv = dv_phys.permute({0, 2, 1, 3});
     auto delta = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
 
     const int64_t rows = (int64_t)B * H * S;
@@ -661,33 +711,35 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
     dim3 grid_a(S / BLOCK_N, H, B);
     dim3 grid_b(S / BLOCK_M, H, B);
 
+#define STR3(t) t.stride(0), t.stride(1), t.stride(2)
 #define LAUNCH_BWD(DV)                                                        \
     do {                                                                      \
         hipLaunchKernelGGL((attn_bwd_delta_kernel<DV>),                       \
             dim3((rows + 3) / 4), dim3(256), 0, stream,                       \
             reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
             reinterpret_cast<const bf16*>(o.data_ptr()),                      \
-            delta.data_ptr<float>(), rows);                                   \
-        hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DV, 32>),\
-            grid_a, dim3(256), 0,                                             \
-            stream,                                                           \
+            delta.data_ptr<float>(), rows, H, S, STR3(dout), STR3(o));        \
+        hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DV, 32>),                    \
+            grid_a, dim3(256), 0, stream,                                     \
             reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
             reinterpret_cast<const bf16*>(q.data_ptr()),                      \
             reinterpret_cast<const bf16*>(k.data_ptr()),                      \
             reinterpret_cast<const bf16*>(v.data_ptr()),                      \
             lse.data_ptr<float>(), delta.data_ptr<float>(),                   \
             slopes.data_ptr<float>(), (float)scale,                           \
-            reinterpret_cast<bf16*>(dk.data_ptr()),                           \
-            reinterpret_cast<bf16*>(dv.data_ptr()), B, H, S);                 \
-        hipLaunchKernelGGL((attn_bwd_dq_kernel<DV, 32>),  \
-            grid_b, dim3(256), 0, stream,                                                           \
+            reinterpret_cast<bf16*>(dk_phys.data_ptr()),                      \
+            reinterpret_cast<bf16*>(dv_phys.data_ptr()), B, H, S,             \
+            STR3(dout), STR3(q), STR3(k), STR3(v), STR3(dk));                 \
+        hipLaunchKernelGGL((attn_bwd_dq_kernel<DV, 32>),                      \
+            grid_b, dim3(256), 0, stream,                                     \
             reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
             reinterpret_cast<const bf16*>(q.data_ptr()),                      \
             reinterpret_cast<const bf16*>(k.data_ptr()),                      \
             reinterpret_cast<const bf16*>(v.data_ptr()),                      \
             lse.data_ptr<float>(), delta.data_ptr<float>(),                   \
             slopes.data_ptr<float>(), (float)scale,                           \
-            reinterpret_cast<bf16*>(dq.data_ptr()), B, H, S);                 \
+            reinterpret_cast<bf16*>(dq_phys.data_ptr()), B, H, S,             \
+            STR3(dout), STR3(q), STR3(k), STR3(v), STR3(dq));                 \
     } while (0)
 
     if (D == 64) {
@@ -697,6 +749,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
         LAUNCH_BWD(128);
     }
 #undef LAUNCH_BWD
+#undef STR3
     HIP_CHECK_LAUNCH();
     return {dq, dk, dv};
 }

@@ -243,7 +243,7 @@ std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
     // fp32 atomics combine partials (512 adds/column max — low contention).
     const bool vec = (H % 8 == 0);
     const int64_t cols = vec ? H / 8 : H;
-    int grid_y = (int)std::min<int64_t>(N, 512);
+    int grid_y = (int)std::min<int64_t>(N, 4096);  // fill the chip: 256 CUs want >>512 blocks when grid.x is 1-2
     dim3 grid_dw((cols + BLOCK - 1) / BLOCK, grid_y);
 
     if (x.scalar_type() == torch::kBFloat16) {

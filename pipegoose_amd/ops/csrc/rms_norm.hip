@@ -137,7 +137,7 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
 
     constexpr int BLOCK = 256;
     auto stream = at::cuda::getCurrentCUDAStream();
-    int grid_y = (int)std::min<int64_t>(N, 512);
+    int grid_y = (int)std::min<int64_t>(N, 4096);  // fill the chip: 256 CUs want >>512 blocks when grid.x is 1-2
     dim3 grid_dw((H + BLOCK - 1) / BLOCK, grid_y);
     if (x.scalar_type() == torch::kBFloat16) {
         hipLaunchKernelGGL((rms_norm_bwd_dx_kernel<__hip_bfloat16, BLOCK>),
