@@ -21,6 +21,13 @@ class PipelineP2P:
         self.pc = parallel_context
         self.group = parallel_context.get_group(ParallelMode.PIPELINE)
         self._shape_cache = {}
+        # Sends ride a dedicated HIP stream: an RCCL send kernel SPINS until
+        # the peer's recv kernel is resident, so a recv queued behind an
+        # unmatched send on the same stream deadlocks (classic 1F1B cycle:
+        # rank i [send_grad -> recv_act] vs rank i-1 [send_act -> recv_grad]).
+        # A separate send stream breaks every such cycle.
+        self._send_stream = torch.cuda.Stream() \
+            if torch.cuda.is_available() else None
 
     def _device(self):
         if dist.get_backend(self.group) == "nccl":
@@ -30,7 +37,13 @@ class PipelineP2P:
     # Fixed-shape fast path: caller guarantees shape/dtype via negotiate()
     def send_activation(self, tensor: torch.Tensor, dst_global: int, tag: int = 0):
         payload = tensor.detach().contiguous().to(self._device())
-        work = dist.isend(payload, dst=dst_global, group=self.group, tag=tag)
+        if self._send_stream is not None and payload.is_cuda:
+            self._send_stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self._send_stream):
+                work = dist.isend(payload, dst=dst_global, group=self.group,
+                                  tag=tag)
+        else:
+            work = dist.isend(payload, dst=dst_global, group=self.group, tag=tag)
         return work, payload  # keep payload alive until work completes
 
     def recv_activation(self, shape, dtype, src_global: int, tag: int = 0,

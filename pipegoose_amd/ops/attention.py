@@ -28,6 +28,8 @@ def _kernel_supported(q: torch.Tensor) -> bool:
 def _d_contig(t: torch.Tensor) -> torch.Tensor:
     # kernels take arbitrary [B,H,S,D] strides as long as d is contiguous —
     # the model's transposed views qualify, so no copies happen here
+    if os.environ.get("PG_ATTN_CONTIG") == "1":
+        return t.contiguous()
     return t if t.stride(-1) == 1 else t.contiguous()
 
 
