@@ -121,7 +121,18 @@ class BloomAttention(nn.Module):
             self._bias_cache = {key: cached}
         return cached
 
-    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
+    def _alibi_bias_rect(self, q_len: int, k_len: int, device, dtype):
+        """Decode-path bias [1, H, q_len, k_len]: query rows sit at absolute
+        positions k_len-q_len .. k_len-1."""
+        qpos = torch.arange(k_len - q_len, k_len, device=device)
+        kpos = torch.arange(k_len, device=device)
+        rel = kpos[None, :] - qpos[:, None]                 # j - i_abs
+        bias = self.alibi_slopes.to(device=device, dtype=torch.float32)[
+            :, None, None] * rel[None].float()
+        bias = bias.masked_fill(rel[None] > 0, float("-inf"))
+        return bias.to(dtype)[None]
+
+    def forward(self, hidden: torch.Tensor, past_kv=None, use_cache: bool = False):
         B = hidden.size(0)
         fused = self.query_key_value(hidden)  # [B, S, local_heads * 3 * hd]
         S = fused.size(1)  # full sequence (SP mode all-gathered inside qkv)
@@ -130,11 +141,23 @@ class BloomAttention(nn.Module):
         k = fused[..., 1, :].transpose(1, 2)
         v = fused[..., 2, :].transpose(1, 2)
 
-        from pipegoose_amd.ops.attention import alibi_attention
-        out = alibi_attention(q, k, v, self.alibi_slopes, self.inv_norm,
-                              mask_fallback=self._alibi_bias)
+        if past_kv is not None:
+            k = torch.cat([past_kv[0], k], dim=2)
+            v = torch.cat([past_kv[1], v], dim=2)
+        present = (k, v) if use_cache else None
+
+        if k.size(2) != S:
+            # incremental decode: rectangular mask, torch sdpa
+            bias = self._alibi_bias_rect(S, k.size(2), q.device, q.dtype)
+            out = TF.scaled_dot_product_attention(
+                q, k, v, attn_mask=bias, scale=self.inv_norm)
+        else:
+            from pipegoose_amd.ops.attention import alibi_attention
+            out = alibi_attention(q, k, v, self.alibi_slopes, self.inv_norm,
+                                  mask_fallback=self._alibi_bias)
         out = out.transpose(1, 2).reshape(B, S, self.num_heads * self.head_dim)
-        return self.dense(out)
+        out = self.dense(out)
+        return (out, present) if use_cache else out
 
 
 class BloomMLP(nn.Module):
@@ -167,12 +190,18 @@ class BloomBlock(nn.Module):
             parallel_context=parallel_context if sp else None)
         self.mlp = BloomMLP(config, parallel_context)
 
-    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
-        attn_out = self.self_attention(self.input_layernorm(hidden))
+    def forward(self, hidden: torch.Tensor, past_kv=None,
+                use_cache: bool = False):
+        attn_out = self.self_attention(self.input_layernorm(hidden),
+                                       past_kv=past_kv, use_cache=use_cache)
+        present = None
+        if use_cache:
+            attn_out, present = attn_out
         # residual add fused into the post-attention norm's HBM pass
         normed, hidden = self.post_attention_layernorm.forward_with_residual(
             attn_out, hidden)
-        return hidden + self.mlp(normed)
+        out = hidden + self.mlp(normed)
+        return (out, present) if use_cache else out
 
     def forward_chained(self, hidden: torch.Tensor, pending):
         """Residual-chained form: the PREVIOUS block's mlp output arrives as
@@ -210,10 +239,18 @@ class BloomModel(nn.Module):
             config.hidden_size, eps=eps, sequence_parallel=sp,
             parallel_context=parallel_context if sp else None)
 
-    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+    def forward(self, input_ids: torch.Tensor, past=None,
+                use_cache: bool = False):
         hidden = self.word_embeddings_layernorm(self.word_embeddings(input_ids))
-        # residual-chained blocks: every residual add fuses into a norm's
-        # HBM pass (2 fused adds per block + the final one into ln_f)
+        if use_cache or past is not None:
+            presents = []
+            for i, block in enumerate(self.h):
+                pk = past[i] if past is not None else None
+                hidden, present = block(hidden, past_kv=pk, use_cache=True)
+                presents.append(present)
+            return self.ln_f(hidden), presents
+        # training fast path — residual-chained blocks: every residual add
+        # fuses into a norm's HBM pass (2 per block + the final one into ln_f)
         pending = None
         for block in self.h:
             hidden, pending = block.forward_chained(hidden, pending)
@@ -273,7 +310,12 @@ class BloomForCausalLM(nn.Module):
         elif isinstance(module, ParallelEmbedding):
             nn.init.normal_(module.weight, mean=0.0, std=std)
 
-    def forward(self, input_ids: torch.Tensor, labels: torch.Tensor = None):
+    def forward(self, input_ids: torch.Tensor, labels: torch.Tensor = None,
+                past=None, use_cache: bool = False):
+        if use_cache or past is not None:
+            hidden, presents = self.transformer(input_ids, past=past,
+                                                use_cache=True)
+            return self.lm_head(hidden), presents
         hidden = self.transformer(input_ids)
         logits = self.lm_head(hidden)
         if labels is None:

@@ -25,14 +25,23 @@ def generate(
     eos_token_id: Optional[int] = None,
 ) -> torch.Tensor:
     """Decode ``max_new_tokens`` tokens (greedy when temperature==0)."""
+    import inspect
     ctx = parallel_context or getattr(model, "parallel_context", None) \
         or ParallelContext.get_context()
     tp = ctx.get_world_size(ParallelMode.TENSOR) if ctx else 1
     model.eval()
     ids = input_ids
     finished = torch.zeros(ids.size(0), dtype=torch.bool, device=ids.device)
+    use_cache = ("use_cache" in inspect.signature(model.forward).parameters
+                 and not getattr(getattr(model, "config", None),
+                                 "sequence_parallel", False))
+    past = None
     for _ in range(max_new_tokens):
-        logits = model(ids)  # [B, S, V_local]
+        if use_cache:
+            step_ids = ids if past is None else ids[:, -1:]
+            logits, past = model(step_ids, past=past, use_cache=True)
+        else:
+            logits = model(ids)  # [B, S, V_local]
         last = logits[:, -1].float()  # [B, V_local]
         if tp > 1:
             vshard = last.size(-1)

@@ -81,11 +81,12 @@ __global__ void rms_norm_bwd_dx_kernel(
     }
 }
 
-// dw[j] = sum_rows dy[r][j] * xhat[r][j] — row-strided blocks + fp32 atomics
+// dw[j] = sum_rows dy[r][j] * xhat[r][j] — per-y-block partials (atomics on
+// shared columns serialize; two-stage partial+reduce is contention-free)
 template <typename T, int BLOCK>
 __global__ void rms_norm_bwd_dw_kernel(
     const T* __restrict__ dy, const T* __restrict__ x,
-    const float* __restrict__ rstd_in, float* __restrict__ dw,
+    const float* __restrict__ rstd_in, float* __restrict__ dw_part,
     int64_t N, int H) {
     const int col = blockIdx.x * BLOCK + threadIdx.x;
     if (col >= H) return;
@@ -93,7 +94,17 @@ __global__ void rms_norm_bwd_dw_kernel(
     for (int64_t r = blockIdx.y; r < N; r += gridDim.y) {
         acc += to_float(dy[r * H + col]) * to_float(x[r * H + col]) * rstd_in[r];
     }
-    atomicAdd(&dw[col], acc);
+    dw_part[(int64_t)blockIdx.y * H + col] = acc;
+}
+
+__global__ void rms_colsum_reduce_kernel(const float* __restrict__ part,
+                                         float* __restrict__ dw,
+                                         int GY, int H) {
+    const int col = blockIdx.x * blockDim.x + threadIdx.x;
+    if (col >= H) return;
+    float acc = 0.f;
+    for (int g = 0; g < GY; ++g) acc += part[(int64_t)g * H + col];
+    dw[col] = acc;
 }
 
 }  // namespace
@@ -137,8 +148,10 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
 
     constexpr int BLOCK = 256;
     auto stream = at::cuda::getCurrentCUDAStream();
-    int grid_y = (int)std::min<int64_t>(N, 4096);  // fill the chip: 256 CUs want >>512 blocks when grid.x is 1-2
+    int grid_y = (int)std::min<int64_t>(N, 256);
     dim3 grid_dw((H + BLOCK - 1) / BLOCK, grid_y);
+    auto part = torch::empty({grid_y, H}, x.options().dtype(torch::kFloat));
+    dim3 grid_red((H + BLOCK - 1) / BLOCK);
     if (x.scalar_type() == torch::kBFloat16) {
         hipLaunchKernelGGL((rms_norm_bwd_dx_kernel<__hip_bfloat16, BLOCK>),
             dim3(N), dim3(BLOCK), 0, stream,
@@ -151,7 +164,10 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
             grid_dw, dim3(BLOCK), 0, stream,
             reinterpret_cast<const __hip_bfloat16*>(dy.data_ptr()),
             reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
-            rstd.data_ptr<float>(), dw.data_ptr<float>(), N, (int)H);
+            rstd.data_ptr<float>(), part.data_ptr<float>(), N, (int)H);
+        hipLaunchKernelGGL((rms_colsum_reduce_kernel), grid_red, dim3(BLOCK),
+            0, stream, part.data_ptr<float>(), dw.data_ptr<float>(),
+            grid_y, (int)H);
     } else if (x.scalar_type() == torch::kFloat) {
         hipLaunchKernelGGL((rms_norm_bwd_dx_kernel<float, BLOCK>),
             dim3(N), dim3(BLOCK), 0, stream,
@@ -160,7 +176,10 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
         hipLaunchKernelGGL((rms_norm_bwd_dw_kernel<float, BLOCK>),
             grid_dw, dim3(BLOCK), 0, stream,
             dy.data_ptr<float>(), x.data_ptr<float>(),
-            rstd.data_ptr<float>(), dw.data_ptr<float>(), N, (int)H);
+            rstd.data_ptr<float>(), part.data_ptr<float>(), N, (int)H);
+        hipLaunchKernelGGL((rms_colsum_reduce_kernel), grid_red, dim3(BLOCK),
+            0, stream, part.data_ptr<float>(), dw.data_ptr<float>(),
+            grid_y, (int)H);
     } else {
         TORCH_CHECK(false, "rms_norm_bwd: unsupported dtype");
     }
