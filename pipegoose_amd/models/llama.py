@@ -87,7 +87,8 @@ class LlamaAttention(nn.Module):
             self._mask_cache = {key: m.to(dtype)[None, None]}
         return self._mask_cache[key]
 
-    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
+    def forward(self, hidden: torch.Tensor, past_kv=None,
+                use_cache: bool = False):
         B = hidden.size(0)
         q = self.q_proj(hidden)
         S = q.size(1)
@@ -96,15 +97,30 @@ class LlamaAttention(nn.Module):
         q = q.view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
         k = k.view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
         v = v.view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
-        q = apply_rope(q, self.rope_theta)
-        k = apply_rope(k, self.rope_theta)
+        offset = past_kv[0].size(2) if past_kv is not None else 0
+        q = apply_rope(q, self.rope_theta, pos_offset=offset)
+        k = apply_rope(k, self.rope_theta, pos_offset=offset)
+        if past_kv is not None:
+            k = torch.cat([past_kv[0], k], dim=2)
+            v = torch.cat([past_kv[1], v], dim=2)
+        present = (k, v) if use_cache else None
 
-        from pipegoose_amd.ops.attention import alibi_attention
-        out = alibi_attention(
-            q, k, v, self.zero_slopes, self.inv_norm,
-            mask_fallback=lambda s, dev, dt: self._causal_mask(s, dev, dt))
+        if k.size(2) != S:
+            # incremental decode: rectangular causal mask, torch sdpa
+            kl = k.size(2)
+            rel = torch.arange(kl, device=q.device)[None, :]                 - torch.arange(kl - S, kl, device=q.device)[:, None]
+            bias = torch.zeros(S, kl, device=q.device, dtype=q.dtype)
+            bias = bias.masked_fill(rel > 0, float("-inf"))[None, None]
+            out = TF.scaled_dot_product_attention(
+                q, k, v, attn_mask=bias, scale=self.inv_norm)
+        else:
+            from pipegoose_amd.ops.attention import alibi_attention
+            out = alibi_attention(
+                q, k, v, self.zero_slopes, self.inv_norm,
+                mask_fallback=lambda s, dev, dt: self._causal_mask(s, dev, dt))
         out = out.transpose(1, 2).reshape(B, S, self.num_heads * self.head_dim)
-        return self.o_proj(out)
+        out = self.o_proj(out)
+        return (out, present) if use_cache else out
 
 
 class LlamaMLP(nn.Module):
@@ -139,10 +155,16 @@ class LlamaBlock(nn.Module):
             sequence_parallel=sp, parallel_context=ctx)
         self.mlp = LlamaMLP(config, parallel_context)
 
-    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
-        hidden = hidden + self.self_attn(self.input_layernorm(hidden))
+    def forward(self, hidden: torch.Tensor, past_kv=None,
+                use_cache: bool = False):
+        attn_out = self.self_attn(self.input_layernorm(hidden),
+                                  past_kv=past_kv, use_cache=use_cache)
+        present = None
+        if use_cache:
+            attn_out, present = attn_out
+        hidden = hidden + attn_out
         hidden = hidden + self.mlp(self.post_attention_layernorm(hidden))
-        return hidden
+        return (hidden, present) if use_cache else hidden
 
 
 class LlamaModel(nn.Module):
@@ -160,8 +182,16 @@ class LlamaModel(nn.Module):
             sequence_parallel=config.sequence_parallel,
             parallel_context=parallel_context if config.sequence_parallel else None)
 
-    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+    def forward(self, input_ids: torch.Tensor, past=None,
+                use_cache: bool = False):
         hidden = self.embed_tokens(input_ids)
+        if use_cache or past is not None:
+            presents = []
+            for i, layer in enumerate(self.layers):
+                pk = past[i] if past is not None else None
+                hidden, present = layer(hidden, past_kv=pk, use_cache=True)
+                presents.append(present)
+            return self.norm(hidden), presents
         for layer in self.layers:
             hidden = layer(hidden)
         return self.norm(hidden)
@@ -191,7 +221,11 @@ class LlamaForCausalLM(nn.Module):
         elif isinstance(module, ParallelEmbedding):
             nn.init.normal_(module.weight, mean=0.0, std=std)
 
-    def forward(self, input_ids: torch.Tensor, labels: torch.Tensor = None):
+    def forward(self, input_ids: torch.Tensor, labels: torch.Tensor = None,
+                past=None, use_cache: bool = False):
+        if use_cache or past is not None:
+            hidden, presents = self.model(input_ids, past=past, use_cache=True)
+            return self.lm_head(hidden), presents
         hidden = self.model(input_ids)
         logits = self.lm_head(hidden)
         if labels is None:

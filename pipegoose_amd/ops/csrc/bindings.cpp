@@ -30,7 +30,8 @@ std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w,
                                         double eps);
 std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
                                         torch::Tensor w, torch::Tensor rstd);
-torch::Tensor rope_apply(torch::Tensor x, double theta_base, bool backward);
+torch::Tensor rope_apply(torch::Tensor x, double theta_base, bool backward,
+                         int64_t pos_offset);
 std::vector<torch::Tensor> router_topk(torch::Tensor logits, int64_t k);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

@@ -15,13 +15,13 @@ namespace {
 template <typename T>
 __global__ void rope_kernel(const T* __restrict__ x, T* __restrict__ y,
                             int S, int D, float theta_base, float sign,
-                            int64_t total_rows) {
+                            int64_t total_rows, int pos_offset) {
     // one wave per (b,h,s) row; lanes cover the D/2 rotation pairs
     const int64_t row = (int64_t)blockIdx.x * (blockDim.x / WAVE_SIZE)
                       + threadIdx.x / WAVE_SIZE;
     if (row >= total_rows) return;
     const int lane = threadIdx.x % WAVE_SIZE;
-    const int s = (int)(row % S);
+    const int s = (int)(row % S) + pos_offset;
     const T* xr = x + row * D;
     T* yr = y + row * D;
     const int half = D / 2;
@@ -40,7 +40,8 @@ __global__ void rope_kernel(const T* __restrict__ x, T* __restrict__ y,
 
 }  // namespace
 
-torch::Tensor rope_apply(torch::Tensor x, double theta_base, bool backward) {
+torch::Tensor rope_apply(torch::Tensor x, double theta_base, bool backward,
+                         int64_t pos_offset) {
     TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
     const int S = x.size(2), D = x.size(3);
     TORCH_CHECK(D % 2 == 0);
@@ -56,11 +57,11 @@ torch::Tensor rope_apply(torch::Tensor x, double theta_base, bool backward) {
             stream,
             reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
             reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
-            S, D, (float)theta_base, sign, rows);
+            S, D, (float)theta_base, sign, rows, (int)pos_offset);
     } else if (x.scalar_type() == torch::kFloat) {
         hipLaunchKernelGGL((rope_kernel<float>), grid, dim3(BLOCK), 0, stream,
             x.data_ptr<float>(), y.data_ptr<float>(),
-            S, D, (float)theta_base, sign, rows);
+            S, D, (float)theta_base, sign, rows, (int)pos_offset);
     } else {
         TORCH_CHECK(false, "rope_apply: unsupported dtype");
     }

@@ -129,3 +129,25 @@ def _run_generate_tp2(rank, world_size, port):
 
 def test_llama_generate_tp2_matches_gathered_argmax():
     spawn(_run_generate_tp2, world_size=2)
+
+
+def _run_llama_kv_cache(rank, world_size, port):
+    ctx = init_parallel_context(rank, world_size, port)
+    torch.manual_seed(61)
+    model = LlamaForCausalLM(_tiny(), ctx)
+    model.eval()
+    ids = torch.randint(0, 256, (2, 6))
+    with torch.no_grad():
+        logits_c, past = model(ids, use_cache=True)
+        logits_f = model(ids)
+        assert torch.allclose(logits_c, logits_f, atol=1e-5)
+        nxt = logits_f[:, -1].argmax(-1, keepdim=True)
+        step_logits, past = model(nxt, past=past, use_cache=True)
+        full = model(torch.cat([ids, nxt], dim=-1))
+        assert torch.allclose(step_logits[:, -1], full[:, -1], atol=1e-5), \
+            (step_logits[:, -1] - full[:, -1]).abs().max()
+    ctx.destroy()
+
+
+def test_llama_kv_cache_decode():
+    spawn(_run_llama_kv_cache, world_size=1)
