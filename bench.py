@@ -39,6 +39,10 @@ def parse_args():
     p.add_argument("--graph", action="store_true",
                    help="hipGraph-capture the training step (measured slower "
                         "on the GPU-bound default configs — opt-in)")
+    p.add_argument("--moe", type=int, default=0, metavar="E",
+                   help="replace MLPs with E-expert Switch MoE layers "
+                        "(BASELINE config 4: bloom-1b7 with 8 experts; "
+                        "all-to-all dispatch over the TP/EP group when tp>1)")
     p.add_argument("--device", type=str, default=None)
     return p.parse_args()
 
@@ -88,6 +92,23 @@ def main():
         cfg.sequence_parallel = True
     torch.manual_seed(1234)
     model = BloomForCausalLM(cfg, ctx)
+    moe_loss_wrap = None
+    if args.moe > 0:
+        from torch import nn
+        from pipegoose_amd.nn import ExpertParallel
+        from pipegoose_amd.nn.expert_parallel import (ExpertLoss,
+                                                      SwitchNoisePolicy,
+                                                      Top1Router)
+        h = cfg.hidden_size
+        dense_expert = nn.Sequential(
+            nn.Linear(h, 4 * h), nn.GELU(), nn.Linear(4 * h, h))
+        model = ExpertParallel(
+            model, args.moe, expert=dense_expert,
+            router=Top1Router(SwitchNoisePolicy(), args.moe, h),
+            enable_tensor_parallel=tp > 1,
+            dispatch="alltoall" if tp > 1 else "mask",
+            parallel_context=ctx).parallelize()
+        moe_loss_wrap = ExpertLoss(lambda loss: loss)
     if pp > 1:
         model = PipelineParallel(
             model, ctx, n_microbatches=args.microbatches,
@@ -125,6 +146,8 @@ def main():
             loss = model(input_ids, input_ids)
         else:
             loss = model(input_ids, labels=input_ids)
+            if moe_loss_wrap is not None:
+                loss = moe_loss_wrap(loss)  # + router aux/z losses
             loss.backward()
         optim.step()
         return loss
@@ -190,7 +213,8 @@ def main():
                 "seq_len": S,
                 "parallelism": f"tp{tp}pp{pp}dp{dp}"
                                + (f"mb{args.microbatches}" if pp > 1 else "")
-                               + ("sp" if args.sp else ""),
+                               + ("sp" if args.sp else "")
+                               + (f"moe{args.moe}" if args.moe else ""),
             },
         }))
 
