@@ -146,13 +146,18 @@ class BloomAttention(nn.Module):
             v = torch.cat([past_kv[1], v], dim=2)
         present = (k, v) if use_cache else None
 
+        from pipegoose_amd.ops.attention import (_kernel_supported,
+                                                 alibi_attention,
+                                                 alibi_attention_qkv)
         if k.size(2) != S:
             # incremental decode: rectangular mask, torch sdpa
             bias = self._alibi_bias_rect(S, k.size(2), q.device, q.dtype)
             out = TF.scaled_dot_product_attention(
                 q, k, v, attn_mask=bias, scale=self.inv_norm)
+        elif past_kv is None and not use_cache and _kernel_supported(q):
+            # training fast path: backward writes one d(fused) buffer
+            out = alibi_attention_qkv(fused, self.alibi_slopes, self.inv_norm)
         else:
-            from pipegoose_amd.ops.attention import alibi_attention
             out = alibi_attention(q, k, v, self.alibi_slopes, self.inv_norm,
                                   mask_fallback=self._alibi_bias)
         out = out.transpose(1, 2).reshape(B, S, self.num_heads * self.head_dim)

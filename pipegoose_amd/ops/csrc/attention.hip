@@ -293,7 +293,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
     // variant (1: 4w/MT1, 2: 4w/MT2, 3: 8w/MT1).
     static const int v128 = [] {
         const char* e = getenv("PG_ATTN_V128");
-        return e ? atoi(e) : 2;
+        return e ? atoi(e) : 3;  // 8-wave MT1: 181 TF vs 153 (4-wave MT2) measured
     }();
 #define LAUNCH_FWD(DV, MTV, WV)                                               \
     do {                                                                      \
@@ -689,21 +689,18 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
 
 }  // namespace
 
-std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
-                                    torch::Tensor k, torch::Tensor v,
-                                    torch::Tensor o, torch::Tensor lse,
-                                    torch::Tensor slopes, double scale) {
+void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
+                   torch::Tensor k, torch::Tensor v,
+                   torch::Tensor o, torch::Tensor lse,
+                   torch::Tensor slopes, double scale,
+                   torch::Tensor dq, torch::Tensor dk, torch::Tensor dv) {
     TORCH_CHECK(dout.is_cuda() && dout.stride(3) == 1 && q.stride(3) == 1 &&
                 k.stride(3) == 1 && v.stride(3) == 1 && o.stride(3) == 1,
                 "attn_bwd: last dim must be contiguous");
+    TORCH_CHECK(dq.stride(3) == 1 && dk.stride(3) == 1 && dv.stride(3) == 1);
+    // dk and dv must share a stride layout (one stride set feeds both writes)
+    TORCH_CHECK(dk.strides() == dv.strides());
     const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
-    // grads physically [B, S, H, D] (the layout the fused-qkv backward wants)
-    auto dq_phys = torch::empty({B, S, H, D}, q.options());
-    auto dk_phys = torch::empty({B, S, H, D}, q.options());
-    auto dv_phys = torch::empty({B, S, H, D}, q.options());
-    auto dq = dq_phys.permute({0, 2, 1, 3});
-    auto dk = dk_phys.permute({0, 2, 1, 3});
-    auto dv = dv_phys.permute({0, 2, 1, 3});
     auto delta = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
 
     const int64_t rows = (int64_t)B * H * S;
@@ -727,8 +724,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
             reinterpret_cast<const bf16*>(v.data_ptr()),                      \
             lse.data_ptr<float>(), delta.data_ptr<float>(),                   \
             slopes.data_ptr<float>(), (float)scale,                           \
-            reinterpret_cast<bf16*>(dk_phys.data_ptr()),                      \
-            reinterpret_cast<bf16*>(dv_phys.data_ptr()), B, H, S,             \
+            reinterpret_cast<bf16*>(dk.data_ptr()),                           \
+            reinterpret_cast<bf16*>(dv.data_ptr()), B, H, S,                  \
             STR3(dout), STR3(q), STR3(k), STR3(v), STR3(dk));                 \
         hipLaunchKernelGGL((attn_bwd_dq_kernel<DV, 32>),                      \
             grid_b, dim3(256), 0, stream,                                     \
@@ -738,7 +735,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
             reinterpret_cast<const bf16*>(v.data_ptr()),                      \
             lse.data_ptr<float>(), delta.data_ptr<float>(),                   \
             slopes.data_ptr<float>(), (float)scale,                           \
-            reinterpret_cast<bf16*>(dq_phys.data_ptr()), B, H, S,             \
+            reinterpret_cast<bf16*>(dq.data_ptr()), B, H, S,                  \
             STR3(dout), STR3(q), STR3(k), STR3(v), STR3(dq));                 \
     } while (0)
 
@@ -751,5 +748,20 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
 #undef LAUNCH_BWD
 #undef STR3
     HIP_CHECK_LAUNCH();
+}
+
+std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
+                                    torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor lse,
+                                    torch::Tensor slopes, double scale) {
+    const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+    // grads physically [B, S, H, D] (the layout the fused-qkv backward wants)
+    auto dq_phys = torch::empty({B, S, H, D}, q.options());
+    auto dk_phys = torch::empty({B, S, H, D}, q.options());
+    auto dv_phys = torch::empty({B, S, H, D}, q.options());
+    auto dq = dq_phys.permute({0, 2, 1, 3});
+    auto dk = dk_phys.permute({0, 2, 1, 3});
+    auto dv = dv_phys.permute({0, 2, 1, 3});
+    attn_bwd_into(dout, q, k, v, o, lse, slopes, scale, dq, dk, dv);
     return {dq, dk, dv};
 }

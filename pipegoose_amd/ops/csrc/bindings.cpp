@@ -25,6 +25,11 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
                                     torch::Tensor slopes, double scale);
+void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
+                   torch::Tensor k, torch::Tensor v,
+                   torch::Tensor o, torch::Tensor lse,
+                   torch::Tensor slopes, double scale,
+                   torch::Tensor dq, torch::Tensor dk, torch::Tensor dv);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor Bt);
 std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w,
                                         double eps);
@@ -49,6 +54,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "flash attention fwd, causal + in-kernel ALiBi (gfx950 MFMA)");
     m.def("attn_bwd", &attn_bwd,
           "flash attention bwd (two-pass, no atomics) (gfx950 MFMA)");
+    m.def("attn_bwd_into", &attn_bwd_into,
+          "flash attention bwd writing grads into caller buffers "
+          "(e.g. fused-qkv gradient slices)");
     m.def("mfma_probe", &mfma_probe,
           "16x16x32 bf16 MFMA fragment-layout probe");
     m.def("rms_norm_fwd", &rms_norm_fwd, "fused RMSNorm forward (gfx950)");

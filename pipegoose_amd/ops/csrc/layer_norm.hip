@@ -168,20 +168,29 @@ __global__ void layer_norm_bwd_dwdb_kernel(
     }
 }
 
-// stage 2: column-sum the [GY, H] partials
+// stage 2: column-sum the [GY, H] partials.  blockIdx.y splits the GY
+// range (16 partials per block) so small H still spawns enough threads;
+// each column gets GY/16 atomics — negligible contention.
 __global__ void colsum_reduce_kernel(const float* __restrict__ part_w,
                                      const float* __restrict__ part_b,
                                      float* __restrict__ dw,
                                      float* __restrict__ db, int GY, int H) {
     const int col = blockIdx.x * blockDim.x + threadIdx.x;
     if (col >= H) return;
+    const int g0 = blockIdx.y * 16;
+    const int g1 = min(g0 + 16, GY);
     float aw = 0.f, ab = 0.f;
-    for (int g = 0; g < GY; ++g) {
+    for (int g = g0; g < g1; ++g) {
         aw += part_w[(int64_t)g * H + col];
         ab += part_b[(int64_t)g * H + col];
     }
-    dw[col] = aw;
-    db[col] = ab;
+    if (gridDim.y == 1) {
+        dw[col] = aw;
+        db[col] = ab;
+    } else {
+        atomicAdd(&dw[col], aw);
+        atomicAdd(&db[col], ab);
+    }
 }
 
 // scalar fallback when H % 8 != 0
@@ -264,7 +273,7 @@ std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
         part_w = torch::empty({grid_y, H}, x.options().dtype(torch::kFloat));
         part_b = torch::empty({grid_y, H}, x.options().dtype(torch::kFloat));
     }
-    dim3 grid_red((H + BLOCK - 1) / BLOCK);
+    dim3 grid_red((H + BLOCK - 1) / BLOCK, (grid_y + 15) / 16);
 
     if (x.scalar_type() == torch::kBFloat16) {
         hipLaunchKernelGGL((layer_norm_bwd_dx_kernel<__hip_bfloat16, BLOCK>), dim3(N), dim3(BLOCK), 0, stream,

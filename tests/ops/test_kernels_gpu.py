@@ -385,3 +385,32 @@ def test_router_fused_matches_eager():
     assert torch.allclose(out_fused.weight, out_eager.weight, atol=1e-5)
     assert torch.allclose(out_fused.aux_loss, out_eager.aux_loss, rtol=1e-3)
     assert torch.allclose(out_fused.z_loss, out_eager.z_loss, rtol=1e-3)
+
+
+def test_attn_fused_qkv_matches_separate():
+    """Training fast path (attn over the fused QKV buffer, backward into one
+    d(fused)) vs the separate-tensor path."""
+    from pipegoose_amd.ops.attention import (alibi_attention,
+                                             alibi_attention_qkv)
+    torch.manual_seed(13)
+    B, S, H, D = 2, 128, 4, 64
+    fused = torch.randn(B, S, H, 3, D, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+    slopes = torch.rand(H, device="cuda") * 0.4
+    scale = D ** -0.5
+
+    o1 = alibi_attention_qkv(fused, slopes, scale)
+    g = torch.randn_like(o1)
+    o1.backward(g)
+    dfused1 = fused.grad.clone()
+
+    fused2 = fused.detach().clone().requires_grad_(True)
+    q = fused2[:, :, :, 0, :].permute(0, 2, 1, 3)
+    k = fused2[:, :, :, 1, :].permute(0, 2, 1, 3)
+    v = fused2[:, :, :, 2, :].permute(0, 2, 1, 3)
+    o2 = alibi_attention(q, k, v, slopes, scale)
+    o2.backward(g)
+
+    assert torch.allclose(o1.float(), o2.float(), atol=1e-3)
+    assert torch.allclose(dfused1.float(), fused2.grad.float(), atol=1e-3), \
+        (dfused1.float() - fused2.grad.float()).abs().max()
