@@ -89,36 +89,12 @@ class PipelineEngine:
         if dp is not None:
             dp.sync_enabled = False
 
+        from pipegoose_amd.utils.tracing import trace_range
         for task in self._actions():
             mb = task.microbatch_idx
-            if task.job_type == JobType.FORWARD:
-                if self.is_first:
-                    x = input_mbs[mb]
-                else:
-                    x = self._recv_forward(mb)
-                    x.requires_grad_(x.is_floating_point())
-                saved_in[mb] = x
-                out = self.stage(x)
-                saved_out[mb] = out
-                if not self.is_last:
-                    pending.append(self._send_forward(out, mb))
-                elif self.loss_fn is None or labels is None:
-                    outputs.append(out.detach())
-            else:  # BACKWARD
-                out = saved_out[mb]
-                if self.is_last:
-                    if self.loss_fn is not None and labels is not None:
-                        loss = self.loss_fn(out, label_mbs[mb]) / m
-                        losses.append(loss.detach())
-                        loss.backward()
-                    # inference-only: nothing to do
-                else:
-                    grad = self._recv_backward(mb, out)
-                    torch.autograd.backward(out, grad_tensors=grad)
-                x = saved_in[mb]
-                if not self.is_first and x is not None and x.grad is not None:
-                    pending.append(self._send_backward(x.grad, mb))
-                saved_in[mb] = saved_out[mb] = None  # free activations
+            with trace_range(f"pp:{task.job_type.name.lower()}:mb{mb}"):
+                self._run_task(task, mb, input_mbs, label_mbs, saved_in,
+                               saved_out, losses, outputs, pending, m)
 
         for work, _payload in pending:
             work.wait()
@@ -134,6 +110,38 @@ class PipelineEngine:
         if self.is_last:
             return torch.cat(outputs, dim=0) if outputs else None
         return None
+
+    def _run_task(self, task, mb, input_mbs, label_mbs, saved_in, saved_out,
+                  losses, outputs, pending, m):
+        has_loss = self.loss_fn is not None and label_mbs[mb] is not None
+        if task.job_type == JobType.FORWARD:
+            if self.is_first:
+                x = input_mbs[mb]
+            else:
+                x = self._recv_forward(mb)
+                x.requires_grad_(x.is_floating_point())
+            saved_in[mb] = x
+            out = self.stage(x)
+            saved_out[mb] = out
+            if not self.is_last:
+                pending.append(self._send_forward(out, mb))
+            elif not has_loss:
+                outputs.append(out.detach())
+        else:  # BACKWARD
+            out = saved_out[mb]
+            if self.is_last:
+                if has_loss:
+                    loss = self.loss_fn(out, label_mbs[mb]) / m
+                    losses.append(loss.detach())
+                    loss.backward()
+                # inference-only: nothing to do
+            else:
+                grad = self._recv_backward(mb, out)
+                torch.autograd.backward(out, grad_tensors=grad)
+            x = saved_in[mb]
+            if not self.is_first and x is not None and x.grad is not None:
+                pending.append(self._send_backward(x.grad, mb))
+            saved_in[mb] = saved_out[mb] = None  # free activations
 
     # ------------------------------------------------------------- transport
 

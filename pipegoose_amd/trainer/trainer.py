@@ -63,17 +63,21 @@ class Trainer:
 
     def train(self, batch) -> float:
         """One optimizer step on one batch; returns the loss value."""
+        from pipegoose_amd.utils.tracing import trace_range
         self.callbacks.fire("on_step_start", self)
         self.model.train()
         self.optimizer.zero_grad()
         tokens = self._count_tokens(batch)
-        loss = self.training_step(dict(batch))
-        loss.backward()
+        with trace_range("trainer:forward"):
+            loss = self.training_step(dict(batch))
+        with trace_range("trainer:backward"):
+            loss.backward()
         if self.max_grad_norm is not None:
             torch.nn.utils.clip_grad_norm_(
                 [p for p in self.model.parameters() if p.requires_grad],
                 self.max_grad_norm)
-        self.optimizer.step()
+        with trace_range("trainer:optimizer"):
+            self.optimizer.step()
         if self.lr_scheduler is not None:
             self.lr_scheduler.step()
         val = float(loss.detach().float().item())
