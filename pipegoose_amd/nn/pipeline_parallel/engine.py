@@ -67,7 +67,8 @@ class PipelineEngine:
 
     # ------------------------------------------------------------------- run
 
-    def run(self, inputs: torch.Tensor, labels: Optional[torch.Tensor] = None):
+    def run(self, inputs: torch.Tensor, labels: Optional[torch.Tensor] = None,
+            dp=None):
         """One training step over the pipeline.
 
         Returns the mean loss on the LAST stage (None elsewhere) when
@@ -85,8 +86,12 @@ class PipelineEngine:
         outputs: List[torch.Tensor] = []
         pending = []  # (work, payload) keep-alives
 
-        dp = getattr(self.stage, "_dp_wrapper", None)
+        if dp is None:
+            dp = getattr(self.stage, "_dp_wrapper", None)
         if dp is not None:
+            # one bucketed sync at the tail instead of per-microbatch
+            # re-reductions (those are numerically idempotent but cost
+            # n_microbatches x the DP wire traffic)
             dp.sync_enabled = False
 
         from pipegoose_amd.utils.tracing import trace_range

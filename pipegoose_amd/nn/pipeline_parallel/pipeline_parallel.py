@@ -25,7 +25,10 @@ class PipelineStageModule(nn.Module):
         self.engine = engine
 
     def forward(self, inputs, labels=None):
-        return self.engine.run(inputs, labels)
+        # DataParallel wraps THIS module; hand its hook-controller to the
+        # engine so grad sync is deferred to the end of the microbatch loop
+        return self.engine.run(inputs, labels,
+                               dp=getattr(self, "_dp_wrapper", None))
 
 
 class PipelineParallel(Parallel):
