@@ -136,12 +136,8 @@ void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
         }
         __syncthreads();
 
-        // causal skip: this wave's rows are all < kvrow0 -> every score is
-        // masked; staging/barriers already done, skip the dead MFMA work
-        const bool wave_active = (qrow0 + MT * 16 - 1) >= kvrow0;
 #pragma unroll
         for (int mt = 0; mt < MT; ++mt) {
-            if (!wave_active) break;
             // S = Q K^T for this row-tile (16 x 64)
             float s_tile[4][4];  // [nsub][reg]
 #pragma unroll
@@ -215,7 +211,6 @@ void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
         // O += P V  (K = 64 in 2 chunks of 32)
 #pragma unroll
         for (int mt = 0; mt < MT; ++mt) {
-            if (!wave_active) break;
 #pragma unroll
             for (int kc = 0; kc < 2; ++kc) {
                 frag_ab aP = *reinterpret_cast<const frag_ab*>(
@@ -483,12 +478,9 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
         }
         __syncthreads();
 
-        // causal skip: all q rows of this chunk < this wave's kv rows
-        const bool wave_active = (q0 + QR - 1) >= kv0;
         // build P^T and dS^T for ALL QR q columns of this chunk
 #pragma unroll
         for (int ms = 0; ms < QR / 16; ++ms) {
-            if (!wave_active) break;
             frag_cd st = frag_cd{0.f, 0.f, 0.f, 0.f};
             frag_cd dpt = frag_cd{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
@@ -517,7 +509,6 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
         // dV += P^T·dO and dK += dS^T·Q over the QR q rows
 #pragma unroll
         for (int kc = 0; kc < QR / 32; ++kc) {
-            if (!wave_active) break;
             frag_ab aPT = *reinterpret_cast<const frag_ab*>(
                 &pt_lds[wave][lcol * TSTRIDE + kc * 32 + 8 * lgrp]);
             frag_ab aDST = *reinterpret_cast<const frag_ab*>(
@@ -643,12 +634,9 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
         }
         __syncthreads();
 
-        // causal skip: every kv col of this chunk > this wave's q rows
-        const bool wave_active = kv0 <= (qrow0 + 15);
         // dS for all KVR kv cols of this chunk
 #pragma unroll
         for (int ns = 0; ns < KVR / 16; ++ns) {
-            if (!wave_active) break;
             frag_cd sacc = frag_cd{0.f, 0.f, 0.f, 0.f};
             frag_cd dpacc = frag_cd{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
@@ -677,7 +665,6 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
         // dQ += dS·K over the KVR kv rows
 #pragma unroll
         for (int kc = 0; kc < KVR / 32; ++kc) {
-            if (!wave_active) break;
             frag_ab aDS = *reinterpret_cast<const frag_ab*>(
                 &ds_lds[wave][lcol * TSTRIDE + kc * 32 + 8 * lgrp]);
 #pragma unroll
