@@ -37,7 +37,7 @@ class _AlibiFlashAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, slopes, scale):
         ext = get_extension(required=True)
-        o, lse = ext.attn_fwd(q, k, v, slopes, scale)
+        o, lse = ext.attn_fwd(q, k, v, slopes, scale, 0)
         ctx.save_for_backward(q, k, v, o, lse, slopes)
         ctx.scale = scale
         return o
@@ -47,7 +47,7 @@ class _AlibiFlashAttention(torch.autograd.Function):
         ext = get_extension(required=True)
         q, k, v, o, lse, slopes = ctx.saved_tensors
         dq, dk, dv = ext.attn_bwd(_d_contig(do), q, k, v, o, lse, slopes,
-                                  ctx.scale)
+                                  ctx.scale, 0)
         return dq, dk, dv, None, None
 
 
@@ -64,7 +64,7 @@ class _AlibiFlashAttentionFused(torch.autograd.Function):
         q = fused[:, :, :, 0, :].permute(0, 2, 1, 3)
         k = fused[:, :, :, 1, :].permute(0, 2, 1, 3)
         v = fused[:, :, :, 2, :].permute(0, 2, 1, 3)
-        o, lse = ext.attn_fwd(q, k, v, slopes, scale)
+        o, lse = ext.attn_fwd(q, k, v, slopes, scale, 0)
         ctx.save_for_backward(fused, o, lse, slopes)
         ctx.scale = scale
         return o
@@ -80,7 +80,7 @@ class _AlibiFlashAttentionFused(torch.autograd.Function):
         ext.attn_bwd_into(_d_contig(do), q, k, v, o, lse, slopes, ctx.scale,
                           dfused[:, :, :, 0, :].permute(0, 2, 1, 3),
                           dfused[:, :, :, 1, :].permute(0, 2, 1, 3),
-                          dfused[:, :, :, 2, :].permute(0, 2, 1, 3))
+                          dfused[:, :, :, 2, :].permute(0, 2, 1, 3), 0)
         return dfused, None, None
 
 

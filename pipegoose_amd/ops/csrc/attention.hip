@@ -41,7 +41,7 @@ void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
                      const bf16* __restrict__ v,
                      const float* __restrict__ slopes, float scale,
                      bf16* __restrict__ o, float* __restrict__ lse,
-                     int B, int H, int S,
+                     int B, int H, int S, int kv_off,
                      int64_t qb, int64_t qh, int64_t qs,
                      int64_t kb, int64_t kh, int64_t ks,
                      int64_t vb, int64_t vh, int64_t vs,
@@ -102,7 +102,11 @@ void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
         for (int r = 0; r < 4; ++r) { m_run[mt][r] = NEG_INF; l_run[mt][r] = 0.f; }
     }
 
-    const int n_kv_blocks = (qblock * ROWS_PER_WG + ROWS_PER_WG) / BLOCK_N;  // causal
+    // kv positions are globally shifted by kv_off (ring attention blocks):
+    // kv_off <= -S means every kv position is visible (no causal cut)
+    int n_kv_blocks = (kv_off <= -S) ? (S / BLOCK_N)
+        : (qblock * ROWS_PER_WG + ROWS_PER_WG - kv_off + BLOCK_N - 1) / BLOCK_N;
+    n_kv_blocks = min(max(n_kv_blocks, 0), S / BLOCK_N);
     for (int nb = 0; nb < n_kv_blocks; ++nb) {
         const int kvrow0 = nb * BLOCK_N;
         __syncthreads();
@@ -154,7 +158,7 @@ void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     const int iq = qrow0 + mt * 16 + 4 * lgrp + r;
-                    const int jk = kvrow0 + ns * 16 + lcol;
+                    const int jk = kvrow0 + ns * 16 + lcol + kv_off;
                     float sv = acc[r] * scale + slope * (float)(jk - iq);
                     s_tile[ns][r] = (jk <= iq) ? sv : NEG_INF;
                 }
@@ -270,7 +274,7 @@ __global__ void mfma_probe_kernel(const bf16* __restrict__ A,
 
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor slopes,
-                                    double scale) {
+                                    double scale, int64_t kv_off) {
     TORCH_CHECK(q.is_cuda() && q.stride(3) == 1 && k.stride(3) == 1 &&
                 v.stride(3) == 1, "attn_fwd: last dim must be contiguous");
     TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "attn_fwd: bf16 only");
@@ -305,7 +309,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
             reinterpret_cast<const bf16*>(v.data_ptr()),                      \
             slopes.data_ptr<float>(), (float)scale,                           \
             reinterpret_cast<bf16*>(o_phys.data_ptr()), lse.data_ptr<float>(),\
-            B, H, S,                                                          \
+            B, H, S, (int)kv_off,                                             \
             q.stride(0), q.stride(1), q.stride(2),                            \
             k.stride(0), k.stride(1), k.stride(2),                            \
             v.stride(0), v.stride(1), v.stride(2),                            \
@@ -389,7 +393,7 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
                           const float* __restrict__ delta,
                           const float* __restrict__ slopes, float scale,
                           bf16* __restrict__ dk, bf16* __restrict__ dv,
-                          int B, int H, int S,
+                          int B, int H, int S, int kv_off,
                           int64_t gb, int64_t gh, int64_t gs,
                           int64_t qb2, int64_t qh2, int64_t qs2,
                           int64_t kb2, int64_t kh2, int64_t ks2,
@@ -443,7 +447,8 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
         accDV[s] = frag_cd{0.f, 0.f, 0.f, 0.f};
     }
 
-    for (int q0 = nb * BLOCK_N; q0 < S; q0 += QR) {
+    const int q_start = (kv_off <= -S) ? 0 : nb * BLOCK_N + kv_off;
+    for (int q0 = q_start < 0 ? 0 : q_start; q0 < S; q0 += QR) {
         __syncthreads();
         {   // stage Q and dO, row-major + transposed, plus lse/delta
             constexpr int PACKETS = (QR / 2) * (D / 8);
@@ -494,7 +499,7 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
             }
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                const int n_glob = kv0 + 4 * lgrp + r;
+                const int n_glob = kv0 + 4 * lgrp + r + kv_off;
                 const int m_loc = ms * 16 + lcol;
                 const int m_glob = q0 + m_loc;
                 float z = st[r] * scale + slope * (float)(n_glob - m_glob);
@@ -550,7 +555,7 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
                         const float* __restrict__ delta,
                         const float* __restrict__ slopes, float scale,
                         bf16* __restrict__ dq,
-                        int B, int H, int S,
+                        int B, int H, int S, int kv_off,
                         int64_t gb, int64_t gh, int64_t gs,
                         int64_t qb2, int64_t qh2, int64_t qs2,
                         int64_t kb2, int64_t kh2, int64_t ks2,
@@ -604,7 +609,8 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
 #pragma unroll
     for (int s = 0; s < DSUB; ++s) accDQ[s] = frag_cd{0.f, 0.f, 0.f, 0.f};
 
-    const int kv_end = qb * BLOCK_M + BLOCK_M;  // causal upper bound
+    const int kv_end = (kv_off <= -S) ? S
+        : min((int64_t)S, (int64_t)qb * BLOCK_M + BLOCK_M - kv_off);  // causal bound
     for (int kv0 = 0; kv0 < kv_end; kv0 += KVR) {
         __syncthreads();
         {   // stage K (row-major + transposed) and V (row-major)
@@ -651,7 +657,7 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int iq = qrow0 + 4 * lgrp + r;
-                const int jk = kv0 + ns * 16 + lcol;
+                const int jk = kv0 + ns * 16 + lcol + kv_off;
                 float z = sacc[r] * scale + slope * (float)(jk - iq);
                 float p = (jk <= iq) ? __expf(z - lse_r[r]) : 0.f;
                 float ds = p * (dpacc[r] - delta_r[r]) * scale;
@@ -693,7 +699,8 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
                    torch::Tensor k, torch::Tensor v,
                    torch::Tensor o, torch::Tensor lse,
                    torch::Tensor slopes, double scale,
-                   torch::Tensor dq, torch::Tensor dk, torch::Tensor dv) {
+                   torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
+                   int64_t kv_off) {
     TORCH_CHECK(dout.is_cuda() && dout.stride(3) == 1 && q.stride(3) == 1 &&
                 k.stride(3) == 1 && v.stride(3) == 1 && o.stride(3) == 1,
                 "attn_bwd: last dim must be contiguous");
@@ -725,7 +732,7 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
             lse.data_ptr<float>(), delta.data_ptr<float>(),                   \
             slopes.data_ptr<float>(), (float)scale,                           \
             reinterpret_cast<bf16*>(dk.data_ptr()),                           \
-            reinterpret_cast<bf16*>(dv.data_ptr()), B, H, S,                  \
+            reinterpret_cast<bf16*>(dv.data_ptr()), B, H, S, (int)kv_off,     \
             STR3(dout), STR3(q), STR3(k), STR3(v), STR3(dk));                 \
         hipLaunchKernelGGL((attn_bwd_dq_kernel<DV, 32>),                      \
             grid_b, dim3(256), 0, stream,                                     \
@@ -735,7 +742,7 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
             reinterpret_cast<const bf16*>(v.data_ptr()),                      \
             lse.data_ptr<float>(), delta.data_ptr<float>(),                   \
             slopes.data_ptr<float>(), (float)scale,                           \
-            reinterpret_cast<bf16*>(dq.data_ptr()), B, H, S,                  \
+            reinterpret_cast<bf16*>(dq.data_ptr()), B, H, S, (int)kv_off,    \
             STR3(dout), STR3(q), STR3(k), STR3(v), STR3(dq));                 \
     } while (0)
 
@@ -753,7 +760,8 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
-                                    torch::Tensor slopes, double scale) {
+                                    torch::Tensor slopes, double scale,
+                                    int64_t kv_off) {
     const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
     // grads physically [B, S, H, D] (the layout the fused-qkv backward wants)
     auto dq_phys = torch::empty({B, S, H, D}, q.options());
@@ -762,6 +770,6 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
     auto dq = dq_phys.permute({0, 2, 1, 3});
     auto dk = dk_phys.permute({0, 2, 1, 3});
     auto dv = dv_phys.permute({0, 2, 1, 3});
-    attn_bwd_into(dout, q, k, v, o, lse, slopes, scale, dq, dk, dv);
+    attn_bwd_into(dout, q, k, v, o, lse, slopes, scale, dq, dk, dv, kv_off);
     return {dq, dk, dv};
 }

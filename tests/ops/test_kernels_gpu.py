@@ -262,7 +262,7 @@ def test_attn_fwd_kernel(shape):
     v = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
     slopes = torch.rand(H, device="cuda") * 0.5
     scale = 1.0 / D ** 0.5
-    o, lse = ext.attn_fwd(q, k, v, slopes, scale)
+    o, lse = ext.attn_fwd(q, k, v, slopes, scale, 0)
     ref = _attn_oracle(q, k, v, slopes, scale)
     err = (o.float() - ref).abs().max()
     assert err < 3e-2, err
@@ -282,8 +282,8 @@ def test_attn_bwd_kernel(shape):
     scale = 1.0 / D ** 0.5
     do = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
 
-    o, lse = ext.attn_fwd(q, k, v, slopes, scale)
-    dq, dk, dv = ext.attn_bwd(do, q, k, v, o, lse, slopes, scale)
+    o, lse = ext.attn_fwd(q, k, v, slopes, scale, 0)
+    dq, dk, dv = ext.attn_bwd(do, q, k, v, o, lse, slopes, scale, 0)
 
     ref = _attn_oracle(q, k, v, slopes, scale)
     ref.backward(do.float())
@@ -414,3 +414,37 @@ def test_attn_fused_qkv_matches_separate():
     assert torch.allclose(o1.float(), o2.float(), atol=1e-3)
     assert torch.allclose(dfused1.float(), fused2.grad.float(), atol=1e-3), \
         (dfused1.float() - fused2.grad.float()).abs().max()
+
+
+@pytest.mark.parametrize("kv_off", [0, -128, -256])
+def test_attn_kernel_kv_offset(kv_off):
+    """Ring-attention block primitive: kernel with shifted kv positions vs
+    the fp32 torch oracle (fwd + all grads)."""
+    from pipegoose_amd.nn.ring_attention import (_BlockAttn,
+                                                 _block_attention_ref)
+    torch.manual_seed(14)
+    B, H, S, D = 2, 3, 128, 64
+    q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn_like(q, requires_grad=True)
+    v = torch.randn_like(q, requires_grad=True)
+    slopes = (torch.rand(H, device="cuda") * 0.3).float()
+    scale = D ** -0.5
+
+    o, lse = _BlockAttn.apply(q, k, v, slopes, scale, kv_off)
+    g = torch.randn_like(o)
+    o.backward(g)
+
+    q2 = q.detach().float().requires_grad_(True)
+    k2 = k.detach().float().requires_grad_(True)
+    v2 = v.detach().float().requires_grad_(True)
+    o_ref, lse_ref = _block_attention_ref(q2, k2, v2, slopes, scale, kv_off)
+    o_ref.backward(g)
+
+    assert torch.allclose(o, o_ref, atol=3e-2), (o - o_ref).abs().max()
+    assert torch.allclose(lse, lse_ref, atol=1e-2)
+    for got, want, name in ((q.grad, q2.grad, "dq"), (k.grad, k2.grad, "dk"),
+                            (v.grad, v2.grad, "dv")):
+        err = (got.float() - want).abs().max()
+        ref_scale = want.abs().max().clamp_min(1.0)
+        assert err / ref_scale < 5e-2, f"{name}: {err}"

@@ -34,9 +34,9 @@ def run_shape(B, H, S, D):
     slopes = (torch.rand(H, device="cuda") * 0.5).float()
     scale = D ** -0.5
 
-    t_fwd = bench(lambda: ext.attn_fwd(q, k, v, slopes, scale))
-    o, lse = ext.attn_fwd(q, k, v, slopes, scale)
-    t_bwd = bench(lambda: ext.attn_bwd(do, q, k, v, o, lse, slopes, scale))
+    t_fwd = bench(lambda: ext.attn_fwd(q, k, v, slopes, scale, 0))
+    o, lse = ext.attn_fwd(q, k, v, slopes, scale, 0)
+    t_bwd = bench(lambda: ext.attn_bwd(do, q, k, v, o, lse, slopes, scale, 0))
 
     # sdpa-with-mask fallback (what the model used before the kernel)
     pos = torch.arange(S, device="cuda")
