@@ -381,11 +381,12 @@ void attn_bwd_delta_kernel(const bf16* __restrict__ dout,
 }
 
 // pass A: dK(n,:) = scale * sum_m dS(m,n) Q(m,:);  dV(n,:) = sum_m P(m,n) dO(m,:)
-// QR = q rows staged per iteration: 32 at D=128 keeps LDS under 1/3 of the
-// 160 KB CU budget (2-3 blocks/CU — at QR=64 the kernel was LDS-capped to
-// ONE wave/SIMD and ran with zero latency hiding).
-template <int D, int QR>
-__global__ __launch_bounds__(256)
+// QR = q rows staged per iteration: 32 keeps LDS under 1/3 of the 160 KB CU
+// budget (at QR=64 the kernel was LDS-capped to ONE wave/SIMD).
+// W = waves per workgroup: 8 waves own 128 kv rows and SHARE each staged
+// q-chunk, halving the Q/dO staging traffic vs 4-wave blocks.
+template <int D, int QR, int W>
+__global__ __launch_bounds__(W * WAVE_SIZE)
 void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
                           const bf16* __restrict__ q, const bf16* __restrict__ k,
                           const bf16* __restrict__ v,
@@ -403,13 +404,15 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
     constexpr int DSUB = D / 16;
     constexpr int RSTRIDE = D + PAD;   // row-major tiles [QR][D+PAD]
     constexpr int TSTRIDE = QR + PAD;  // transposed tiles [D][QR+PAD]
+    constexpr int NT = W * WAVE_SIZE;
+    constexpr int KVROWS = W * 16;     // kv rows per workgroup
 
     __shared__ bf16 do_lds[QR * RSTRIDE];
     __shared__ bf16 q_lds[QR * RSTRIDE];
     __shared__ bf16 qt_lds[D * TSTRIDE];
     __shared__ bf16 dot_lds[D * TSTRIDE];
-    __shared__ bf16 pt_lds[NWAVES][16 * TSTRIDE];   // P^T  (n rows, m cols)
-    __shared__ bf16 dst_lds[NWAVES][16 * TSTRIDE];  // dS^T (n rows, m cols)
+    __shared__ bf16 pt_lds[W][16 * TSTRIDE];   // P^T  (n rows, m cols)
+    __shared__ bf16 dst_lds[W][16 * TSTRIDE];  // dS^T (n rows, m cols)
     __shared__ float lse_lds[QR];
     __shared__ float delta_lds[QR];
 
@@ -426,7 +429,7 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
     const bf16* dop = dout + b * gb + h * gh;
     const float slope = slopes[h];
 
-    const int kv0 = nb * BLOCK_N + wave * 16;  // this wave's 16 kv rows
+    const int kv0 = nb * KVROWS + wave * 16;  // this wave's 16 kv rows
 
     // this wave's K and V rows live in registers (A-fragments over d)
     frag_ab aK[DCH], aV[DCH];
@@ -447,12 +450,12 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
         accDV[s] = frag_cd{0.f, 0.f, 0.f, 0.f};
     }
 
-    const int q_start = (kv_off <= -S) ? 0 : nb * BLOCK_N + kv_off;
+    const int q_start = (kv_off <= -S) ? 0 : nb * KVROWS + kv_off;
     for (int q0 = q_start < 0 ? 0 : q_start; q0 < S; q0 += QR) {
         __syncthreads();
         {   // stage Q and dO, row-major + transposed, plus lse/delta
             constexpr int PACKETS = (QR / 2) * (D / 8);
-            for (int p = tid; p < PACKETS; p += 256) {
+            for (int p = tid; p < PACKETS; p += NT) {
                 const int row = (p / (D / 8)) * 2;
                 const int col = (p % (D / 8)) * 8;
                 frag_ab pq0 = *reinterpret_cast<const frag_ab*>(
@@ -476,7 +479,7 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
                     *reinterpret_cast<uint32_t*>(&dot_lds[(col + j) * TSTRIDE + row]) = b2.u;
                 }
             }
-            for (int i = tid; i < QR; i += 256) {
+            for (int i = tid; i < QR; i += NT) {
                 lse_lds[i] = lse[bh_off + q0 + i];
                 delta_lds[i] = delta[bh_off + q0 + i];
             }
@@ -712,19 +715,12 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
 
     const int64_t rows = (int64_t)B * H * S;
     auto stream = at::cuda::getCurrentCUDAStream();
-    dim3 grid_a(S / BLOCK_N, H, B);
+    const int dkdv_waves = (S % 128 == 0) ? 8 : 4;
+    dim3 grid_a(S / (16 * dkdv_waves), H, B);
     dim3 grid_b(S / BLOCK_M, H, B);
 
 #define STR3(t) t.stride(0), t.stride(1), t.stride(2)
-#define LAUNCH_BWD(DV)                                                        \
-    do {                                                                      \
-        hipLaunchKernelGGL((attn_bwd_delta_kernel<DV>),                       \
-            dim3((rows + 3) / 4), dim3(256), 0, stream,                       \
-            reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
-            reinterpret_cast<const bf16*>(o.data_ptr()),                      \
-            delta.data_ptr<float>(), rows, H, S, STR3(dout), STR3(o));        \
-        hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DV, 32>),                    \
-            grid_a, dim3(256), 0, stream,                                     \
+#define BWD_DKDV_ARGS                                                         \
             reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
             reinterpret_cast<const bf16*>(q.data_ptr()),                      \
             reinterpret_cast<const bf16*>(k.data_ptr()),                      \
@@ -733,7 +729,21 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
             slopes.data_ptr<float>(), (float)scale,                           \
             reinterpret_cast<bf16*>(dk.data_ptr()),                           \
             reinterpret_cast<bf16*>(dv.data_ptr()), B, H, S, (int)kv_off,     \
-            STR3(dout), STR3(q), STR3(k), STR3(v), STR3(dk));                 \
+            STR3(dout), STR3(q), STR3(k), STR3(v), STR3(dk)
+#define LAUNCH_BWD(DV)                                                        \
+    do {                                                                      \
+        hipLaunchKernelGGL((attn_bwd_delta_kernel<DV>),                       \
+            dim3((rows + 3) / 4), dim3(256), 0, stream,                       \
+            reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
+            reinterpret_cast<const bf16*>(o.data_ptr()),                      \
+            delta.data_ptr<float>(), rows, H, S, STR3(dout), STR3(o));        \
+        if (dkdv_waves == 8) {                                                \
+            hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DV, 32, 8>),             \
+                grid_a, dim3(512), 0, stream, BWD_DKDV_ARGS);                 \
+        } else {                                                              \
+            hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DV, 32, 4>),             \
+                grid_a, dim3(256), 0, stream, BWD_DKDV_ARGS);                 \
+        }                                                                     \
         hipLaunchKernelGGL((attn_bwd_dq_kernel<DV, 32>),                      \
             grid_b, dim3(256), 0, stream,                                     \
             reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
@@ -742,7 +752,7 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
             reinterpret_cast<const bf16*>(v.data_ptr()),                      \
             lse.data_ptr<float>(), delta.data_ptr<float>(),                   \
             slopes.data_ptr<float>(), (float)scale,                           \
-            reinterpret_cast<bf16*>(dq.data_ptr()), B, H, S, (int)kv_off,    \
+            reinterpret_cast<bf16*>(dq.data_ptr()), B, H, S, (int)kv_off,     \
             STR3(dout), STR3(q), STR3(k), STR3(v), STR3(dq));                 \
     } while (0)
 
@@ -753,6 +763,7 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
         LAUNCH_BWD(128);
     }
 #undef LAUNCH_BWD
+#undef BWD_DKDV_ARGS
 #undef STR3
     HIP_CHECK_LAUNCH();
 }
