@@ -20,16 +20,18 @@ torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor targets,
                                 int64_t vocab_end);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor slopes,
-                                    double scale);
+                                    double scale, int64_t kv_off);
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
-                                    torch::Tensor slopes, double scale);
+                                    torch::Tensor slopes, double scale,
+                                    int64_t kv_off);
 void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
                    torch::Tensor k, torch::Tensor v,
                    torch::Tensor o, torch::Tensor lse,
                    torch::Tensor slopes, double scale,
-                   torch::Tensor dq, torch::Tensor dk, torch::Tensor dv);
+                   torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
+                   int64_t kv_off);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor Bt);
 std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w,
                                         double eps);
