@@ -257,8 +257,14 @@ class BloomModel(nn.Module):
         # training fast path — residual-chained blocks: every residual add
         # fuses into a norm's HBM pass (2 per block + the final one into ln_f)
         pending = None
-        for block in self.h:
-            hidden, pending = block.forward_chained(hidden, pending)
+        if getattr(self, "gradient_checkpointing", False) and self.training:
+            import torch.utils.checkpoint as ckpt
+            for block in self.h:
+                hidden, pending = ckpt.checkpoint(
+                    block.forward_chained, hidden, pending, use_reentrant=False)
+        else:
+            for block in self.h:
+                hidden, pending = block.forward_chained(hidden, pending)
         normed, _ = self.ln_f.forward_with_residual(pending, hidden) \
             if pending is not None else (self.ln_f(hidden), hidden)
         return normed
@@ -339,6 +345,13 @@ class BloomForCausalLM(nn.Module):
                 shift_logits.float().reshape(-1, shift_logits.size(-1)),
                 shift_labels.reshape(-1))
         return loss
+
+
+    def gradient_checkpointing_enable(self, enabled: bool = True):
+        """Recompute each block in backward instead of storing activations —
+        trades ~30% step time for O(sqrt) activation memory (capability the
+        reference lacked; composes with TP/SP/DP)."""
+        self.transformer.gradient_checkpointing = enabled
 
     @torch.no_grad()
     def generate(self, input_ids, max_new_tokens: int = 20,

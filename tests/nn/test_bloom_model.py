@@ -147,3 +147,31 @@ def _run_kv_cache_decode(rank, world_size, port):
 def test_bloom_kv_cache_decode():
     from pipegoose_amd.testing.utils import spawn
     spawn(_run_kv_cache_decode, world_size=1)
+
+
+def _run_grad_ckpt(rank, world_size, port):
+    import torch
+    from pipegoose_amd.models.bloom import BloomForCausalLM, bloom_tiny
+    from pipegoose_amd.testing.utils import init_parallel_context
+    ctx = init_parallel_context(rank, world_size, port)
+    torch.manual_seed(80)
+    ref = BloomForCausalLM(bloom_tiny(), ctx)
+    torch.manual_seed(80)
+    ck = BloomForCausalLM(bloom_tiny(), ctx)
+    ck.gradient_checkpointing_enable()
+    ids = torch.randint(0, 256, (2, 12))
+    l1 = ref(ids, labels=ids)
+    l2 = ck(ids, labels=ids)
+    assert torch.allclose(l1, l2, atol=1e-6)
+    l1.backward()
+    l2.backward()
+    for (n, p1), p2 in zip(ref.named_parameters(), ck.parameters()):
+        if p1.grad is None:
+            continue
+        assert torch.allclose(p1.grad, p2.grad, atol=1e-5), n
+    ctx.destroy()
+
+
+def test_bloom_gradient_checkpointing_parity():
+    from pipegoose_amd.testing.utils import spawn
+    spawn(_run_grad_ckpt, world_size=1)
