@@ -548,9 +548,10 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
 }
 
 // pass B: dQ(m,:) = scale * sum_n dS(m,n) K(n,:)
-// KVR = kv rows staged per iteration (same LDS-occupancy reasoning as QR).
-template <int D, int KVR>
-__global__ __launch_bounds__(256)
+// KVR = kv rows staged per iteration (same LDS-occupancy reasoning as QR);
+// W waves (8 when S allows) share each staged kv chunk.
+template <int D, int KVR, int W>
+__global__ __launch_bounds__(W * WAVE_SIZE)
 void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
                         const bf16* __restrict__ q, const bf16* __restrict__ k,
                         const bf16* __restrict__ v,
@@ -568,11 +569,13 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
     constexpr int DSUB = D / 16;
     constexpr int RSTRIDE = D + PAD;
     constexpr int TSTRIDE = KVR + PAD;
+    constexpr int NT = W * WAVE_SIZE;
+    constexpr int QROWS = W * 16;          // q rows per workgroup
 
     __shared__ bf16 k_lds[KVR * RSTRIDE];   // row-major K (QK^T B-frags)
     __shared__ bf16 v_lds[KVR * RSTRIDE];   // row-major V (dP B-frags)
     __shared__ bf16 kt_lds[D * TSTRIDE];    // transposed K (dQ B-frags)
-    __shared__ bf16 ds_lds[NWAVES][16 * TSTRIDE];  // dS (m rows, n cols)
+    __shared__ bf16 ds_lds[W][16 * TSTRIDE];  // dS (m rows, n cols)
 
     const int qb = blockIdx.x;
     const int h = blockIdx.y;
@@ -587,7 +590,7 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
     const bf16* vp = v + b * vb2 + h * vh2;
     const float slope = slopes[h];
 
-    const int qrow0 = qb * BLOCK_M + wave * 16;
+    const int qrow0 = qb * QROWS + wave * 16;
 
     // Q and dO rows of this wave in registers; lse/delta per owned row
     frag_ab aQ[DCH], aDO[DCH];
@@ -613,12 +616,12 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
     for (int s = 0; s < DSUB; ++s) accDQ[s] = frag_cd{0.f, 0.f, 0.f, 0.f};
 
     const int kv_end = (kv_off <= -S) ? S
-        : min((int64_t)S, (int64_t)qb * BLOCK_M + BLOCK_M - kv_off);  // causal bound
+        : min((int64_t)S, (int64_t)qb * QROWS + QROWS - kv_off);  // causal bound
     for (int kv0 = 0; kv0 < kv_end; kv0 += KVR) {
         __syncthreads();
         {   // stage K (row-major + transposed) and V (row-major)
             constexpr int PACKETS = (KVR / 2) * (D / 8);
-            for (int p = tid; p < PACKETS; p += 256) {
+            for (int p = tid; p < PACKETS; p += NT) {
                 const int row = (p / (D / 8)) * 2;
                 const int col = (p % (D / 8)) * 8;
                 frag_ab pk0 = *reinterpret_cast<const frag_ab*>(
@@ -717,7 +720,7 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
     auto stream = at::cuda::getCurrentCUDAStream();
     const int dkdv_waves = (S % 128 == 0) ? 8 : 4;
     dim3 grid_a(S / (16 * dkdv_waves), H, B);
-    dim3 grid_b(S / BLOCK_M, H, B);
+    dim3 grid_b(S / (16 * dkdv_waves), H, B);
 
 #define STR3(t) t.stride(0), t.stride(1), t.stride(2)
 #define BWD_DKDV_ARGS                                                         \
@@ -730,6 +733,15 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
             reinterpret_cast<bf16*>(dk.data_ptr()),                           \
             reinterpret_cast<bf16*>(dv.data_ptr()), B, H, S, (int)kv_off,     \
             STR3(dout), STR3(q), STR3(k), STR3(v), STR3(dk)
+#define BWD_DQ_ARGS                                                           \
+            reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
+            reinterpret_cast<const bf16*>(q.data_ptr()),                      \
+            reinterpret_cast<const bf16*>(k.data_ptr()),                      \
+            reinterpret_cast<const bf16*>(v.data_ptr()),                      \
+            lse.data_ptr<float>(), delta.data_ptr<float>(),                   \
+            slopes.data_ptr<float>(), (float)scale,                           \
+            reinterpret_cast<bf16*>(dq.data_ptr()), B, H, S, (int)kv_off,     \
+            STR3(dout), STR3(q), STR3(k), STR3(v), STR3(dq)
 #define LAUNCH_BWD(DV)                                                        \
     do {                                                                      \
         hipLaunchKernelGGL((attn_bwd_delta_kernel<DV>),                       \
@@ -744,16 +756,13 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
             hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DV, 32, 4>),             \
                 grid_a, dim3(256), 0, stream, BWD_DKDV_ARGS);                 \
         }                                                                     \
-        hipLaunchKernelGGL((attn_bwd_dq_kernel<DV, 32>),                      \
-            grid_b, dim3(256), 0, stream,                                     \
-            reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
-            reinterpret_cast<const bf16*>(q.data_ptr()),                      \
-            reinterpret_cast<const bf16*>(k.data_ptr()),                      \
-            reinterpret_cast<const bf16*>(v.data_ptr()),                      \
-            lse.data_ptr<float>(), delta.data_ptr<float>(),                   \
-            slopes.data_ptr<float>(), (float)scale,                           \
-            reinterpret_cast<bf16*>(dq.data_ptr()), B, H, S, (int)kv_off,     \
-            STR3(dout), STR3(q), STR3(k), STR3(v), STR3(dq));                 \
+        if (dkdv_waves == 8) {                                                \
+            hipLaunchKernelGGL((attn_bwd_dq_kernel<DV, 32, 8>),               \
+                grid_b, dim3(512), 0, stream, BWD_DQ_ARGS);                   \
+        } else {                                                              \
+            hipLaunchKernelGGL((attn_bwd_dq_kernel<DV, 32, 4>),               \
+                grid_b, dim3(256), 0, stream, BWD_DQ_ARGS);                   \
+        }                                                                     \
     } while (0)
 
     if (D == 64) {
@@ -764,6 +773,7 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
     }
 #undef LAUNCH_BWD
 #undef BWD_DKDV_ARGS
+#undef BWD_DQ_ARGS
 #undef STR3
     HIP_CHECK_LAUNCH();
 }
