@@ -7,7 +7,7 @@ snapshot to GPU boxes.
 """
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -27,7 +27,7 @@ SRC = [
 setup(
     name="pipegoose_amd",
     version="0.1.0",
-    packages=["pipegoose_amd"],
+    packages=find_packages(include=["pipegoose_amd*"]),
     ext_modules=[
         CUDAExtension(
             name="pipegoose_amd.ops._C",

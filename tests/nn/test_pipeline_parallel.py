@@ -163,3 +163,24 @@ def ctx2_placeholder(ctx):
 
 def test_pp2_bloom_matches_oracle():
     spawn(run_pp_bloom, world_size=2)
+
+
+def test_1f1b_edge_schedules():
+    """1F1B tables at the boundaries: m == 1, m < stages, m == stages."""
+    from pipegoose_amd.nn.pipeline_parallel.scheduler import (JobType,
+                                                              OneFOneBScheduler)
+    for m, p in [(1, 2), (2, 4), (4, 4), (8, 2)]:
+        sched = OneFOneBScheduler(m, p)
+        for rank in range(p):
+            acts = sched.get_rank_schedule(rank)
+            fwd = [t.microbatch_idx for t in acts if t.job_type == JobType.FORWARD]
+            bwd = [t.microbatch_idx for t in acts if t.job_type == JobType.BACKWARD]
+            # every microbatch exactly once in each direction, fwd before bwd
+            assert sorted(fwd) == list(range(m)), (m, p, rank, fwd)
+            assert sorted(bwd) == list(range(m)), (m, p, rank, bwd)
+            seen_f = set()
+            for t in acts:
+                if t.job_type == JobType.FORWARD:
+                    seen_f.add(t.microbatch_idx)
+                else:
+                    assert t.microbatch_idx in seen_f, (m, p, rank)
