@@ -270,6 +270,15 @@ class BloomModel(nn.Module):
         return normed
 
 
+
+
+def _shifted_labels(labels: torch.Tensor) -> torch.Tensor:
+    """labels shifted left with a -100 pad at the end: CE over the FULL logits
+    equals shift-slice CE but skips materializing logits[:, :-1].contiguous()
+    (8.2 GB/step at BLOOM vocab) and its backward zeros+copy."""
+    pad = labels.new_full((labels.size(0), 1), -100)
+    return torch.cat([labels[:, 1:], pad], dim=1)
+
 def make_causal_lm_loss(parallel_context: ParallelContext):
     """Shift-CE loss fn over (possibly vocab-sharded) logits — for the
     pipeline engine's last stage."""
@@ -278,18 +287,16 @@ def make_causal_lm_loss(parallel_context: ParallelContext):
         if tp > 1 else None
 
     def loss_fn(logits, labels):
-        shift_logits = logits[:, :-1].contiguous()
-        shift_labels = labels[:, 1:].contiguous()
+        shift_labels = _shifted_labels(labels)
         if vp_ce is not None:
-            return vp_ce(shift_logits, shift_labels)
+            return vp_ce(logits, shift_labels)
         if logits.is_cuda:
             from pipegoose_amd.ops.cross_entropy import fused_cross_entropy
             return fused_cross_entropy(
-                shift_logits.reshape(-1, shift_logits.size(-1)),
-                shift_labels.reshape(-1))
+                logits.reshape(-1, logits.size(-1)), shift_labels.reshape(-1))
         return TF.cross_entropy(
-            shift_logits.float().reshape(-1, shift_logits.size(-1)),
-            shift_labels.reshape(-1))
+            logits.float().reshape(-1, logits.size(-1)),
+            shift_labels.reshape(-1), ignore_index=-100)
 
     return loss_fn
 
@@ -331,19 +338,17 @@ class BloomForCausalLM(nn.Module):
         logits = self.lm_head(hidden)
         if labels is None:
             return logits
-        shift_logits = logits[:, :-1].contiguous()
-        shift_labels = labels[:, 1:].contiguous()
+        shift_labels = _shifted_labels(labels)
         if self.loss_fn is not None:
-            loss = self.loss_fn(shift_logits, shift_labels)
+            loss = self.loss_fn(logits, shift_labels)
         elif logits.is_cuda:
             from pipegoose_amd.ops.cross_entropy import fused_cross_entropy
             loss = fused_cross_entropy(
-                shift_logits.reshape(-1, shift_logits.size(-1)),
-                shift_labels.reshape(-1))
+                logits.reshape(-1, logits.size(-1)), shift_labels.reshape(-1))
         else:
             loss = TF.cross_entropy(
-                shift_logits.float().reshape(-1, shift_logits.size(-1)),
-                shift_labels.reshape(-1))
+                logits.float().reshape(-1, logits.size(-1)),
+                shift_labels.reshape(-1), ignore_index=-100)
         return loss
 
 

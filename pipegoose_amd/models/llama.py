@@ -235,18 +235,17 @@ class LlamaForCausalLM(nn.Module):
         logits = self.lm_head(hidden)
         if labels is None:
             return logits
-        shift_logits = logits[:, :-1].contiguous()
-        shift_labels = labels[:, 1:].contiguous()
+        from pipegoose_amd.models.bloom import _shifted_labels
+        shift_labels = _shifted_labels(labels)
         if self.loss_fn is not None:
-            return self.loss_fn(shift_logits, shift_labels)
+            return self.loss_fn(logits, shift_labels)
         if logits.is_cuda:
             from pipegoose_amd.ops.cross_entropy import fused_cross_entropy
             return fused_cross_entropy(
-                shift_logits.reshape(-1, shift_logits.size(-1)),
-                shift_labels.reshape(-1))
+                logits.reshape(-1, logits.size(-1)), shift_labels.reshape(-1))
         return TF.cross_entropy(
-            shift_logits.float().reshape(-1, shift_logits.size(-1)),
-            shift_labels.reshape(-1))
+            logits.float().reshape(-1, logits.size(-1)),
+            shift_labels.reshape(-1), ignore_index=-100)
 
 
     def gradient_checkpointing_enable(self, enabled: bool = True):

@@ -86,12 +86,14 @@ class VocabParallelCrossEntropy(nn.Module):
                 logits.size(-1), rank)
             loss = fused_cross_entropy(logits, targets, start, end,
                                        self.parallel_context, reduction="none")
+        else:
+            loss = _VocabParallelCrossEntropy.apply(logits, targets,
+                                                    self.parallel_context)
+        valid = targets >= 0  # ignore_index support (negative labels)
+        if not bool(valid.all()):
+            loss = loss * valid
             if self.reduction == "mean":
-                return loss.mean()
-            if self.reduction == "sum":
-                return loss.sum()
-            return loss.reshape(orig_shape)
-        loss = _VocabParallelCrossEntropy.apply(logits, targets, self.parallel_context)
+                return loss.sum() / valid.sum().clamp(min=1)
         if self.reduction == "mean":
             return loss.mean()
         if self.reduction == "sum":

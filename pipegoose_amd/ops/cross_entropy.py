@@ -62,6 +62,16 @@ def fused_cross_entropy(
         vocab_end = vocab_start + logits.size(-1)
     loss = _FusedCrossEntropy.apply(logits.contiguous(), targets.contiguous(),
                                     vocab_start, vocab_end, parallel_context)
+    # ignore_index (-100 style): negative targets contribute 0 loss and 0
+    # grad (the masking is part of the autograd graph, so the backward kernel
+    # sees gscale 0 on ignored rows).  Lets callers run CE over UNSLICED
+    # logits with a padded shifted-label tensor instead of materializing
+    # logits[:, :-1].contiguous() (8.2 GB per step at bloom vocab).
+    valid = targets.reshape(-1) >= 0
+    if not bool(valid.all()):
+        loss = loss * valid
+        if reduction == "mean":
+            return loss.sum() / valid.sum().clamp(min=1)
     if reduction == "mean":
         return loss.mean()
     if reduction == "sum":
