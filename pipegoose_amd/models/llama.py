@@ -32,6 +32,7 @@ class LlamaConfig:
     intermediate_size: int = 11008
     n_layer: int = 32
     n_head: int = 32
+    n_kv_head: int = 0        # 0 = MHA; else GQA (q heads share kv heads)
     rope_theta: float = 10000.0
     rms_norm_eps: float = 1e-6
     initializer_range: float = 0.02
@@ -62,17 +63,24 @@ class LlamaAttention(nn.Module):
         super().__init__()
         tp = parallel_context.get_world_size(ParallelMode.TENSOR)
         assert config.n_head % tp == 0
+        n_kv = config.n_kv_head or config.n_head
+        assert config.n_head % n_kv == 0 and n_kv % tp == 0
         self.num_heads = config.n_head // tp
+        self.num_kv_heads = n_kv // tp
+        self.kv_group = self.num_heads // self.num_kv_heads
         self.head_dim = config.head_dim
         self.rope_theta = config.rope_theta
         self.inv_norm = 1.0 / math.sqrt(self.head_dim)
         sp = config.sequence_parallel
         h = config.hidden_size
+        kv_out = n_kv * config.head_dim
         self.q_proj = ColumnParallelLinear(h, h, bias=False, sequence_parallel=sp,
                                            parallel_context=parallel_context)
-        self.k_proj = ColumnParallelLinear(h, h, bias=False, sequence_parallel=sp,
+        self.k_proj = ColumnParallelLinear(h, kv_out, bias=False,
+                                           sequence_parallel=sp,
                                            parallel_context=parallel_context)
-        self.v_proj = ColumnParallelLinear(h, h, bias=False, sequence_parallel=sp,
+        self.v_proj = ColumnParallelLinear(h, kv_out, bias=False,
+                                           sequence_parallel=sp,
                                            parallel_context=parallel_context)
         self.o_proj = RowParallelLinear(h, h, bias=False, sequence_parallel=sp,
                                         parallel_context=parallel_context)
@@ -95,8 +103,8 @@ class LlamaAttention(nn.Module):
         k = self.k_proj(hidden)
         v = self.v_proj(hidden)
         q = q.view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
-        k = k.view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
-        v = v.view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
+        k = k.view(B, S, self.num_kv_heads, self.head_dim).transpose(1, 2)
+        v = v.view(B, S, self.num_kv_heads, self.head_dim).transpose(1, 2)
         offset = past_kv[0].size(2) if past_kv is not None else 0
         q = apply_rope(q, self.rope_theta, pos_offset=offset)
         k = apply_rope(k, self.rope_theta, pos_offset=offset)
@@ -105,16 +113,26 @@ class LlamaAttention(nn.Module):
             v = torch.cat([past_kv[1], v], dim=2)
         present = (k, v) if use_cache else None
 
-        if k.size(2) != S:
-            # incremental decode: rectangular causal mask, torch sdpa
-            kl = k.size(2)
-            rel = torch.arange(kl, device=q.device)[None, :]                 - torch.arange(kl - S, kl, device=q.device)[:, None]
+        from pipegoose_amd.ops.attention import (_kernel_supported,
+                                                 alibi_attention)
+        kernel_ok = (_kernel_supported(q) and past_kv is None
+                     and not use_cache)
+        if k.size(2) != S or (self.kv_group > 1 and not kernel_ok):
+            # decode / CPU-GQA fallback: expand kv heads, rect mask, sdpa
+            ke = k.repeat_interleave(self.kv_group, dim=1) \
+                if self.kv_group > 1 else k
+            ve = v.repeat_interleave(self.kv_group, dim=1) \
+                if self.kv_group > 1 else v
+            kl = ke.size(2)
+            rel = torch.arange(kl, device=q.device)[None, :] \
+                - torch.arange(kl - S, kl, device=q.device)[:, None]
             bias = torch.zeros(S, kl, device=q.device, dtype=q.dtype)
             bias = bias.masked_fill(rel > 0, float("-inf"))[None, None]
             out = TF.scaled_dot_product_attention(
-                q, k, v, attn_mask=bias, scale=self.inv_norm)
+                q, ke, ve, attn_mask=bias, scale=self.inv_norm)
         else:
-            from pipegoose_amd.ops.attention import alibi_attention
+            # the hand-written kernel handles GQA natively (kv head =
+            # q head / group — attention.hip)
             out = alibi_attention(
                 q, k, v, self.zero_slopes, self.inv_norm,
                 mask_fallback=lambda s, dev, dt: self._causal_mask(s, dev, dt))

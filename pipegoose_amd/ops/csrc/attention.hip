@@ -41,7 +41,7 @@ void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
                      const bf16* __restrict__ v,
                      const float* __restrict__ slopes, float scale,
                      bf16* __restrict__ o, float* __restrict__ lse,
-                     int B, int H, int S, int kv_off,
+                     int B, int H, int S, int kv_off, int kv_group,
                      int64_t qb, int64_t qh, int64_t qs,
                      int64_t kb, int64_t kh, int64_t ks,
                      int64_t vb, int64_t vh, int64_t vs,
@@ -73,9 +73,10 @@ void attn_fwd_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
     const int lcol = lane & 15;       // 0..15
 
     const int64_t bh_off = ((int64_t)b * H + h) * S;  // lse layout
+    const int hk = h / kv_group;  // GQA: q heads share kv heads
     const bf16* qp = q + b * qb + h * qh;
-    const bf16* kp = k + b * kb + h * kh;
-    const bf16* vp = v + b * vb + h * vh;
+    const bf16* kp = k + b * kb + hk * kh;
+    const bf16* vp = v + b * vb + hk * vh;
     bf16* op = o + b * ob + h * oh;
 
     // wave owns rows [qrow0, qrow0 + MT*16)
@@ -279,6 +280,9 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                 v.stride(3) == 1, "attn_fwd: last dim must be contiguous");
     TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "attn_fwd: bf16 only");
     const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+    const int Hkv = k.size(1);
+    TORCH_CHECK(H % Hkv == 0, "attn_fwd: q heads must be a multiple of kv heads");
+    const int kv_group = H / Hkv;
     TORCH_CHECK(S % BLOCK_M == 0, "attn_fwd: S must be a multiple of 64");
     TORCH_CHECK(D == 64 || D == 128, "attn_fwd: head dim 64 or 128");
     TORCH_CHECK(slopes.numel() == H && slopes.scalar_type() == torch::kFloat);
@@ -309,7 +313,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
             reinterpret_cast<const bf16*>(v.data_ptr()),                      \
             slopes.data_ptr<float>(), (float)scale,                           \
             reinterpret_cast<bf16*>(o_phys.data_ptr()), lse.data_ptr<float>(),\
-            B, H, S, (int)kv_off,                                             \
+            B, H, S, (int)kv_off, kv_group,                                   \
             q.stride(0), q.stride(1), q.stride(2),                            \
             k.stride(0), k.stride(1), k.stride(2),                            \
             v.stride(0), v.stride(1), v.stride(2),                            \
@@ -394,7 +398,7 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
                           const float* __restrict__ delta,
                           const float* __restrict__ slopes, float scale,
                           bf16* __restrict__ dk, bf16* __restrict__ dv,
-                          int B, int H, int S, int kv_off,
+                          int B, int H, int S, int kv_off, int kv_group,
                           int64_t gb, int64_t gh, int64_t gs,
                           int64_t qb2, int64_t qh2, int64_t qs2,
                           int64_t kb2, int64_t kh2, int64_t ks2,
@@ -424,11 +428,8 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
     const int lane = tid % WAVE_SIZE;
     const int lgrp = lane >> 4, lcol = lane & 15;
 
-    const int64_t bh_off = ((int64_t)b * H + h) * S;  // lse/delta layout
-    const bf16* qp = q + b * qb2 + h * qh2;
-    const bf16* dop = dout + b * gb + h * gh;
-    const float slope = slopes[h];
-
+    // blockIdx.y is the KV head; the group's q heads accumulate into the
+    // same dk/dv (GQA — looping here avoids cross-block write races)
     const int kv0 = nb * KVROWS + wave * 16;  // this wave's 16 kv rows
 
     // this wave's K and V rows live in registers (A-fragments over d)
@@ -449,6 +450,13 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
         accDK[s] = frag_cd{0.f, 0.f, 0.f, 0.f};
         accDV[s] = frag_cd{0.f, 0.f, 0.f, 0.f};
     }
+
+    for (int gq = 0; gq < kv_group; ++gq) {
+    const int hq = h * kv_group + gq;
+    const int64_t bh_off = ((int64_t)b * H + hq) * S;  // lse/delta layout
+    const bf16* qp = q + b * qb2 + hq * qh2;
+    const bf16* dop = dout + b * gb + hq * gh;
+    const float slope = slopes[hq];
 
     const int q_start = (kv_off <= -S) ? 0 : nb * KVROWS + kv_off;
     for (int q0 = q_start < 0 ? 0 : q_start; q0 < S; q0 += QR) {
@@ -533,6 +541,8 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
         }
     }
 
+    }  // gq (GQA group loop)
+
     // epilogue: C row = this wave's kv row (4*lgrp+reg), col = d
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -559,7 +569,7 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
                         const float* __restrict__ delta,
                         const float* __restrict__ slopes, float scale,
                         bf16* __restrict__ dq,
-                        int B, int H, int S, int kv_off,
+                        int B, int H, int S, int kv_off, int kv_group,
                         int64_t gb, int64_t gh, int64_t gs,
                         int64_t qb2, int64_t qh2, int64_t qs2,
                         int64_t kb2, int64_t kh2, int64_t ks2,
@@ -586,8 +596,9 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
     const int lgrp = lane >> 4, lcol = lane & 15;
 
     const int64_t bh_off = ((int64_t)b * H + h) * S;  // lse/delta layout
-    const bf16* kp = k + b * kb2 + h * kh2;
-    const bf16* vp = v + b * vb2 + h * vh2;
+    const int hk = h / kv_group;  // GQA
+    const bf16* kp = k + b * kb2 + hk * kh2;
+    const bf16* vp = v + b * vb2 + hk * vh2;
     const float slope = slopes[h];
 
     const int qrow0 = qb * QROWS + wave * 16;
@@ -714,12 +725,15 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
     // dk and dv must share a stride layout (one stride set feeds both writes)
     TORCH_CHECK(dk.strides() == dv.strides());
     const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+    const int Hkv = k.size(1);
+    TORCH_CHECK(H % Hkv == 0);
+    const int kv_group = H / Hkv;
     auto delta = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
 
     const int64_t rows = (int64_t)B * H * S;
     auto stream = at::cuda::getCurrentCUDAStream();
     const int dkdv_waves = (S % 128 == 0) ? 8 : 4;
-    dim3 grid_a(S / (16 * dkdv_waves), H, B);
+    dim3 grid_a(S / (16 * dkdv_waves), Hkv, B);
     dim3 grid_b(S / (16 * dkdv_waves), H, B);
 
 #define STR3(t) t.stride(0), t.stride(1), t.stride(2)
@@ -732,6 +746,7 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
             slopes.data_ptr<float>(), (float)scale,                           \
             reinterpret_cast<bf16*>(dk.data_ptr()),                           \
             reinterpret_cast<bf16*>(dv.data_ptr()), B, H, S, (int)kv_off,     \
+            kv_group,                                                         \
             STR3(dout), STR3(q), STR3(k), STR3(v), STR3(dk)
 #define BWD_DQ_ARGS                                                           \
             reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
@@ -741,6 +756,7 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
             lse.data_ptr<float>(), delta.data_ptr<float>(),                   \
             slopes.data_ptr<float>(), (float)scale,                           \
             reinterpret_cast<bf16*>(dq.data_ptr()), B, H, S, (int)kv_off,     \
+            kv_group,                                                         \
             STR3(dout), STR3(q), STR3(k), STR3(v), STR3(dq)
 #define LAUNCH_BWD(DV)                                                        \
     do {                                                                      \
@@ -784,10 +800,11 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor slopes, double scale,
                                     int64_t kv_off) {
     const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+    const int Hkv = k.size(1);
     // grads physically [B, S, H, D] (the layout the fused-qkv backward wants)
     auto dq_phys = torch::empty({B, S, H, D}, q.options());
-    auto dk_phys = torch::empty({B, S, H, D}, q.options());
-    auto dv_phys = torch::empty({B, S, H, D}, q.options());
+    auto dk_phys = torch::empty({B, S, Hkv, D}, q.options());
+    auto dv_phys = torch::empty({B, S, Hkv, D}, q.options());
     auto dq = dq_phys.permute({0, 2, 1, 3});
     auto dk = dk_phys.permute({0, 2, 1, 3});
     auto dv = dv_phys.permute({0, 2, 1, 3});

@@ -448,3 +448,44 @@ def test_attn_kernel_kv_offset(kv_off):
         err = (got.float() - want).abs().max()
         ref_scale = want.abs().max().clamp_min(1.0)
         assert err / ref_scale < 5e-2, f"{name}: {err}"
+
+
+def test_attn_kernel_gqa():
+    """GQA: kernel with Hkv < H vs expanded-kv fp32 oracle (fwd + grads)."""
+    ext = _ext()
+    torch.manual_seed(15)
+    B, H, Hkv, S, D = 2, 8, 2, 128, 64
+    group = H // Hkv
+    q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(B, Hkv, S, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn_like(k, requires_grad=True)
+    slopes = (torch.rand(H, device="cuda") * 0.3).float()
+    scale = D ** -0.5
+
+    o, lse = ext.attn_fwd(q, k, v, slopes, scale, 0)
+    do = torch.randn_like(o)
+    dq, dk, dv = ext.attn_bwd(do, q, k, v, o, lse, slopes, scale, 0)
+
+    q2 = q.detach().float().requires_grad_(True)
+    k2 = k.detach().float().requires_grad_(True)
+    v2 = v.detach().float().requires_grad_(True)
+    ke = k2.repeat_interleave(group, dim=1)
+    ve = v2.repeat_interleave(group, dim=1)
+    ref = _attn_oracle(q2.bfloat16(), ke.bfloat16(), ve.bfloat16(), slopes, scale)
+    # recompute oracle in fp32 with grads
+    pos = torch.arange(S, device="cuda")
+    rel = (pos[None, :] - pos[:, None]).float()
+    bias = slopes[:, None, None] * rel[None]
+    bias = bias + torch.triu(torch.full((S, S), float("-inf"), device="cuda"), 1)[None]
+    scores = (q2 @ ke.transpose(-1, -2)) * scale + bias[None]
+    o_ref = torch.softmax(scores, -1) @ ve
+    o_ref.backward(do.float())
+
+    assert (o.float() - o_ref).abs().max() < 3e-2
+    for got, want, name in ((dq, q2.grad, "dq"), (dk, k2.grad, "dk"),
+                            (dv, v2.grad, "dv")):
+        err = (got.float() - want).abs().max()
+        sc = want.abs().max().clamp_min(1.0)
+        assert err / sc < 5e-2, f"{name}: {err}"
