@@ -141,10 +141,16 @@ class BloomAttention(nn.Module):
         k = fused[..., 1, :].transpose(1, 2)
         v = fused[..., 2, :].transpose(1, 2)
 
+        present = None
         if past_kv is not None:
-            k = torch.cat([past_kv[0], k], dim=2)
-            v = torch.cat([past_kv[1], v], dim=2)
-        present = (k, v) if use_cache else None
+            if hasattr(past_kv, "append"):     # StaticKVCache layer
+                k, v = past_kv.append(k, v)
+                present = past_kv
+            else:
+                k = torch.cat([past_kv[0], k], dim=2)
+                v = torch.cat([past_kv[1], v], dim=2)
+        if use_cache and present is None:
+            present = (k, v)
 
         from pipegoose_amd.ops.attention import (_kernel_supported,
                                                  alibi_attention,
@@ -158,6 +164,7 @@ class BloomAttention(nn.Module):
             # training fast path: backward writes one d(fused) buffer
             out = alibi_attention_qkv(fused, self.alibi_slopes, self.inv_norm)
         else:
+            # cached prefill / plain path: kernel when shapes allow, else sdpa
             out = alibi_attention(q, k, v, self.alibi_slopes, self.inv_norm,
                                   mask_fallback=self._alibi_bias)
         out = out.transpose(1, 2).reshape(B, S, self.num_heads * self.head_dim)
@@ -351,6 +358,15 @@ class BloomForCausalLM(nn.Module):
                 shift_labels.reshape(-1), ignore_index=-100)
         return loss
 
+
+    def new_kv_cache(self, batch_size: int, max_len: int):
+        """Preallocated static KV cache for decode (models/kv_cache.py)."""
+        from pipegoose_amd.models.kv_cache import StaticKVCache
+        attn = self.transformer.h[0].self_attention
+        p = next(self.parameters())
+        return StaticKVCache(len(self.transformer.h), batch_size,
+                             attn.num_heads, max_len, attn.head_dim,
+                             p.dtype, p.device)
 
     def gradient_checkpointing_enable(self, enabled: bool = True):
         """Recompute each block in backward instead of storing activations —

@@ -35,11 +35,18 @@ def generate(
     use_cache = ("use_cache" in inspect.signature(model.forward).parameters
                  and not getattr(getattr(model, "config", None),
                                  "sequence_parallel", False))
-    past = None
+    # static preallocated cache when the model provides one: no per-step
+    # cat of the whole history
+    past = model.new_kv_cache(ids.size(0), ids.size(1) + max_new_tokens)         if use_cache and hasattr(model, "new_kv_cache") else None
+    static = past is not None
+    first = True
     for _ in range(max_new_tokens):
         if use_cache:
-            step_ids = ids if past is None else ids[:, -1:]
-            logits, past = model(step_ids, past=past, use_cache=True)
+            step_ids = ids if first else ids[:, -1:]
+            first = False
+            logits, presents = model(step_ids, past=past, use_cache=True)
+            if not static:
+                past = presents
         else:
             logits = model(ids)  # [B, S, V_local]
         last = logits[:, -1].float()  # [B, V_local]

@@ -105,13 +105,24 @@ class LlamaAttention(nn.Module):
         q = q.view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
         k = k.view(B, S, self.num_kv_heads, self.head_dim).transpose(1, 2)
         v = v.view(B, S, self.num_kv_heads, self.head_dim).transpose(1, 2)
-        offset = past_kv[0].size(2) if past_kv is not None else 0
+        if past_kv is None:
+            offset = 0
+        elif hasattr(past_kv, "len"):
+            offset = past_kv.len
+        else:
+            offset = past_kv[0].size(2)
         q = apply_rope(q, self.rope_theta, pos_offset=offset)
         k = apply_rope(k, self.rope_theta, pos_offset=offset)
+        present = None
         if past_kv is not None:
-            k = torch.cat([past_kv[0], k], dim=2)
-            v = torch.cat([past_kv[1], v], dim=2)
-        present = (k, v) if use_cache else None
+            if hasattr(past_kv, "append"):     # StaticKVCache layer
+                k, v = past_kv.append(k, v)
+                present = past_kv
+            else:
+                k = torch.cat([past_kv[0], k], dim=2)
+                v = torch.cat([past_kv[1], v], dim=2)
+        if use_cache and present is None:
+            present = (k, v)
 
         from pipegoose_amd.ops.attention import (_kernel_supported,
                                                  alibi_attention)
@@ -265,6 +276,15 @@ class LlamaForCausalLM(nn.Module):
             logits.float().reshape(-1, logits.size(-1)),
             shift_labels.reshape(-1), ignore_index=-100)
 
+
+    def new_kv_cache(self, batch_size: int, max_len: int):
+        """Preallocated static KV cache for decode (models/kv_cache.py)."""
+        from pipegoose_amd.models.kv_cache import StaticKVCache
+        attn = self.model.layers[0].self_attn
+        p = next(self.parameters())
+        return StaticKVCache(len(self.model.layers), batch_size,
+                             attn.num_kv_heads, max_len, attn.head_dim,
+                             p.dtype, p.device)
 
     def gradient_checkpointing_enable(self, enabled: bool = True):
         """Recompute each block in backward instead of storing activations —
