@@ -32,7 +32,10 @@ def parse_args():
     p.add_argument("--tp", type=int, default=0, help="0 = auto by world size")
     p.add_argument("--pp", type=int, default=0)
     p.add_argument("--dp", type=int, default=0)
-    p.add_argument("--microbatches", type=int, default=4, help="pipeline microbatches")
+    p.add_argument("--microbatches", type=int, default=8,
+                   help="pipeline microbatches (1F1B bubble = (pp-1)/(m+pp-1): "
+                        "m=8 at pp=2 -> 11%% vs 20%% at m=4; per-rank "
+                        "microbatch stays 2048 tokens)")
     p.add_argument("--sp", action="store_true",
                    help="Megatron-style sequence parallelism over the TP group "
                         "(BASELINE config 5)")

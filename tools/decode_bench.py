@@ -49,8 +49,9 @@ def main():
         if use_gpu:
             torch.cuda.synchronize()
 
+        past = model.new_kv_cache(B, P + N)  # static: no per-step history cat
         t0 = time.perf_counter()
-        logits, past = model(ids, use_cache=True)
+        logits, _ = model(ids, past=past, use_cache=True)
         if use_gpu:
             torch.cuda.synchronize()
         t_prefill = time.perf_counter() - t0
@@ -58,7 +59,7 @@ def main():
         nxt = logits[:, -1].argmax(-1, keepdim=True)
         t0 = time.perf_counter()
         for _ in range(N):
-            logits, past = model(nxt, past=past, use_cache=True)
+            logits, _ = model(nxt, past=past, use_cache=True)
             nxt = logits[:, -1].argmax(-1, keepdim=True)
         if use_gpu:
             torch.cuda.synchronize()
