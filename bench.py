@@ -135,7 +135,8 @@ def main():
     else:
         optim = torch.optim.AdamW(model.parameters(), foreach=True, **opt_kw)
     if dp > 1:
-        optim = DistributedOptimizer(optim, ctx)
+        # grads REDUCE to their ZeRO shard owner (half the all-reduce bytes)
+        optim = DistributedOptimizer(optim, ctx, grad_reduce="shard")
 
     B, S = args.micro_batch, args.seq_len
     # synthetic data, fixed per rank (weak scaling: per-GPU work constant)

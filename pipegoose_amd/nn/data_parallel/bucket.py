@@ -58,62 +58,81 @@ class Bucket:
 
 
 class BucketManager:
-    """Per-(mode, dtype) buckets with async flush over the mode's group."""
+    """Per-(mode, dtype, owner) buckets with async flush over the mode's group.
+
+    ``owner`` is None for plain data parallelism (all-reduce everywhere) or a
+    ZeRO shard-owner local rank (params tagged ``_zero_owner`` by
+    DistributedOptimizer(grad_reduce="shard")): those buckets are REDUCED to
+    the owner only — half the wire bytes of an all-reduce, and exactly what
+    ZeRO-1 needs since only the owner steps those params."""
 
     def __init__(self, parallel_context: ParallelContext,
                  bucket_size_mb: int = BUCKET_SIZE_MB):
         self.parallel_context = parallel_context
         self.bucket_bytes = bucket_size_mb * 1024 * 1024
-        self.buckets: Dict[Tuple[ParallelMode, torch.dtype], Bucket] = {}
-        self.pending_works: List[Tuple[object, torch.Tensor, int]] = []
+        self.buckets: Dict[Tuple[ParallelMode, torch.dtype, Optional[int]],
+                           Bucket] = {}
+        self.pending_works: List[Tuple[object, torch.Tensor, int, bool]] = []
 
-    def _get_bucket(self, mode: ParallelMode, dtype: torch.dtype,
-                    device: torch.device, min_numel: int) -> Bucket:
-        key = (mode, dtype)
-        numel = max(self.bucket_bytes // dtype.itemsize, min_numel)
+    def _get_bucket(self, key, device: torch.device, min_numel: int) -> Bucket:
+        numel = max(self.bucket_bytes // key[1].itemsize, min_numel)
         if key not in self.buckets:
-            self.buckets[key] = Bucket(numel, dtype, device)
+            self.buckets[key] = Bucket(numel, key[1], device)
         return self.buckets[key]
 
     def add_param(self, param: torch.nn.Parameter, mode: ParallelMode):
         """Queue a param's grad; flush the bucket first if it would overflow."""
         numel = param.grad.numel()
-        bucket = self._get_bucket(mode, param.grad.dtype, param.grad.device, numel)
+        owner = getattr(param, "_zero_owner", None)             if mode == ParallelMode.DATA else None
+        key = (mode, param.grad.dtype, owner)
+        bucket = self._get_bucket(key, param.grad.device, numel)
         if bucket.is_full_with(numel):
-            self.flush(mode, param.grad.dtype)
-            bucket = self._get_bucket(mode, param.grad.dtype, param.grad.device, numel)
+            self._flush_key(key)
+            bucket = self._get_bucket(key, param.grad.device, numel)
         bucket.add_grad(param)
+
+    def _flush_key(self, key):
+        mode, dt, owner = key
+        bucket = self.buckets.get(key)
+        if bucket is None or bucket.offset == 0:
+            return
+        world = self.parallel_context.get_world_size(mode)
+        my_rank = self.parallel_context.get_local_rank(mode)
+        work = None
+        if world > 1:
+            group = self.parallel_context.get_group(mode)
+            if owner is None:
+                work = dist.all_reduce(bucket.filled_view(), group=group,
+                                       async_op=True)
+            else:
+                dst = self.parallel_context.get_ranks_in_group(mode)[owner]
+                work = dist.reduce(bucket.filled_view(), dst=dst, group=group,
+                                   async_op=True)
+        needs_divide = world > 1 and (owner is None or owner == my_rank)
+        self.pending_works.append(
+            (work, bucket.filled_view(), world, needs_divide))
+        # New bucket for further grads this step: re-point to fresh storage
+        # so the in-flight buffer isn't overwritten.
+        self.buckets[key] = Bucket(bucket.size, dt, bucket.buffer.device)
 
     def flush(self, mode: Optional[ParallelMode] = None,
               dtype: Optional[torch.dtype] = None):
-        """Launch async all-reduce on matching non-empty buckets."""
-        for (m, dt), bucket in self.buckets.items():
-            if mode is not None and m != mode:
+        """Launch async reduces on matching non-empty buckets."""
+        for key in list(self.buckets.keys()):
+            if mode is not None and key[0] != mode:
                 continue
-            if dtype is not None and dt != dtype:
+            if dtype is not None and key[1] != dtype:
                 continue
-            if bucket.offset == 0:
-                continue
-            world = self.parallel_context.get_world_size(m)
-            if world > 1:
-                work = dist.all_reduce(
-                    bucket.filled_view(),
-                    group=self.parallel_context.get_group(m),
-                    async_op=True,
-                )
-            else:
-                work = None
-            self.pending_works.append((work, bucket.filled_view(), world))
-            # New bucket for further grads this step: re-point to fresh storage
-            # so the in-flight buffer isn't overwritten.
-            self.buckets[(m, dt)] = Bucket(bucket.size, dt, bucket.buffer.device)
+            self._flush_key(key)
 
     def wait_all(self):
         """Complete outstanding reduces and average (post-divide: better bf16
-        precision than the reference's pre-divide, data_parallel.py:34-43)."""
-        for work, view, world in self.pending_works:
+        precision than the reference's pre-divide, data_parallel.py:34-43).
+        Owner-routed buckets only divide on the owner (elsewhere the buffer
+        holds partial sums nobody reads)."""
+        for work, view, world, needs_divide in self.pending_works:
             if work is not None:
                 work.wait()
-            if world > 1:
+            if needs_divide:
                 view.div_(world)
         self.pending_works.clear()

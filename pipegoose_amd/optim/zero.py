@@ -113,7 +113,9 @@ class DistributedOptimizer(BaseDistributedOptimizer):
         if dev is None:
             dev = torch.device("cpu")
             sq = sq.to(dev)
-        if world > 1 and self.grad_reduce == "shard":
+        if world > 1:
+            # shards partition the params, so the shard-norms always combine
+            # to the global norm (both grad_reduce modes)
             import torch.distributed as dist
             comm = sq if dist.get_backend() != "gloo" else sq.cpu()
             dist.all_reduce(comm, group=pc.get_group(self.parallel_mode))

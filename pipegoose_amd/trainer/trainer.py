@@ -73,9 +73,13 @@ class Trainer:
         with trace_range("trainer:backward"):
             loss.backward()
         if self.max_grad_norm is not None:
-            torch.nn.utils.clip_grad_norm_(
-                [p for p in self.model.parameters() if p.requires_grad],
-                self.max_grad_norm)
+            if hasattr(self.optimizer, "clip_grad_norm_"):
+                # ZeRO shard-aware clip (valid grads may live only on owners)
+                self.optimizer.clip_grad_norm_(self.max_grad_norm)
+            else:
+                torch.nn.utils.clip_grad_norm_(
+                    [p for p in self.model.parameters() if p.requires_grad],
+                    self.max_grad_norm)
         with trace_range("trainer:optimizer"):
             self.optimizer.step()
         if self.lr_scheduler is not None:

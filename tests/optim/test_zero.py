@@ -68,3 +68,40 @@ def run_zero_step(rank, world_size, port):
 
 def test_zero1_matches_full_adam():
     spawn(run_zero_step, world_size=2)
+
+
+def _run_shard_reduce_parity(rank, world_size, port):
+    """grad_reduce="shard" training == grad_reduce="replicate" training."""
+    import torch
+    from pipegoose_amd.models.bloom import BloomForCausalLM, bloom_tiny
+    from pipegoose_amd.nn import DataParallel
+    from pipegoose_amd.optim import DistributedOptimizer
+    from pipegoose_amd.testing.utils import init_parallel_context
+    ctx = init_parallel_context(rank, world_size, port, data_parallel_size=2)
+
+    def train(mode):
+        torch.manual_seed(7)
+        model = BloomForCausalLM(bloom_tiny(), ctx)
+        model = DataParallel(model, ctx).parallelize()
+        optim = DistributedOptimizer(
+            torch.optim.Adam(model.parameters(), lr=1e-3), ctx,
+            grad_reduce=mode)
+        for step in range(3):
+            torch.manual_seed(100 + 10 * step + rank)
+            ids = torch.randint(0, 256, (2, 12))
+            optim.zero_grad()
+            model(ids, labels=ids).backward()
+            optim.clip_grad_norm_(1.0)
+            optim.step()
+        return [p.detach().clone() for p in model.parameters()]
+
+    ref = train("replicate")
+    got = train("shard")
+    for a, b in zip(ref, got):
+        assert torch.allclose(a, b, atol=1e-5), (a - b).abs().max()
+    ctx.destroy()
+
+
+def test_zero_shard_grad_reduce_matches_replicate_dp2():
+    from pipegoose_amd.testing.utils import spawn
+    spawn(_run_shard_reduce_parity, world_size=2)
