@@ -319,8 +319,14 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
             v.stride(0), v.stride(1), v.stride(2),                            \
             o.stride(0), o.stride(1), o.stride(2));                           \
     } while (0)
+    static const int v64 = [] {
+        const char* e = getenv("PG_ATTN_V64");
+        return e ? atoi(e) : 3;  // 8w/MT1: 141 TF vs 135 (4w/MT2) measured
+    }();
     if (D == 64) {
-        if (S % 128 == 0) LAUNCH_FWD(64, 2, 4); else LAUNCH_FWD(64, 1, 4);
+        if (S % 128 != 0) LAUNCH_FWD(64, 1, 4);
+        else if (v64 == 3) LAUNCH_FWD(64, 1, 8);
+        else LAUNCH_FWD(64, 2, 4);
     } else if (S % 128 != 0) {
         LAUNCH_FWD(128, 1, 4);
     } else if (v128 == 3) {
