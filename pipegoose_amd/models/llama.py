@@ -168,7 +168,10 @@ class LlamaMLP(nn.Module):
                                            parallel_context=parallel_context)
 
     def forward(self, hidden: torch.Tensor) -> torch.Tensor:
-        return self.down_proj(TF.silu(self.gate_proj(hidden)) * self.up_proj(hidden))
+        from pipegoose_amd.ops.silu_mul import silu_mul
+        # fused silu(gate)*up: one HBM pass (ops/csrc/silu_mul.hip)
+        return self.down_proj(silu_mul(self.gate_proj(hidden),
+                                       self.up_proj(hidden)))
 
 
 class LlamaBlock(nn.Module):

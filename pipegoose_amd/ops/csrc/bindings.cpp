@@ -40,6 +40,9 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
 torch::Tensor rope_apply(torch::Tensor x, double theta_base, bool backward,
                          int64_t pos_offset);
 std::vector<torch::Tensor> router_topk(torch::Tensor logits, int64_t k);
+torch::Tensor silu_mul_fwd(torch::Tensor gate, torch::Tensor up);
+std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dy, torch::Tensor gate,
+                                        torch::Tensor up);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
@@ -67,4 +70,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "rotary embedding with in-kernel cos/sin (gfx950)");
     m.def("router_topk", &router_topk,
           "fused Switch router: softmax+topk+colsum+lse in one pass (gfx950)");
+    m.def("silu_mul_fwd", &silu_mul_fwd, "fused SwiGLU silu(g)*u (gfx950)");
+    m.def("silu_mul_bwd", &silu_mul_bwd, "fused SwiGLU backward (gfx950)");
 }

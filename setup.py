@@ -22,6 +22,7 @@ SRC = [
     "pipegoose_amd/ops/csrc/rms_norm.hip",
     "pipegoose_amd/ops/csrc/rope.hip",
     "pipegoose_amd/ops/csrc/router.hip",
+    "pipegoose_amd/ops/csrc/silu_mul.hip",
 ]
 
 setup(

@@ -489,3 +489,23 @@ def test_attn_kernel_gqa():
         err = (got.float() - want).abs().max()
         sc = want.abs().max().clamp_min(1.0)
         assert err / sc < 5e-2, f"{name}: {err}"
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_silu_mul(dtype):
+    from pipegoose_amd.ops.silu_mul import silu_mul
+    torch.manual_seed(16)
+    g = torch.randn(64, 1024, device="cuda", dtype=dtype, requires_grad=True)
+    u = torch.randn_like(g, requires_grad=True)
+    y = silu_mul(g, u)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    g2 = g.detach().float().requires_grad_(True)
+    u2 = u.detach().float().requires_grad_(True)
+    (torch.nn.functional.silu(g2) * u2).backward(dy.float())
+    tol = 1e-5 if dtype == torch.float32 else 3e-2
+    assert torch.allclose(y.float(), torch.nn.functional.silu(
+        g.detach().float()) * u.detach().float(), atol=tol)
+    assert torch.allclose(g.grad.float(), g2.grad, atol=tol * 3)
+    assert torch.allclose(u.grad.float(), u2.grad, atol=tol * 3)
