@@ -7,6 +7,12 @@ Launch (single 8-GPU node):
 """
 import argparse
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import torch
 
 from pipegoose_amd import ParallelContext
