@@ -103,7 +103,6 @@ class BloomAttention(nn.Module):
         local = slopes.chunk(tp)[tp_rank].clone()
         self.register_buffer("alibi_slopes", local, persistent=False)
         self._bias_cache = {}
-        self._fold_cache_store = {}
 
     def _alibi_bias(self, seq_len: int, device, dtype) -> torch.Tensor:
         key = (seq_len, device, dtype)
