@@ -75,3 +75,41 @@ def run_dp_pp(rank, world_size, port):
 
 def test_pp2_dp2_replicas_stay_synced():
     spawn(run_dp_pp, world_size=4)
+
+
+def run_tp_pp_sp(rank, world_size, port):
+    """Config 5 shape: TP2 x PP2 with sequence parallelism — loss parity
+    with the non-SP hybrid (same seeds, same data)."""
+    import torch
+    from pipegoose_amd.distributed.parallel_mode import ParallelMode
+    from pipegoose_amd.models.bloom import (BloomConfig, BloomForCausalLM,
+                                            make_causal_lm_loss)
+    from pipegoose_amd.nn.pipeline_parallel import PipelineParallel
+    from pipegoose_amd.testing.utils import init_parallel_context
+
+    ctx = init_parallel_context(rank, world_size, port,
+                                tensor_parallel_size=2,
+                                pipeline_parallel_size=2)
+
+    def build(sp):
+        cfg = BloomConfig(vocab_size=256, hidden_size=64, n_layer=2, n_head=4,
+                          sequence_parallel=sp)
+        torch.manual_seed(12)
+        model = BloomForCausalLM(cfg, ctx)
+        return PipelineParallel(model, ctx, n_microbatches=2,
+                                loss_fn=make_causal_lm_loss(ctx)).parallelize()
+
+    torch.manual_seed(13)
+    ids = torch.randint(0, 256, (4, 16))
+
+    loss_ref = build(sp=False)(ids, ids)
+    loss_sp = build(sp=True)(ids, ids)
+    if ctx.is_last_rank(ParallelMode.PIPELINE):
+        assert loss_ref is not None and loss_sp is not None
+        assert torch.allclose(loss_ref, loss_sp, atol=1e-5), \
+            (loss_ref, loss_sp)
+    ctx.destroy()
+
+
+def test_tp2_pp2_sequence_parallel_matches():
+    spawn(run_tp_pp_sp, world_size=4)
