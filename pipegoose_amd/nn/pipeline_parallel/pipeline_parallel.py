@@ -24,7 +24,10 @@ class PipelineStageModule(nn.Module):
         self.stage = engine.stage  # registers stage params
         self.engine = engine
 
-    def forward(self, inputs, labels=None):
+    def forward(self, inputs=None, labels=None, input_ids=None):
+        # accept both positional and Trainer-style input_ids= kwargs
+        if inputs is None:
+            inputs = input_ids
         # DataParallel wraps THIS module; hand its hook-controller to the
         # engine so grad sync is deferred to the end of the microbatch loop
         return self.engine.run(inputs, labels,

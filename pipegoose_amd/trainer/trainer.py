@@ -52,6 +52,8 @@ class Trainer:
         CausalLM called with labels=)."""
         if self.loss_fn is None:
             out = self.model(**batch)
+            if out is None:
+                return None  # pipeline non-last stage (engine ran backward)
             return out if torch.is_tensor(out) else out["loss"]
         labels = batch.pop("labels")
         logits = self.model(**batch)
@@ -70,8 +72,11 @@ class Trainer:
         tokens = self._count_tokens(batch)
         with trace_range("trainer:forward"):
             loss = self.training_step(dict(batch))
-        with trace_range("trainer:backward"):
-            loss.backward()
+        if loss is not None and loss.requires_grad:
+            with trace_range("trainer:backward"):
+                loss.backward()
+        # else: a pipeline engine already ran backward internally (its
+        # returned loss is detached; non-last stages return None)
         if self.max_grad_norm is not None:
             if hasattr(self.optimizer, "clip_grad_norm_"):
                 # ZeRO shard-aware clip (valid grads may live only on owners)
@@ -84,13 +89,15 @@ class Trainer:
             self.optimizer.step()
         if self.lr_scheduler is not None:
             self.lr_scheduler.step()
-        val = float(loss.detach().float().item())
+        val = float(loss.detach().float().item()) if loss is not None \
+            else float("nan")
         self.state.global_step += 1
         self.state.tokens_seen += tokens
         self.state.last_loss = val
         self.meter.update(tokens)
         self.callbacks.fire("on_step_end", self, val)
-        if self.log_interval and self.state.global_step % self.log_interval == 0:
+        if self.log_interval and not math.isnan(val) \
+                and self.state.global_step % self.log_interval == 0:
             tps = self.meter.tokens_per_sec
             self.logger.info(
                 f"step {self.state.global_step} loss {val:.4f}"

@@ -71,3 +71,30 @@ def test_throughput_meter():
 def test_logger_rank_prefix(capsys):
     log = DistributedLogger("t_logger", parallel_context=None)
     log.info("hello")  # no context → always logs, no crash
+
+
+def _run_trainer_with_pp2(rank, world_size, port):
+    """Trainer driving a pipeline-parallel model (the examples path)."""
+    from pipegoose_amd.models.bloom import (BloomForCausalLM, bloom_tiny,
+                                            make_causal_lm_loss)
+    from pipegoose_amd.nn.pipeline_parallel import PipelineParallel
+    ctx = init_parallel_context(rank, world_size, port,
+                                pipeline_parallel_size=2)
+    torch.manual_seed(6)
+    model = BloomForCausalLM(bloom_tiny(), ctx)
+    model = PipelineParallel(model, ctx, n_microbatches=2,
+                             loss_fn=make_causal_lm_loss(ctx)).parallelize()
+    optim = torch.optim.SGD(model.parameters(), lr=0.01)
+    trainer = Trainer(model, optim, parallel_context=ctx, log_interval=0)
+
+    torch.manual_seed(7)
+    batches = [{"input_ids": torch.randint(0, 256, (4, 16)),
+                "labels": torch.randint(0, 256, (4, 16))} for _ in range(3)]
+    # PP: loss returned only on the last stage; trainer must tolerate None
+    state = trainer.fit(batches, epochs=1)
+    assert state.global_step == 3
+    ctx.destroy()
+
+
+def test_trainer_with_pipeline_parallel():
+    spawn(_run_trainer_with_pp2, world_size=2)
