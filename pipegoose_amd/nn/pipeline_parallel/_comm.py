@@ -26,8 +26,8 @@ class PipelineP2P:
         # unmatched send on the same stream deadlocks (classic 1F1B cycle:
         # rank i [send_grad -> recv_act] vs rank i-1 [send_act -> recv_grad]).
         # A separate send stream breaks every such cycle.
-        self._send_stream = torch.cuda.Stream() \
-            if torch.cuda.is_available() else None
+        self._send_streams = {}  # per-channel, created lazily
+        self._recv_streams = {}
 
     def _device(self):
         if dist.get_backend(self.group) == "nccl":
@@ -35,11 +35,15 @@ class PipelineP2P:
         return torch.device("cpu")
 
     # Fixed-shape fast path: caller guarantees shape/dtype via negotiate()
-    def send_activation(self, tensor: torch.Tensor, dst_global: int, tag: int = 0):
+    def send_activation(self, tensor: torch.Tensor, dst_global: int,
+                        tag: int = 0, channel: int = 0):
         payload = tensor.detach().contiguous().to(self._device())
-        if self._send_stream is not None and payload.is_cuda:
-            self._send_stream.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(self._send_stream):
+        stream = self._send_streams.get(channel)
+        if stream is None and torch.cuda.is_available():
+            stream = self._send_streams[channel] = torch.cuda.Stream()
+        if stream is not None and payload.is_cuda:
+            stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(stream):
                 work = dist.isend(payload, dst=dst_global, group=self.group,
                                   tag=tag)
         else:
@@ -52,3 +56,24 @@ class PipelineP2P:
         dist.recv(buf, src=src_global, group=self.group, tag=tag)
         buf.requires_grad_(requires_grad)
         return buf
+
+    def recv_activation_async(self, shape, dtype, src_global: int,
+                              tag: int = 0, channel: int = 0):
+        """Post an irecv on a dedicated per-CHANNEL recv stream (an RCCL recv
+        kernel spins until matched — posted on the compute stream it would
+        block everything queued behind it, and forward/backward prefetches
+        must not queue behind each other either: in 1F1B a grad can be
+        needed before later activations have even been sent).  Returns
+        (work, buf); consume with ``work.wait()`` from the compute stream
+        (inserts a stream dependency, not a host stall)."""
+        stream = self._recv_streams.get(channel)
+        if stream is None and torch.cuda.is_available():
+            stream = self._recv_streams[channel] = torch.cuda.Stream()
+        buf = torch.empty(shape, dtype=dtype, device=self._device())
+        if stream is not None and buf.is_cuda:
+            with torch.cuda.stream(stream):
+                work = dist.irecv(buf, src=src_global, group=self.group,
+                                  tag=tag)
+        else:
+            work = dist.irecv(buf, src=src_global, group=self.group, tag=tag)
+        return work, buf

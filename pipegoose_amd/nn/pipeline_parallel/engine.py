@@ -82,6 +82,15 @@ class PipelineEngine:
 
         saved_in: List[Optional[torch.Tensor]] = [None] * m
         saved_out: List[Optional[torch.Tensor]] = [None] * m
+        acts = self._actions()
+        # recv prefetch bookkeeping: posts must follow the consumption order
+        # (= this rank's schedule order = the peer's send order)
+        self._fwd_order = [t.microbatch_idx for t in acts
+                           if t.job_type == JobType.FORWARD]
+        self._bwd_order = [t.microbatch_idx for t in acts
+                           if t.job_type == JobType.BACKWARD]
+        self._fwd_prefetch = {}
+        self._bwd_prefetch = {}
         losses: List[torch.Tensor] = []
         outputs: List[torch.Tensor] = []
         pending = []  # (work, payload) keep-alives
@@ -95,7 +104,7 @@ class PipelineEngine:
             dp.sync_enabled = False
 
         from pipegoose_amd.utils.tracing import trace_range
-        for task in self._actions():
+        for task in acts:
             mb = task.microbatch_idx
             with trace_range(f"pp:{task.job_type.name.lower()}:mb{mb}"):
                 self._run_task(task, mb, input_mbs, label_mbs, saved_in,
@@ -154,22 +163,54 @@ class PipelineEngine:
         if mb == 0:
             self.codec.send(out, self.next_rank)
             return _NullWork(), None
-        return self.p2p.send_activation(out, self.next_rank)
+        return self.p2p.send_activation(out, self.next_rank, channel=0)
 
     def _recv_forward(self, mb):
-        if mb == 0:
+        if mb == self._fwd_order[0]:
+            # first forward: negotiate shapes, then prefetch every remaining
+            # forward recv on the recv stream IN SCHEDULE ORDER (= the
+            # sender's send order) — P2P latency overlaps compute
             t = self.codec.recv(self.prev_rank)
             self._act_shape = tuple(t.shape)
             self._act_dtype = t.dtype
+            for nxt in self._fwd_order[1:]:
+                self._fwd_prefetch[nxt] = self.p2p.recv_activation_async(
+                    self._act_shape, self._act_dtype, self.prev_rank,
+                    channel=0)
             return t
-        return self.p2p.recv_activation(self._act_shape, self._act_dtype,
-                                        self.prev_rank)
+        work, buf = self._fwd_prefetch.pop(mb)
+        work.wait()  # on CUDA: a stream dependency, not a host stall
+        if buf.is_cuda:
+            buf.record_stream(torch.cuda.current_stream())
+        buf.requires_grad_(buf.is_floating_point())
+        return buf
 
     def _send_backward(self, grad, mb):
-        return self.p2p.send_activation(grad, self.prev_rank)
+        return self.p2p.send_activation(grad, self.prev_rank, channel=1)
 
     def _recv_backward(self, mb, out):
-        return self.p2p.recv_activation(tuple(out.shape), out.dtype, self.next_rank)
+        if mb in self._bwd_prefetch:
+            work, buf = self._bwd_prefetch.pop(mb)
+            work.wait()
+            if buf.is_cuda:
+                buf.record_stream(torch.cuda.current_stream())
+            return buf
+        # first backward: post the CURRENT recv first (the peer sends it
+        # first), then prefetch the rest in schedule order
+        work, buf = self.p2p.recv_activation_async(
+            tuple(out.shape), out.dtype, self.next_rank, channel=1)
+        started = False
+        for nxt in self._bwd_order:
+            if nxt == mb:
+                started = True
+                continue
+            if started:
+                self._bwd_prefetch[nxt] = self.p2p.recv_activation_async(
+                    tuple(out.shape), out.dtype, self.next_rank, channel=1)
+        work.wait()
+        if buf.is_cuda:
+            buf.record_stream(torch.cuda.current_stream())
+        return buf
 
 
 class _NullWork:
