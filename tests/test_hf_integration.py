@@ -135,3 +135,43 @@ def _run_tp2_hf_llama(rank, world_size, port):
 
 def test_tensor_parallel_hf_llama_tp2():
     spawn(_run_tp2_hf_llama, world_size=2)
+
+
+def _hf_llama_gqa():
+    from transformers import LlamaConfig, LlamaForCausalLM
+    cfg = LlamaConfig(vocab_size=512, hidden_size=64, intermediate_size=128,
+                      num_hidden_layers=2, num_attention_heads=4,
+                      num_key_value_heads=2, attn_implementation="eager")
+    torch.manual_seed(35)
+    return LlamaForCausalLM(cfg)
+
+
+def _run_tp2_hf_llama_gqa(rank, world_size, port):
+    """Mistral/Qwen-style GQA surgery: kv projections column-split while the
+    q/kv group RATIO stays TP-invariant in HF's attention math."""
+    ctx = init_parallel_context(rank, world_size, port, tensor_parallel_size=2)
+    model = _hf_llama_gqa()
+    ref = _hf_llama_gqa()
+    torch.manual_seed(36)
+    ids = torch.randint(0, 512, (2, 10))
+    with torch.no_grad():
+        ref_logits = ref(ids).logits
+    model = TensorParallel(model, ctx).parallelize()
+    attn = model.model.layers[0].self_attn
+    assert attn.k_proj.weight.shape[0] == 1 * 16  # 1 local kv head * head_dim
+    with torch.no_grad():
+        out = model(ids).logits
+    if out.size(-1) == ref_logits.size(-1) // 2:
+        import torch.distributed as dist
+        from pipegoose_amd.distributed.parallel_mode import ParallelMode
+        shards = [torch.empty_like(out) for _ in range(2)]
+        dist.all_gather(shards, out.contiguous(),
+                        group=ctx.get_group(ParallelMode.TENSOR))
+        out = torch.cat(shards, dim=-1)
+    assert torch.allclose(out, ref_logits, atol=1e-4), \
+        (out - ref_logits).abs().max()
+    ctx.destroy()
+
+
+def test_tensor_parallel_hf_llama_gqa_tp2():
+    spawn(_run_tp2_hf_llama_gqa, world_size=2)
