@@ -5,7 +5,7 @@ layer (O(S) copies per token).  A StaticKVCache preallocates
 [B, H, max_len, D] once and index-writes the new step — no growth copies,
 stable addresses (hipGraph-friendly for a future captured decode step).
 """
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 

@@ -3,7 +3,6 @@ ExpertLayers (reference: nn/expert_parallel/expert_parallel.py:53-79)."""
 import re
 from typing import Callable, List, Optional, Union
 
-import torch
 from torch import nn
 
 from pipegoose_amd.distributed.parallel_context import ParallelContext

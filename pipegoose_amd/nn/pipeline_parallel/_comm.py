@@ -7,8 +7,6 @@ send/recv rides xGMI directly; shapes are negotiated once per engine run
 (same [B/m, S, H] every microbatch), so steady-state transfers are a single
 payload message with no metadata round trips.
 """
-from typing import Optional
-
 import torch
 import torch.distributed as dist
 

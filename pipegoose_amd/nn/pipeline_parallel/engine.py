@@ -14,7 +14,6 @@ Backward IS an explicit engine phase: saved (input, output) per microbatch,
 from typing import Callable, List, Optional
 
 import torch
-import torch.distributed as dist
 
 from pipegoose_amd.distributed.p2p import P2P
 from pipegoose_amd.distributed.parallel_context import ParallelContext

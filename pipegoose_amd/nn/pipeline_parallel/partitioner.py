@@ -12,7 +12,6 @@ per-architecture via `register_structure`.
 """
 from typing import Callable, List, Optional
 
-import torch
 from torch import nn
 
 from pipegoose_amd.distributed.parallel_context import ParallelContext

@@ -8,7 +8,6 @@ user-facing shape: wrap, parallelize, call the model).  The returned module's
 """
 from typing import Callable, Optional
 
-import torch
 from torch import nn
 
 from pipegoose_amd.distributed.parallel_context import ParallelContext

@@ -8,7 +8,6 @@ model in place.
 import torch
 from torch import nn
 
-from pipegoose_amd.distributed.parallel_context import ParallelContext
 from pipegoose_amd.distributed.parallel_mode import ParallelMode
 from pipegoose_amd.nn.parallel import Parallel
 from pipegoose_amd.nn.tensor_parallel.parallelizer import (

@@ -2,7 +2,6 @@
 credit there to fairscale OSS)."""
 from typing import Dict, List
 
-import torch
 
 from pipegoose_amd.distributed.parallel_context import ParallelContext
 from pipegoose_amd.distributed.parallel_mode import ParallelMode

@@ -7,7 +7,7 @@ optim.py:62-66), every rank launches async broadcasts of ONE flat buffer per
 owner rank, all in flight together over xGMI, then unpacks.  On RCCL these
 ride separate channels and saturate the 7 P2P links.
 """
-from typing import Dict, List
+from typing import List
 
 import torch
 import torch.distributed as dist

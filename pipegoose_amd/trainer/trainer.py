@@ -14,7 +14,7 @@ from typing import Callable, Iterable, Optional
 import torch
 
 from pipegoose_amd.distributed.parallel_context import ParallelContext
-from pipegoose_amd.trainer.callback import Callback, CallbackList
+from pipegoose_amd.trainer.callback import CallbackList
 from pipegoose_amd.trainer.logger import DistributedLogger, ThroughputMeter
 from pipegoose_amd.trainer.state import TrainerStage, TrainerState
 
