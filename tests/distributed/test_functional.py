@@ -97,3 +97,30 @@ def test_broadcast_scatter_barrier():
 
 def test_send_recv():
     spawn(run_send_recv, world_size=2)
+
+
+def run_variable_a2a(rank, world_size, port):
+    from pipegoose_amd.distributed import functional as F
+    ctx = init_parallel_context(rank, world_size, port)
+    # rank 0 sends [1, 2] rows; rank 1 sends [3, 0] rows
+    in_splits = [1, 2] if rank == 0 else [3, 0]
+    torch.manual_seed(rank)
+    x = torch.arange(sum(in_splits) * 4, dtype=torch.float32).reshape(-1, 4) \
+        + 100 * rank
+    out, out_splits = F.all_to_all_variable(x, in_splits, ctx, MODE)
+    if rank == 0:
+        assert out_splits == [1, 3]
+        assert torch.equal(out[:1], x[:1])                       # own slice
+        assert torch.equal(out[1:], torch.arange(12, dtype=torch.float32)
+                           .reshape(3, 4) + 100)                 # rank1's 3
+    else:
+        assert out_splits == [2, 0]
+        assert torch.equal(out, torch.arange(4, 12, dtype=torch.float32)
+                           .reshape(2, 4))                       # rank0's 2
+    # exchange_splits alone
+    assert F.exchange_splits(in_splits, ctx, MODE) == out_splits
+    ctx.destroy()
+
+
+def test_all_to_all_variable_uneven_with_zero():
+    spawn(run_variable_a2a, world_size=2)
