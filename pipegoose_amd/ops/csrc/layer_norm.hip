@@ -26,10 +26,16 @@ __global__ void layer_norm_fwd_kernel(
 
     using V = typename vec8<T>::type;
     const int HV = H / 8;
+    // cache this thread's packets across the two phases: H <= CACHE*8*BLOCK
+    // rows never re-read x from HBM (one read + one write per element)
+    constexpr int CACHE = 4;
+    V reg[CACHE];
+    const bool cached = HV <= CACHE * BLOCK;
 
     float sum = 0.f, sumsq = 0.f;
-    for (int i = threadIdx.x; i < HV; i += BLOCK) {
+    for (int i = threadIdx.x, c = 0; i < HV; i += BLOCK, ++c) {
         V pkt = reinterpret_cast<const V*>(xr)[i];
+        if (cached) reg[c] = pkt;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
             float f = to_float(pkt.v[j]);
@@ -53,8 +59,8 @@ __global__ void layer_norm_fwd_kernel(
         rstd_out[row] = rstd;
     }
 
-    for (int i = threadIdx.x; i < HV; i += BLOCK) {
-        V pkt = reinterpret_cast<const V*>(xr)[i];
+    for (int i = threadIdx.x, c = 0; i < HV; i += BLOCK, ++c) {
+        V pkt = cached ? reg[c] : reinterpret_cast<const V*>(xr)[i];
         V wp = reinterpret_cast<const V*>(w)[i];
         V bp = reinterpret_cast<const V*>(b)[i];
         V out;
@@ -347,9 +353,12 @@ __global__ void layer_norm_res_fwd_kernel(
 
     using V = typename vec8<T>::type;
     const int HV = H / 8;
+    constexpr int CACHE = 4;
+    V reg[CACHE];
+    const bool cached = HV <= CACHE * BLOCK;
 
     float sum = 0.f, sumsq = 0.f;
-    for (int i = threadIdx.x; i < HV; i += BLOCK) {
+    for (int i = threadIdx.x, c = 0; i < HV; i += BLOCK, ++c) {
         V px = reinterpret_cast<const V*>(xr)[i];
         V pr = reinterpret_cast<const V*>(rr)[i];
         V ps;
@@ -364,6 +373,7 @@ __global__ void layer_norm_res_fwd_kernel(
             sumsq += fs * fs;
         }
         reinterpret_cast<V*>(sr)[i] = ps;
+        if (cached) reg[c] = ps;
     }
     for (int i = HV * 8 + threadIdx.x; i < H; i += BLOCK) {
         T sv = from_float<T>(to_float(xr[i]) + to_float(rr[i]));
@@ -383,8 +393,8 @@ __global__ void layer_norm_res_fwd_kernel(
         rstd_out[row] = rstd;
     }
 
-    for (int i = threadIdx.x; i < HV; i += BLOCK) {
-        V ps = reinterpret_cast<const V*>(sr)[i];
+    for (int i = threadIdx.x, c = 0; i < HV; i += BLOCK, ++c) {
+        V ps = cached ? reg[c] : reinterpret_cast<const V*>(sr)[i];
         V wp = reinterpret_cast<const V*>(w)[i];
         V bp = reinterpret_cast<const V*>(b)[i];
         V out;

@@ -21,10 +21,14 @@ __global__ void rms_norm_fwd_kernel(
 
     using V = typename vec8<T>::type;
     const int HV = H / 8;
+    constexpr int CACHE = 4;
+    V reg[CACHE];
+    const bool cached = HV <= CACHE * BLOCK;
 
     float sumsq = 0.f;
-    for (int i = threadIdx.x; i < HV; i += BLOCK) {
+    for (int i = threadIdx.x, c = 0; i < HV; i += BLOCK, ++c) {
         V pkt = reinterpret_cast<const V*>(xr)[i];
+        if (cached) reg[c] = pkt;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
             float f = to_float(pkt.v[j]);
@@ -39,8 +43,8 @@ __global__ void rms_norm_fwd_kernel(
     const float rstd = rsqrtf(sumsq / H + eps);
     if (threadIdx.x == 0) rstd_out[row] = rstd;
 
-    for (int i = threadIdx.x; i < HV; i += BLOCK) {
-        V pkt = reinterpret_cast<const V*>(xr)[i];
+    for (int i = threadIdx.x, c = 0; i < HV; i += BLOCK, ++c) {
+        V pkt = cached ? reg[c] : reinterpret_cast<const V*>(xr)[i];
         V wp = reinterpret_cast<const V*>(w)[i];
         V out;
 #pragma unroll
