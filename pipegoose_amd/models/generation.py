@@ -69,6 +69,12 @@ def generate(
                 full = F.all_gather(last, dim=-1, parallel_context=ctx,
                                     parallel_mode=ParallelMode.TENSOR)
                 next_tok = _sample(full, temperature, top_k)
+                # sampling is stochastic: all TP ranks must decode the SAME
+                # token — rank 0 of the group decides
+                F.broadcast(next_tok,
+                            src=ctx.get_ranks_in_group(ParallelMode.TENSOR)[0],
+                            parallel_context=ctx,
+                            parallel_mode=ParallelMode.TENSOR)
         else:
             next_tok = last.argmax(dim=-1) if temperature == 0.0 \
                 else _sample(last, temperature, top_k)
