@@ -22,12 +22,32 @@ class RouterOutput:
 
 
 class SwitchNoisePolicy:
-    def __init__(self, eps: float = 0.1):
+    """Multiplicative jitter noise in [1-eps, 1+eps).
+
+    Draws come from a dedicated generator seeded identically on every rank:
+    under tensor parallelism the default CUDA seeds are deliberately
+    TP-decorrelated (Megatron dropout convention), but ROUTING must agree
+    across the expert group — with divergent noise the mask-dispatch path
+    double- or zero-counts tokens."""
+
+    def __init__(self, eps: float = 0.1, seed: int = 1234):
         self.eps = eps
+        self.seed = seed
+        self._gens = {}
+
+    def _gen(self, device: torch.device) -> torch.Generator:
+        key = (device.type, device.index)
+        g = self._gens.get(key)
+        if g is None:
+            g = torch.Generator(device=device)
+            g.manual_seed(self.seed)
+            self._gens[key] = g
+        return g
 
     def sample_like(self, logits: torch.Tensor) -> torch.Tensor:
-        # multiplicative uniform noise in [1-eps, 1+eps)
-        return torch.empty_like(logits).uniform_(1.0 - self.eps, 1.0 + self.eps)
+        noise = torch.rand(logits.shape, device=logits.device,
+                           dtype=logits.dtype, generator=self._gen(logits.device))
+        return 1.0 - self.eps + 2.0 * self.eps * noise
 
 
 class _TopKRouter(nn.Module):

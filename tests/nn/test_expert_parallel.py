@@ -248,3 +248,16 @@ def _run_hybrid_ep(rank, world_size, port):
 
 def test_hybrid_ep_tp_dp_world4():
     spawn(_run_hybrid_ep, world_size=4)
+
+
+def test_noise_policy_identical_across_identical_states():
+    """Same-seeded policies draw identical noise (the TP-sync guarantee)."""
+    torch.manual_seed(1)  # global seed must NOT influence the policy
+    n1 = SwitchNoisePolicy(seed=7)
+    torch.manual_seed(2)
+    n2 = SwitchNoisePolicy(seed=7)
+    x = torch.zeros(4, 8)
+    a = n1.sample_like(x)
+    b = n2.sample_like(x)
+    assert torch.equal(a, b)
+    assert (a >= 0.9).all() and (a <= 1.1).all()
