@@ -35,7 +35,11 @@ class PipelineEngine:
         n_microbatches: int,
         schedule: str = "1f1b",
         loss_fn: Optional[Callable] = None,
+        moe_aux_weight: float = 0.01,
+        moe_z_weight: float = 0.1,
     ):
+        self.moe_aux_weight = moe_aux_weight
+        self.moe_z_weight = moe_z_weight
         self.stage = stage_module
         self.pc = parallel_context
         self.n_microbatches = n_microbatches
@@ -81,6 +85,7 @@ class PipelineEngine:
 
         saved_in: List[Optional[torch.Tensor]] = [None] * m
         saved_out: List[Optional[torch.Tensor]] = [None] * m
+        self._saved_moe = [None] * m  # per-microbatch weighted aux+z losses
         acts = self._actions()
         # recv prefetch bookkeeping: posts must follow the consumption order
         # (= this rank's schedule order = the peer's send order)
@@ -136,25 +141,58 @@ class PipelineEngine:
             saved_in[mb] = x
             out = self.stage(x)
             saved_out[mb] = out
+            # Snapshot MoE router aux/z losses pushed during THIS forward.
+            # They share graph nodes with `out` (the gate reads the stage's
+            # hidden states), so they must ride the SAME backward call as
+            # this microbatch — a separate end-of-run backward would hit an
+            # already-freed graph.  Under PP the last stage's ExpertLoss
+            # wrapper only ever sees its own stage's losses; the engine owns
+            # the weights for every stage instead (moe_aux_weight/moe_z_weight).
+            self._saved_moe[mb] = self._pop_moe_losses()
             if not self.is_last:
                 pending.append(self._send_forward(out, mb))
             elif not has_loss:
                 outputs.append(out.detach())
         else:  # BACKWARD
             out = saved_out[mb]
+            moe = self._saved_moe[mb]
             if self.is_last:
                 if has_loss:
                     loss = self.loss_fn(out, label_mbs[mb]) / m
                     losses.append(loss.detach())
+                    if moe is not None and moe.requires_grad:
+                        loss = loss + moe / m
                     loss.backward()
                 # inference-only: nothing to do
             else:
                 grad = self._recv_backward(mb, out)
-                torch.autograd.backward(out, grad_tensors=grad)
+                if moe is not None and moe.requires_grad:
+                    torch.autograd.backward([out, moe / m],
+                                            grad_tensors=[grad, None])
+                else:
+                    torch.autograd.backward(out, grad_tensors=grad)
             x = saved_in[mb]
             if not self.is_first and x is not None and x.grad is not None:
                 pending.append(self._send_backward(x.grad, mb))
             saved_in[mb] = saved_out[mb] = None  # free activations
+            self._saved_moe[mb] = None
+
+    def _pop_moe_losses(self):
+        """Drain ExpertContext into one weighted scalar (or None)."""
+        from pipegoose_amd.nn.expert_parallel import ExpertContext
+        ectx = ExpertContext.get_instance()
+        aux = ectx.pop_all_aux_loss()
+        zl = ectx.pop_all_z_loss()
+        if not aux and not zl:
+            return None
+        total = None
+        for a in aux:
+            t = a * self.moe_aux_weight
+            total = t if total is None else total + t
+        for z in zl:
+            t = z * self.moe_z_weight
+            total = t if total is None else total + t
+        return total
 
     # ------------------------------------------------------------- transport
 

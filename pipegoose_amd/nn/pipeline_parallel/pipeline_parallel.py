@@ -42,12 +42,18 @@ class PipelineParallel(Parallel):
         schedule: str = "1f1b",
         loss_fn: Optional[Callable] = None,
         partition_sizes=None,
+        moe_aux_weight: float = 0.01,
+        moe_z_weight: float = 0.1,
     ):
         super().__init__(module, parallel_context)
         self.n_microbatches = n_microbatches
         self.schedule = schedule
         self.loss_fn = loss_fn
         self.partition_sizes = partition_sizes
+        # Under PP, MoE aux/z losses are consumed by the ENGINE per stage and
+        # per microbatch (ExpertLoss can't see non-last stages' routers):
+        self.moe_aux_weight = moe_aux_weight
+        self.moe_z_weight = moe_z_weight
 
     def parallelize(self) -> nn.Module:
         pp = self.parallel_context.get_world_size(ParallelMode.PIPELINE)
@@ -59,7 +65,9 @@ class PipelineParallel(Parallel):
         ).get_model_partition()
         engine = PipelineEngine(
             stage, self.parallel_context, self.n_microbatches,
-            schedule=self.schedule, loss_fn=self.loss_fn)
+            schedule=self.schedule, loss_fn=self.loss_fn,
+            moe_aux_weight=self.moe_aux_weight,
+            moe_z_weight=self.moe_z_weight)
         wrapped = PipelineStageModule(engine)
         self._save_metadata(wrapped, self.parallel_context)
         return wrapped

@@ -184,3 +184,86 @@ def test_1f1b_edge_schedules():
                     seen_f.add(t.microbatch_idx)
                 else:
                     assert t.microbatch_idx in seen_f, (m, p, rank)
+
+
+# ------------------------------------------------------------------ MoE + PP
+
+def run_pp_moe(rank, world_size, port):
+    """MoE layers on BOTH stages: the engine must fold each stage's router
+    aux/z losses into that microbatch's backward (ExpertLoss on the last
+    stage can never see stage-0's routers).  Oracle: per-microbatch loop on
+    one process with the same weights."""
+    from pipegoose_amd.nn.expert_parallel import (ExpertContext, ExpertLayer,
+                                                  SwitchNoisePolicy, Top1Router)
+    from pipegoose_amd.nn.pipeline_parallel import PipelineParallel
+    from pipegoose_amd.nn.pipeline_parallel.partitioner import UniformPartitioner
+
+    ctx = init_parallel_context(rank, world_size, port,
+                                pipeline_parallel_size=world_size)
+
+    def build(seed):
+        torch.manual_seed(seed)
+        def expert():
+            return nn.Sequential(nn.Linear(HID, HID), nn.GELU(),
+                                 nn.Linear(HID, HID))
+        def moe():
+            return ExpertLayer(
+                2, expert(),
+                Top1Router(SwitchNoisePolicy(eps=0.0), 2, HID),
+                enable_tensor_parallel=False, parallel_context=ctx)
+        return nn.Sequential(nn.Linear(HID, HID), moe(), nn.Tanh(),
+                             nn.Linear(HID, HID), moe(), nn.Linear(HID, HID))
+
+    model = build(11)
+    ref = build(11)
+
+    torch.manual_seed(50)
+    x = torch.randn(8, 1, HID)  # [B, S=1, H] — ExpertLayer expects 3-D
+    target = torch.randn(8, 1, HID)
+
+    AUX_W, Z_W = 0.01, 0.1
+    m = 2
+    ExpertContext.get_instance().pop_all_aux_loss()
+    ExpertContext.get_instance().pop_all_z_loss()
+
+    pp = PipelineParallel(model, ctx, n_microbatches=m, loss_fn=_loss_fn,
+                          moe_aux_weight=AUX_W, moe_z_weight=Z_W).parallelize()
+    loss = pp(x, target)
+    # engine must leave the context drained
+    assert ExpertContext.get_instance().aux_losses == []
+
+    # oracle: identical per-microbatch accumulation on one process
+    ectx = ExpertContext.get_instance()
+    ref_losses = []
+    for x_mb, t_mb in zip(x.chunk(m), target.chunk(m)):
+        out = ref(x_mb)
+        lm = _loss_fn(out, t_mb) / m
+        ref_losses.append(lm.detach())
+        aux = sum(ectx.pop_all_aux_loss())
+        zl = sum(ectx.pop_all_z_loss())
+        (lm + (AUX_W * aux + Z_W * zl) / m).backward()
+
+    if rank == world_size - 1:
+        ref_loss = torch.stack(ref_losses).sum()
+        assert torch.allclose(loss, ref_loss, atol=1e-6), (loss, ref_loss)
+
+    stages = UniformPartitioner(ref, ctx).split(world_size)
+    own_params = dict(pp.named_parameters())
+    router_grads = 0
+    for name, p_ref in stages[rank].named_parameters():
+        p = own_params["stage." + name]
+        if p_ref.grad is None:  # expert that received zero tokens
+            assert p.grad is None or p.grad.abs().max() == 0, name
+            continue
+        assert p.grad is not None, f"stage {rank}: no grad for {name}"
+        assert torch.allclose(p.grad, p_ref.grad, atol=1e-6), \
+            f"{name}: {(p.grad - p_ref.grad).abs().max()}"
+        if "router.gate" in name:
+            router_grads += 1
+            assert p.grad.abs().sum() > 0, f"router grad zero: {name}"
+    assert router_grads > 0, f"stage {rank} holds no router (bad split)"
+    ctx.destroy()
+
+
+def test_pp2_moe_aux_losses_reach_both_stages():
+    spawn(run_pp_moe, world_size=2)
