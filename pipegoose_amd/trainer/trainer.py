@@ -31,6 +31,7 @@ class Trainer:
         max_grad_norm: Optional[float] = None,
         log_interval: int = 10,
         logger: Optional[DistributedLogger] = None,
+        hang_timeout_s: Optional[float] = None,
     ):
         self.model = model
         self.optimizer = optimizer
@@ -43,6 +44,11 @@ class Trainer:
         self.state = TrainerState()
         self.meter = ThroughputMeter()
         self.logger = logger or DistributedLogger(parallel_context=self.ctx)
+        # failure detection: a desynced collective hangs every
+        # rank silently; with hang_timeout_s set, fit() arms a watchdog that
+        # dumps stacks and exits the process if a step stops making progress
+        self.hang_timeout_s = hang_timeout_s
+        self._watchdog = None
 
     # ----------------------------------------------------------------- steps
 
@@ -110,16 +116,26 @@ class Trainer:
             max_steps: Optional[int] = None):
         self.state.stage = TrainerStage.TRAINING
         self.callbacks.fire("on_fit_start", self)
-        for epoch in range(epochs):
-            self.state.epoch = epoch
-            self.callbacks.fire("on_epoch_start", self)
-            for batch in train_loader:
-                self.train(batch)
+        if self.hang_timeout_s is not None:
+            from pipegoose_amd.utils.watchdog import HangWatchdog
+            self._watchdog = HangWatchdog(self.hang_timeout_s).start()
+        try:
+            for epoch in range(epochs):
+                self.state.epoch = epoch
+                self.callbacks.fire("on_epoch_start", self)
+                for batch in train_loader:
+                    self.train(batch)
+                    if self._watchdog is not None:
+                        self._watchdog.tick()
+                    if max_steps is not None and self.state.global_step >= max_steps:
+                        break
+                self.callbacks.fire("on_epoch_end", self)
                 if max_steps is not None and self.state.global_step >= max_steps:
                     break
-            self.callbacks.fire("on_epoch_end", self)
-            if max_steps is not None and self.state.global_step >= max_steps:
-                break
+        finally:
+            if self._watchdog is not None:
+                self._watchdog.stop()
+                self._watchdog = None
         self.state.stage = TrainerStage.FINISHED
         self.callbacks.fire("on_fit_end", self)
         return self.state
