@@ -1,0 +1,92 @@
+"""Convergence smoke tests (CPU-sized versions of the reference's headline
+evidence, README.md:87-92 — DP/ZeRO convergence W&B runs; its TP convergence
+was retracted, so the hybrid test here is coverage the reference never had).
+
+Criterion: overfitting a fixed tiny batch must cut the loss by a large factor
+in a few dozen steps — catches broken grads, desynced replicas, and optimizer
+state bugs that single-step parity tests can miss."""
+import torch
+import torch.nn.functional as TF
+
+from pipegoose_amd.models.bloom import BloomForCausalLM, bloom_tiny
+from pipegoose_amd.nn import DataParallel
+from pipegoose_amd.optim import DistributedOptimizer
+from pipegoose_amd.testing import init_parallel_context, spawn
+
+
+def lm_loss(logits, labels):
+    shift_logits = logits[:, :-1].reshape(-1, logits.size(-1)).float()
+    shift_labels = labels[:, 1:].reshape(-1)
+    return TF.cross_entropy(shift_logits, shift_labels)
+
+
+def _run_single(rank, world_size, port):
+    ctx = init_parallel_context(rank, world_size, port)
+    torch.manual_seed(0)
+    model = BloomForCausalLM(bloom_tiny(), ctx)
+    opt = torch.optim.Adam(model.parameters(), lr=3e-3)
+    torch.manual_seed(1)
+    ids = torch.randint(0, 256, (4, 16))
+
+    first = None
+    for _ in range(30):
+        opt.zero_grad()
+        loss = lm_loss(model(ids), ids)
+        if first is None:
+            first = loss.item()
+        loss.backward()
+        opt.step()
+    assert torch.isfinite(loss)
+    assert loss.item() < 0.35 * first, (first, loss.item())
+    ctx.destroy()
+
+
+def test_single_process_overfit():
+    spawn(_run_single, world_size=1)
+
+
+def _run_tp_dp_zero(rank, world_size, port):
+    """TP2 x DP2 + ZeRO-1 (shard-routed reduce) + grad clip, 15 steps on a
+    fixed batch: loss must drop sharply and DP replicas must stay bit-synced
+    through optimizer state, clipping, and shard broadcasts."""
+    import torch.distributed as dist
+    from pipegoose_amd.distributed.parallel_mode import ParallelMode
+    from pipegoose_amd.nn.tensor_parallel import VocabParallelCrossEntropy
+
+    ctx = init_parallel_context(rank, world_size, port,
+                                tensor_parallel_size=2, data_parallel_size=2)
+    torch.manual_seed(0)
+    model = BloomForCausalLM(bloom_tiny(), ctx)
+    model = DataParallel(model, ctx).parallelize()
+    opt = DistributedOptimizer(torch.optim.Adam(model.parameters(), lr=3e-3),
+                               ctx, grad_reduce="shard")
+    vp_ce = VocabParallelCrossEntropy(parallel_context=ctx)
+
+    # same batch on both DP replicas -> losses must agree exactly
+    torch.manual_seed(1)
+    ids = torch.randint(0, 256, (4, 16))
+
+    first = last = None
+    for _ in range(15):
+        opt.zero_grad()
+        logits = model(ids)
+        loss = vp_ce(logits[:, :-1].contiguous(), ids[:, 1:].contiguous())
+        if first is None:
+            first = loss.item()
+        loss.backward()
+        opt.clip_grad_norm_(1.0)
+        opt.step()
+        last = loss.item()
+    assert last < 0.5 * first, (first, last)
+
+    # replicas identical across DATA, shards identical across TENSOR-pairs
+    flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    peers = [torch.empty_like(flat) for _ in
+             range(ctx.get_world_size(ParallelMode.DATA))]
+    dist.all_gather(peers, flat, group=ctx.get_group(ParallelMode.DATA))
+    assert torch.equal(peers[0], peers[1]), "DP replicas diverged during training"
+    ctx.destroy()
+
+
+def test_tp2_dp2_zero1_convergence_and_replica_sync():
+    spawn(_run_tp_dp_zero, world_size=4)
