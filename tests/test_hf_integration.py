@@ -175,3 +175,35 @@ def _run_tp2_hf_llama_gqa(rank, world_size, port):
 
 def test_tensor_parallel_hf_llama_gqa_tp2():
     spawn(_run_tp2_hf_llama_gqa, world_size=2)
+
+
+def _hf_albert():
+    from transformers import AlbertConfig, AlbertModel
+    cfg = AlbertConfig(vocab_size=128, embedding_size=32, hidden_size=64,
+                       num_hidden_layers=2, num_attention_heads=4,
+                       intermediate_size=128)
+    torch.manual_seed(90)
+    return AlbertModel(cfg)
+
+
+def _run_tp2_hf_albert(rank, world_size, port):
+    """reference ships an albert TP mapping (nn/parallel_mapping.py:24-31)
+    but never ran it; here it is parity-tested: TP2 output == full model."""
+    ctx = init_parallel_context(rank, world_size, port, tensor_parallel_size=2)
+    model = _hf_albert()
+    ref = _hf_albert()
+    torch.manual_seed(91)
+    ids = torch.randint(0, 128, (2, 12))
+
+    with torch.no_grad():
+        ref_out = ref(ids).last_hidden_state
+
+    model = TensorParallel(model, ctx).parallelize()
+    with torch.no_grad():
+        out = model(ids).last_hidden_state
+    assert torch.allclose(out, ref_out, atol=1e-4), (out - ref_out).abs().max()
+    ctx.destroy()
+
+
+def test_tensor_parallel_hf_albert_tp2():
+    spawn(_run_tp2_hf_albert, world_size=2)
