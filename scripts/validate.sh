@@ -16,6 +16,10 @@ python -m torch.distributed.run --nnodes=1 --nproc-per-node 2 \
     --master-addr 127.0.0.1 --master-port 29801 \
     bench.py --gpus 2 --steps 2 --warmup 1 --model bloom-tiny \
     --seq-len 128 --micro-batch 4
+python -m torch.distributed.run --nnodes=1 --nproc-per-node 4 \
+    --master-addr 127.0.0.1 --master-port 29803 \
+    bench.py --gpus 4 --steps 2 --warmup 1 --model bloom-tiny \
+    --tp 2 --dp 2 --seq-len 128 --micro-batch 4
 python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
     --master-addr 127.0.0.1 --master-port 29802 \
     bench.py --gpus 8 --steps 2 --warmup 1 --model bloom-tiny \
