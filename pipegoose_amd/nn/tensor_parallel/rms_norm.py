@@ -3,6 +3,8 @@
 GPU: fused CDNA4 kernel (ops/csrc/rms_norm.hip); CPU: fp32 eager oracle.
 Same sequence-parallel treatment as LayerNorm: under SP the (replicated)
 weight grad is partial over local tokens, fixed in-graph via _Broadcast.
+
+Beyond-reference (the reference had no RMSNorm/llama support).
 """
 import torch
 from torch import nn

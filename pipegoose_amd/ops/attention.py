@@ -5,6 +5,9 @@ slope*(j-i) computed in-kernel from the per-head slopes (no [H,S,S] mask
 tensor) — see csrc/attention.hip.  Until/unless the kernel supports a given
 shape, falls back to torch sdpa with an explicit additive mask.
 CPU path: sdpa math with the mask (numerics oracle).
+
+No reference counterpart (the reference has no attention kernels at all);
+this is MI355X-native.
 """
 import os
 

@@ -3,6 +3,9 @@
 On GPU: GEMM via hipBLASLt (torch.matmul) + one fused HIP kernel applying
 bias+GeLU (tanh approx) in a single HBM pass, with a fused dgelu backward.
 CPU path: eager bias + torch GELU (the numerics oracle).
+
+MI355X-native fusion of the reference's separate bias-add + GELU modules
+('fused bias-GeLU for mlp.dense_h_to_4h').
 """
 import os
 

@@ -2,6 +2,9 @@
 
 CPU path: torch.nn.functional.layer_norm (the numerics oracle the GPU kernel
 is tested against, tests/ops/test_layer_norm.py).
+
+MI355X-native kernel wrapper, no reference counterpart (the reference used
+plain nn.LayerNorm — nn/tensor_parallel/layer_norm.py:23-25).
 """
 import os
 

@@ -1,4 +1,7 @@
-"""Fused RMSNorm (CPU oracle: explicit fp32 rms computation)."""
+"""Fused RMSNorm (CPU oracle: explicit fp32 rms computation).
+
+Beyond-reference kernel (no RMSNorm anywhere in the reference).
+"""
 import os
 
 import torch

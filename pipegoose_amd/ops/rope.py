@@ -2,6 +2,9 @@
 
 GPU: in-kernel cos/sin from the base frequency (no [S, D] tables in HBM);
 backward is the inverse rotation.  CPU oracle: explicit fp32 rotation.
+
+Beyond-reference (the reference had no RoPE/llama support); SURVEY §2.7 item 1
+names RoPE as a required native kernel.
 """
 import os
 

@@ -1,4 +1,7 @@
-"""Fused SwiGLU: silu(gate) * up in one HBM pass (llama-family MLPs)."""
+"""Fused SwiGLU: silu(gate) * up in one HBM pass (llama-family MLPs).
+
+Beyond-reference (no llama family in the reference).
+"""
 import os
 
 import torch
