@@ -131,6 +131,23 @@ class BloomAttention(nn.Module):
         bias = bias.masked_fill(rel[None] > 0, float("-inf"))
         return bias.to(dtype)[None]
 
+    def _alibi_bias_graph(self, k_len: int, pos_t: torch.Tensor, dtype):
+        """Graph-mode decode bias [1, H, 1, k_len]: the single query sits at
+        absolute position ``pos_t`` (a device tensor — no host value enters
+        the graph); keys beyond it (unfilled cache slots) get -inf."""
+        if not hasattr(self, "_kpos_cache"):
+            self._kpos_cache = {}
+        key = (k_len, pos_t.device)
+        kpos = self._kpos_cache.get(key)
+        if kpos is None:
+            kpos = torch.arange(k_len, device=pos_t.device)
+            self._kpos_cache[key] = kpos
+        rel = (kpos - pos_t).float()                         # [k_len]
+        bias = self.alibi_slopes.to(device=pos_t.device,
+                                    dtype=torch.float32)[:, None, None] * rel
+        bias = bias.masked_fill(kpos > pos_t, float("-inf"))  # [H, 1, k_len]
+        return bias.to(dtype)[None]
+
     def forward(self, hidden: torch.Tensor, past_kv=None, use_cache: bool = False):
         B = hidden.size(0)
         fused = self.query_key_value(hidden)  # [B, S, local_heads * 3 * hd]
@@ -154,7 +171,13 @@ class BloomAttention(nn.Module):
         from pipegoose_amd.ops.attention import (_kernel_supported,
                                                  alibi_attention,
                                                  alibi_attention_qkv)
-        if k.size(2) != S:
+        if present is not None and getattr(present, "graph_mode", False):
+            # hipGraph-capturable decode: constant shapes (full cache
+            # length), position enters as DATA via the cache's device pos_t
+            bias = self._alibi_bias_graph(k.size(2), present.pos_t, q.dtype)
+            out = TF.scaled_dot_product_attention(
+                q, k, v, attn_mask=bias, scale=self.inv_norm)
+        elif k.size(2) != S:
             # incremental decode: rectangular mask, torch sdpa
             bias = self._alibi_bias_rect(S, k.size(2), q.device, q.dtype)
             out = TF.scaled_dot_product_attention(
@@ -366,6 +389,15 @@ class BloomForCausalLM(nn.Module):
         return StaticKVCache(len(self.transformer.h), batch_size,
                              attn.num_heads, max_len, attn.head_dim,
                              p.dtype, p.device)
+
+    def new_graph_kv_cache(self, batch_size: int, max_len: int):
+        """Cache bank for hipGraph-captured decode (models/graph_decode.py)."""
+        from pipegoose_amd.models.kv_cache import GraphKVCache
+        attn = self.transformer.h[0].self_attention
+        p = next(self.parameters())
+        return GraphKVCache(len(self.transformer.h), batch_size,
+                            attn.num_heads, max_len, attn.head_dim,
+                            p.dtype, p.device)
 
     def gradient_checkpointing_enable(self, enabled: bool = True):
         """Recompute each block in backward instead of storing activations —

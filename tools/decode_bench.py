@@ -20,6 +20,8 @@ def main():
     p.add_argument("--batch", type=int, default=8)
     p.add_argument("--prompt-len", type=int, default=512)
     p.add_argument("--new-tokens", type=int, default=64)
+    p.add_argument("--graph", action="store_true",
+                   help="also time the hipGraph-captured decode loop")
     args = p.parse_args()
 
     os.environ.setdefault("RANK", "0")
@@ -69,6 +71,23 @@ def main():
           f"prefill {B * P / t_prefill:,.0f} tok/s ({t_prefill * 1e3:.1f} ms) | "
           f"decode {t_decode / N * 1e3:.2f} ms/step "
           f"= {B * N / t_decode:,.0f} tok/s")
+
+    if args.graph:
+        from pipegoose_amd.models.graph_decode import GraphDecoder
+        dec = GraphDecoder(model, batch_size=B, max_len=P + 2 * N + 8)
+        with torch.no_grad():
+            dec.generate(ids, max_new_tokens=N)  # warmup + capture
+            if use_gpu:
+                torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            out = dec.generate(ids, max_new_tokens=N)
+            if use_gpu:
+                torch.cuda.synchronize()
+            t_graph = time.perf_counter() - t0
+        mode = "hipGraph" if dec._graph is not None else "eager-fallback"
+        print(f"  graph decode ({mode}): includes prefill; "
+              f"{(t_graph - t_prefill) / N * 1e3:.2f} ms/step "
+              f"= {B * N / (t_graph - t_prefill):,.0f} tok/s")
     ctx.destroy()
 
 
