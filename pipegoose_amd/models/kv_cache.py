@@ -59,6 +59,11 @@ class GraphLayerKVCache(LayerKVCache):
 
     def __init__(self, B, H, max_len, D, dtype, device, pos_t: torch.Tensor):
         super().__init__(B, H, max_len, D, dtype, device)
+        # full-length attention reads UNFILLED slots (masked by -inf bias);
+        # they must be finite — inf garbage in q·k would make inf + (-inf)
+        # = NaN and poison the whole softmax row
+        self.k.zero_()
+        self.v.zero_()
         self.pos_t = pos_t          # int64 [1], shared across layers
         self.graph_mode = False
 
