@@ -105,14 +105,23 @@ class LlamaAttention(nn.Module):
         q = q.view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
         k = k.view(B, S, self.num_kv_heads, self.head_dim).transpose(1, 2)
         v = v.view(B, S, self.num_kv_heads, self.head_dim).transpose(1, 2)
-        if past_kv is None:
-            offset = 0
-        elif hasattr(past_kv, "len"):
-            offset = past_kv.len
+        graph_mode = past_kv is not None and getattr(past_kv, "graph_mode",
+                                                     False)
+        if graph_mode:
+            # hipGraph decode: the absolute position is the cache's device
+            # pos_t — no host value may enter the captured step
+            from pipegoose_amd.ops.rope import rope_at_position
+            q = rope_at_position(q, self.rope_theta, past_kv.pos_t)
+            k = rope_at_position(k, self.rope_theta, past_kv.pos_t)
         else:
-            offset = past_kv[0].size(2)
-        q = apply_rope(q, self.rope_theta, pos_offset=offset)
-        k = apply_rope(k, self.rope_theta, pos_offset=offset)
+            if past_kv is None:
+                offset = 0
+            elif hasattr(past_kv, "len"):
+                offset = past_kv.len
+            else:
+                offset = past_kv[0].size(2)
+            q = apply_rope(q, self.rope_theta, pos_offset=offset)
+            k = apply_rope(k, self.rope_theta, pos_offset=offset)
         present = None
         if past_kv is not None:
             if hasattr(past_kv, "append"):     # StaticKVCache layer
@@ -128,7 +137,26 @@ class LlamaAttention(nn.Module):
                                                  alibi_attention)
         kernel_ok = (_kernel_supported(q) and past_kv is None
                      and not use_cache)
-        if k.size(2) != S or (self.kv_group > 1 and not kernel_ok):
+        if graph_mode:
+            # full-length masked attention over the static cache (constant
+            # shapes; unfilled slots are zero + masked by -inf)
+            ke = k.repeat_interleave(self.kv_group, dim=1) \
+                if self.kv_group > 1 else k
+            ve = v.repeat_interleave(self.kv_group, dim=1) \
+                if self.kv_group > 1 else v
+            kl = ke.size(2)
+            if not hasattr(self, "_kpos_cache"):
+                self._kpos_cache = {}
+            kpos = self._kpos_cache.get((kl, q.device))
+            if kpos is None:
+                kpos = torch.arange(kl, device=q.device)
+                self._kpos_cache[(kl, q.device)] = kpos
+            bias = torch.zeros(1, 1, 1, kl, device=q.device, dtype=q.dtype)
+            bias = bias.masked_fill((kpos > past_kv.pos_t)[None, None, None],
+                                    float("-inf"))
+            out = TF.scaled_dot_product_attention(
+                q, ke, ve, attn_mask=bias, scale=self.inv_norm)
+        elif k.size(2) != S or (self.kv_group > 1 and not kernel_ok):
             # decode / CPU-GQA fallback: expand kv heads, rect mask, sdpa
             ke = k.repeat_interleave(self.kv_group, dim=1) \
                 if self.kv_group > 1 else k
@@ -288,6 +316,15 @@ class LlamaForCausalLM(nn.Module):
         return StaticKVCache(len(self.model.layers), batch_size,
                              attn.num_kv_heads, max_len, attn.head_dim,
                              p.dtype, p.device)
+
+    def new_graph_kv_cache(self, batch_size: int, max_len: int):
+        """Cache bank for hipGraph-captured decode (models/graph_decode.py)."""
+        from pipegoose_amd.models.kv_cache import GraphKVCache
+        attn = self.model.layers[0].self_attn
+        p = next(self.parameters())
+        return GraphKVCache(len(self.model.layers), batch_size,
+                            attn.num_kv_heads, max_len, attn.head_dim,
+                            p.dtype, p.device)
 
     def gradient_checkpointing_enable(self, enabled: bool = True):
         """Recompute each block in backward instead of storing activations —

@@ -50,3 +50,19 @@ def apply_rope(x: torch.Tensor, theta_base: float = 10000.0,
     if x.is_cuda and os.environ.get("PIPEGOOSE_DISABLE_EXT") != "1":
         return _Rope.apply(x.contiguous(), theta_base, pos_offset)
     return _rope_ref(x, theta_base, pos_offset)
+
+
+def rope_at_position(x: torch.Tensor, theta_base: float,
+                     pos_t: torch.Tensor) -> torch.Tensor:
+    """x: [B, H, 1, D], pos_t: int64 device tensor [1] — rotate the single
+    decode step at a DATA-carried position (hipGraph-capturable: no host
+    value enters; same half-split math as _rope_ref)."""
+    half = x.size(-1) // 2
+    d = torch.arange(half, device=x.device, dtype=torch.float32)
+    freqs = theta_base ** (-2.0 * d / x.size(-1))
+    ang = pos_t.float()[:, None] * freqs          # [1, half]
+    cos, sin = ang.cos(), ang.sin()
+    x1, x2 = x[..., :half].float(), x[..., half:].float()
+    y1 = x1 * cos - x2 * sin
+    y2 = x2 * cos + x1 * sin
+    return torch.cat([y1, y2], dim=-1).to(x.dtype)

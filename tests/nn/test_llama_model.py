@@ -225,3 +225,28 @@ def _run_sample_sync_tp2(rank, world_size, port):
 
 def test_sampling_synced_across_tp_ranks():
     spawn(_run_sample_sync_tp2, world_size=2)
+
+
+def _run_llama_graph_decoder(rank, world_size, port):
+    """Graph-mode static decode path (device-pos RoPE + full-length masked
+    attention, incl. the GQA expand) must match generate()'s greedy tokens."""
+    import dataclasses
+    from pipegoose_amd.models.graph_decode import GraphDecoder
+
+    ctx = init_parallel_context(rank, world_size, port)
+    torch.manual_seed(55)
+    cfg = dataclasses.replace(llama_tiny(), n_kv_head=2)  # exercise GQA
+    model = LlamaForCausalLM(cfg, ctx)
+    model.eval()
+    torch.manual_seed(56)
+    prompt = torch.randint(0, 256, (2, 9))
+
+    ref = model.generate(prompt, max_new_tokens=7)
+    dec = GraphDecoder(model, batch_size=2, max_len=32)
+    out = dec.generate(prompt, max_new_tokens=7)
+    assert torch.equal(out, ref[:, -7:]), (out, ref)
+    ctx.destroy()
+
+
+def test_llama_graph_decoder_matches_generate_cpu():
+    spawn(_run_llama_graph_decoder, world_size=1)

@@ -539,3 +539,31 @@ def test_graph_decoder_capture_parity_gpu():
     out2 = dec.generate(prompt, max_new_tokens=12)
     assert torch.equal(out2, ref)
     ctx.destroy()
+
+
+def test_llama_graph_decoder_capture_parity_gpu():
+    """hipGraph capture of the llama decode step (device-pos RoPE + GQA
+    expand + full-length masked sdpa) must match eager greedy tokens."""
+    import dataclasses
+    import os
+    from pipegoose_amd.models.llama import LlamaForCausalLM, llama_tiny
+    from pipegoose_amd.models.graph_decode import GraphDecoder
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29881")
+    os.environ.setdefault("RANK", "0")
+    os.environ.setdefault("WORLD_SIZE", "1")
+    from pipegoose_amd import ParallelContext
+    ctx = ParallelContext.from_torch()
+    torch.manual_seed(72)
+    cfg = dataclasses.replace(llama_tiny(), n_kv_head=2)
+    model = LlamaForCausalLM(cfg, ctx).to("cuda", torch.bfloat16).eval()
+    torch.manual_seed(73)
+    prompt = torch.randint(0, 256, (2, 16), device="cuda")
+
+    ref = model.generate(prompt, max_new_tokens=12)[:, -12:]
+    dec = GraphDecoder(model, batch_size=2, max_len=64)
+    out = dec.generate(prompt, max_new_tokens=12)
+    assert dec._graph is not None, "hipGraph capture failed (fell back to eager)"
+    assert torch.equal(out, ref), (out, ref)
+    ctx.destroy()
