@@ -48,6 +48,24 @@ def main():
         print(f"generated {n_new} tokens x batch {args.batch} "
               f"in {dt_s * 1e3:.1f} ms "
               f"({args.batch * n_new / dt_s:,.0f} tokens/s decode incl. prefill)")
+
+    if ctx.get_world_size() == 1:
+        # single-GPU fast path: hipGraph-captured decode loop (2.24x measured
+        # on MI355X — models/graph_decode.py)
+        from pipegoose_amd.models.graph_decode import GraphDecoder
+        dec = GraphDecoder(model, batch_size=args.batch,
+                           max_len=args.prompt_len + 2 * args.new_tokens + 8)
+        dec.generate(ids, max_new_tokens=args.new_tokens)  # warmup + capture
+        if dev == "cuda":
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        out2 = dec.generate(ids, max_new_tokens=args.new_tokens)
+        if dev == "cuda":
+            torch.cuda.synchronize()
+        dt_g = time.perf_counter() - t0
+        mode = "hipGraph" if dec._graph is not None else "eager"
+        print(f"graph decoder ({mode}): {args.batch * out2.size(1) / dt_g:,.0f} "
+              f"tokens/s incl. prefill")
     ctx.destroy()
 
 
