@@ -66,16 +66,29 @@ class GraphDecoder:
 
     @torch.no_grad()
     def _capture(self):
-        s = torch.cuda.Stream()
-        s.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(s):  # allocator warmup, runs 1 REAL step
+        """Run ONE real warmup step (allocator priming), then record the
+        graph.  Always returns the warmup step's token — even when capture
+        fails, that step advanced the state and its token must be kept."""
+        try:
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+        except Exception:       # no device: nothing ran, nothing to return
+            self.use_graph = False
+            self._graph = None
+            return None
+        with torch.cuda.stream(s):
             self._step()
         torch.cuda.current_stream().wait_stream(s)
         warmup_token = self.ids.clone()
-        g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):   # records, does not execute
-            self._step()
-        self._graph = g
+        try:
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):   # records, does not execute
+                self._step()
+            self._graph = g
+        except Exception:
+            # capture unsupported here: permanent eager fallback
+            self.use_graph = False
+            self._graph = None
         return warmup_token
 
     # -------------------------------------------------------------- generate
@@ -99,19 +112,13 @@ class GraphDecoder:
 
         tokens = [self.ids.clone()]
         remaining = max_new_tokens - 1
-        if self.use_graph and remaining > 1:
-            if self._graph is None:
-                try:
-                    tokens.append(self._capture())
-                    remaining -= 1
-                except Exception:
-                    # capture unsupported here: permanent eager fallback
-                    self.use_graph = False
-                    self._graph = None
-            else:
-                # a previously captured graph replays against the CURRENT
-                # buffer contents — no re-capture needed across calls
-                pass
+        if self.use_graph and remaining > 1 and self._graph is None:
+            warmup_token = self._capture()
+            if warmup_token is not None:  # the warmup step really ran
+                tokens.append(warmup_token)
+                remaining -= 1
+        # a previously captured graph replays against the CURRENT buffer
+        # contents — no re-capture needed across calls
         for _ in range(remaining):
             if self._graph is not None:
                 self._graph.replay()

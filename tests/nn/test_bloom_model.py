@@ -207,3 +207,40 @@ def _run_graph_decoder_parity(rank, world_size, port):
 
 def test_graph_decoder_matches_generate_cpu():
     spawn(_run_graph_decoder_parity, world_size=1)
+
+
+def _run_graph_decoder_edges(rank, world_size, port):
+    """Boundary + guard behavior: exact max_len fill, batch/overflow asserts,
+    and use_graph=True falling back cleanly on CPU (capture impossible)."""
+    from pipegoose_amd.models.graph_decode import GraphDecoder
+
+    ctx = init_parallel_context(rank, world_size, port)
+    torch.manual_seed(81)
+    model = BloomForCausalLM(bloom_tiny(), ctx).eval()
+    torch.manual_seed(82)
+    prompt = torch.randint(0, 256, (2, 10))
+
+    # exact fill: P + n == max_len must work
+    dec = GraphDecoder(model, batch_size=2, max_len=16)
+    out = dec.generate(prompt, max_new_tokens=6)
+    assert out.shape == (2, 6)
+    ref = model.generate(prompt, max_new_tokens=6)[:, -6:]
+    assert torch.equal(out, ref)
+
+    # overflow and batch-mismatch guards
+    import pytest as _pytest
+    with _pytest.raises(AssertionError):
+        dec.generate(prompt, max_new_tokens=7)
+    with _pytest.raises(AssertionError):
+        dec.generate(prompt[:1], max_new_tokens=2)
+
+    # forcing use_graph on CPU: capture fails, falls back, tokens still right
+    dec2 = GraphDecoder(model, batch_size=2, max_len=32, use_graph=True)
+    out2 = dec2.generate(prompt, max_new_tokens=6)
+    assert dec2._graph is None and not dec2.use_graph
+    assert torch.equal(out2, ref)
+    ctx.destroy()
+
+
+def test_graph_decoder_edge_cases():
+    spawn(_run_graph_decoder_edges, world_size=1)
