@@ -19,6 +19,16 @@ class LMHead(ParallelInfo):
     pass
 
 
+class FusedColumn(ParallelInfo):
+    """Column-parallel layer whose output dim stacks n projections BLOCKWISE
+    (GPT-2 ``c_attn``: [Q|K|V] along the out dim) — each block must be sliced
+    separately, a plain chunk would hand rank 0 all of Q."""
+
+    def __init__(self, *names, n: int = 3, **kwargs):
+        super().__init__(*names, **kwargs)
+        self.n = n
+
+
 class TensorParallelMapping(ParallelMapping):
     __MAPPING__ = {
         "bloom-560m": [
@@ -36,6 +46,12 @@ class TensorParallelMapping(ParallelMapping):
             Row("self_attn.o_proj", "mlp.down_proj"),
             LMHead("lm_head"),
         ],
+        "gpt2": [
+            FusedColumn("attn.c_attn", n=3),  # Conv1D, blockwise q|k|v
+            Column("mlp.c_fc"),
+            Row("attn.c_proj", "mlp.c_proj"),
+            LMHead("lm_head"),
+        ],
         # pipegoose_amd native models (models/{bloom,llama}.py) use the same
         # names as the HF architectures.
     }
@@ -51,3 +67,12 @@ class TensorParallelMapping(ParallelMapping):
     @classmethod
     def is_lm_head(cls, module_name: str) -> bool:
         return isinstance(cls._search(module_name), LMHead)
+
+    @classmethod
+    def is_fused_column(cls, module_name: str) -> bool:
+        return isinstance(cls._search(module_name), FusedColumn)
+
+    @classmethod
+    def get_fused_count(cls, module_name: str) -> int:
+        info = cls._search(module_name)
+        return info.n if isinstance(info, FusedColumn) else 1

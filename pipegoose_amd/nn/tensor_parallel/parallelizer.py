@@ -69,6 +69,57 @@ class LinearParallelizer(ModuleParallelizer):
         return module
 
 
+class Conv1DParallelizer(ModuleParallelizer):
+    """HF GPT-2 ``Conv1D`` (weight stored [in, out] — transposed vs
+    nn.Linear).  Converted in place to Column/RowParallelLinear with the
+    weight transposed and sliced.  Fused ``c_attn`` slices each of the n
+    stacked blocks separately: GPT-2 lays q|k|v out BLOCKWISE along the
+    output dim, so a plain chunk would give rank 0 all of Q (the reference
+    never supported Conv1D models at all)."""
+
+    @staticmethod
+    def is_parallelizable(module_name, module):
+        return (type(module).__name__ == "Conv1D" and hasattr(module, "nf")
+                and (TensorParallelMapping.is_column_parallel(module_name)
+                     or TensorParallelMapping.is_row_parallel(module_name)
+                     or TensorParallelMapping.is_fused_column(module_name)))
+
+    def parallelize(self):
+        module, pc = self.module, self.parallel_context
+        w = module.weight.data.t().contiguous()   # -> [out, in]
+        b = module.bias.data if getattr(module, "bias", None) is not None \
+            else None
+        if TensorParallelMapping.is_fused_column(self.module_name):
+            n = TensorParallelMapping.get_fused_count(self.module_name)
+            out = w.shape[0]
+            assert out % n == 0
+            blocks = w.view(n, out // n, w.shape[1])
+            module.weight = nn.Parameter(torch.cat(
+                [get_partition(blk, 0, pc) for blk in blocks], dim=0))
+            if b is not None:
+                bb = b.view(n, out // n)
+                module.bias = nn.Parameter(torch.cat(
+                    [get_partition(x, 0, pc) for x in bb], dim=0))
+            module.__class__ = ColumnParallelLinear
+            module.gather_output = False
+            module.out_features = module.weight.shape[0]
+        elif TensorParallelMapping.is_column_parallel(self.module_name):
+            module.weight = nn.Parameter(get_partition(w, 0, pc))
+            if b is not None:
+                module.bias = nn.Parameter(get_partition(b, 0, pc))
+            module.__class__ = ColumnParallelLinear
+            module.gather_output = False
+            module.out_features = module.weight.shape[0]
+        else:  # row-parallel: split the input dim, bias kept whole
+            module.weight = nn.Parameter(get_partition(w, 1, pc))
+            if b is not None:
+                module.bias = nn.Parameter(b.clone())
+            module.__class__ = RowParallelLinear
+            module.in_features = module.weight.shape[1]
+        module.parallel_context = pc
+        return module
+
+
 class EmbeddingParallelizer(ModuleParallelizer):
     @staticmethod
     def is_parallelizable(module_name, module):

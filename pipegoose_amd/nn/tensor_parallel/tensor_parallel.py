@@ -11,6 +11,7 @@ from torch import nn
 from pipegoose_amd.distributed.parallel_mode import ParallelMode
 from pipegoose_amd.nn.parallel import Parallel
 from pipegoose_amd.nn.tensor_parallel.parallelizer import (
+    Conv1DParallelizer,
     EmbeddingParallelizer,
     LayerNormParallelizer,
     LinearParallelizer,
@@ -21,6 +22,7 @@ PARALLELIZERS = [
     EmbeddingParallelizer,
     LMHeadParallelizer,   # before LinearParallelizer: lm_head is also a Linear
     LinearParallelizer,
+    Conv1DParallelizer,   # GPT-2 family (transposed-weight Conv1D)
     LayerNormParallelizer,
 ]
 
@@ -53,7 +55,8 @@ class TensorParallel(Parallel):
                           for c in mod.children())
             if not has_cpl:
                 continue
-            for attr in ("num_heads", "num_attention_heads", "num_key_value_heads"):
+            for attr in ("num_heads", "num_attention_heads",
+                         "num_key_value_heads", "split_size", "all_head_size"):
                 val = getattr(mod, attr, None)
                 if isinstance(val, int) and val % tp == 0 and val >= tp:
                     setattr(mod, attr, val // tp)

@@ -207,3 +207,43 @@ def _run_tp2_hf_albert(rank, world_size, port):
 
 def test_tensor_parallel_hf_albert_tp2():
     spawn(_run_tp2_hf_albert, world_size=2)
+
+
+def _hf_gpt2():
+    from transformers import GPT2Config, GPT2LMHeadModel
+    cfg = GPT2Config(vocab_size=256, n_positions=64, n_embd=64, n_layer=2,
+                     n_head=4)
+    torch.manual_seed(95)
+    return GPT2LMHeadModel(cfg)
+
+
+def _run_tp2_hf_gpt2(rank, world_size, port):
+    """GPT-2 family: Conv1D (transposed weights) + blockwise-fused c_attn —
+    capability the reference never had (its parallelizer only knew
+    nn.Linear).  TP2 logits must match the unparallelized model."""
+    ctx = init_parallel_context(rank, world_size, port, tensor_parallel_size=2)
+    model = _hf_gpt2().eval()  # gpt2 defaults to dropout 0.1 — disable
+    ref = _hf_gpt2().eval()
+    torch.manual_seed(96)
+    ids = torch.randint(0, 256, (2, 12))
+
+    with torch.no_grad():
+        ref_logits = ref(ids).logits
+
+    model = TensorParallel(model, ctx).parallelize()
+    with torch.no_grad():
+        out = model(ids).logits
+    if out.size(-1) == ref_logits.size(-1) // 2:  # sharded lm_head
+        import torch.distributed as dist
+        from pipegoose_amd.distributed.parallel_mode import ParallelMode
+        shards = [torch.empty_like(out) for _ in range(2)]
+        dist.all_gather(shards, out.contiguous(),
+                        group=ctx.get_group(ParallelMode.TENSOR))
+        out = torch.cat(shards, dim=-1)
+    assert torch.allclose(out, ref_logits, atol=1e-4), \
+        (out - ref_logits).abs().max()
+    ctx.destroy()
+
+
+def test_tensor_parallel_hf_gpt2_tp2():
+    spawn(_run_tp2_hf_gpt2, world_size=2)
