@@ -390,3 +390,43 @@ def test_interleaved_pp2_v3():
 
 def test_interleaved_pp4_v2():
     spawn(run_interleaved_deep, world_size=4, v=2, m=8, n_layers=8)
+
+
+def run_interleaved_bloom(rank, world_size, port):
+    """Interleaved pp2 x v2 over the native BLOOM model (structural
+    partitioner must produce 4 balanced chunks) — loss parity with the
+    single-process model."""
+    from pipegoose_amd.models.bloom import BloomForCausalLM, bloom_tiny
+    from pipegoose_amd.nn.pipeline_parallel import PipelineParallel
+
+    import dataclasses
+    ctx = init_parallel_context(rank, world_size, port,
+                                pipeline_parallel_size=world_size)
+    cfg = dataclasses.replace(bloom_tiny(), n_layer=4)  # >= p*v blocks
+    torch.manual_seed(124)
+    model = BloomForCausalLM(cfg, ctx)
+    ref_state = {k: v.clone() for k, v in model.state_dict().items()}
+
+    def lm_loss(logits, labels):
+        import torch.nn.functional as TF
+        return TF.cross_entropy(
+            logits[:, :-1].reshape(-1, logits.size(-1)).float(),
+            labels[:, 1:].reshape(-1))
+
+    torch.manual_seed(8)
+    ids = torch.randint(0, 256, (4, 16))
+    pp = PipelineParallel(model, ctx, n_microbatches=2,
+                          schedule="interleaved", virtual_stages=2,
+                          loss_fn=lm_loss).parallelize()
+    loss = pp(ids, ids)
+    if rank == world_size - 1:
+        torch.manual_seed(124)
+        ref = BloomForCausalLM(cfg, ctx)
+        ref.load_state_dict(ref_state)
+        ref_loss = lm_loss(ref(ids), ids)
+        assert torch.allclose(loss, ref_loss, atol=1e-5), (loss, ref_loss)
+    ctx.destroy()
+
+
+def test_interleaved_pp2_v2_bloom():
+    spawn(run_interleaved_bloom, world_size=2)
