@@ -32,6 +32,9 @@ def parse_args():
     p.add_argument("--tp", type=int, default=0, help="0 = auto by world size")
     p.add_argument("--pp", type=int, default=0)
     p.add_argument("--dp", type=int, default=0)
+    p.add_argument("--virtual-stages", type=int, default=1,
+                   help=">1 switches pp to the interleaved 1F1B schedule "
+                        "(bubble (pp-1)/(v*m); experimental on RCCL)")
     p.add_argument("--microbatches", type=int, default=8,
                    help="pipeline microbatches (1F1B bubble = (pp-1)/(m+pp-1): "
                         "m=8 at pp=2 -> 11%% vs 20%% at m=4; per-rank "
@@ -113,8 +116,10 @@ def main():
             parallel_context=ctx).parallelize()
         moe_loss_wrap = ExpertLoss(lambda loss: loss)
     if pp > 1:
+        v = max(1, args.virtual_stages)
         model = PipelineParallel(
             model, ctx, n_microbatches=args.microbatches,
+            schedule="interleaved" if v > 1 else "1f1b", virtual_stages=v,
             loss_fn=make_causal_lm_loss(ctx)).parallelize()
     model = model.to(device=device, dtype=dtype)
     if dp > 1:
