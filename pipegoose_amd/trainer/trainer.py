@@ -110,6 +110,18 @@ class Trainer:
                 + (f" tokens/s {tps:,.0f}" if not math.isnan(tps) else ""))
         return val
 
+    def resume_from(self, ckpt_dir: str) -> dict:
+        """Restore model weights + optimizer/scheduler/step state written by
+        ``CheckpointCallback`` (or save_pretrained/save_training_state) and
+        continue counting from the saved step."""
+        from pipegoose_amd.nn.utils import from_pretrained, load_training_state
+        from_pretrained(self.model, ckpt_dir, parallel_context=self.ctx)
+        payload = load_training_state(self.optimizer, ckpt_dir,
+                                      parallel_context=self.ctx,
+                                      lr_scheduler=self.lr_scheduler)
+        self.state.global_step = payload["step"]
+        return payload
+
     # ------------------------------------------------------------------- fit
 
     def fit(self, train_loader: Iterable, epochs: int = 1,
