@@ -75,7 +75,10 @@ class InterleavedPipelineEngine:
 
     def __init__(self, chunks: List[torch.nn.Module],
                  parallel_context: ParallelContext, n_microbatches: int,
-                 loss_fn=None):
+                 loss_fn=None, moe_aux_weight: float = 0.01,
+                 moe_z_weight: float = 0.1):
+        self.moe_aux_weight = moe_aux_weight
+        self.moe_z_weight = moe_z_weight
         self.chunks = chunks
         self.v = len(chunks)
         self.pc = parallel_context
@@ -95,6 +98,22 @@ class InterleavedPipelineEngine:
         self._shape = None
         self._dtype = None
 
+    def _pop_moe_losses(self):
+        from pipegoose_amd.nn.expert_parallel import ExpertContext
+        ectx = ExpertContext.get_instance()
+        aux = ectx.pop_all_aux_loss()
+        zl = ectx.pop_all_z_loss()
+        if not aux and not zl:
+            return None
+        total = None
+        for a in aux:
+            t = a * self.moe_aux_weight
+            total = t if total is None else total + t
+        for z in zl:
+            t = z * self.moe_z_weight
+            total = t if total is None else total + t
+        return total
+
     # stage-role helpers -----------------------------------------------------
 
     def _is_first_stage(self, c: int) -> bool:
@@ -113,6 +132,10 @@ class InterleavedPipelineEngine:
 
         saved_in = [[None] * m for _ in range(v)]
         saved_out = [[None] * m for _ in range(v)]
+        # MoE router aux/z losses share graph nodes with the chunk output —
+        # snapshot per (chunk, mb) at forward, ride the SAME backward call
+        # (identical treatment to PipelineEngine, engine.py)
+        self._saved_moe = [[None] * m for _ in range(v)]
         losses: List[torch.Tensor] = []
         outputs: List[torch.Tensor] = []
         pending = []
@@ -170,6 +193,7 @@ class InterleavedPipelineEngine:
         saved_in[c][mb] = x
         out = self.chunks[c](x)
         saved_out[c][mb] = out
+        self._saved_moe[c][mb] = self._pop_moe_losses()
         if not self._is_last_stage(c):
             self._send_act(out, pending)
         elif self.loss_fn is None or label_mbs[mb] is None:
@@ -178,15 +202,23 @@ class InterleavedPipelineEngine:
     def _backward(self, c, mb, saved_in, saved_out, label_mbs, losses,
                   pending):
         out = saved_out[c][mb]
+        moe = self._saved_moe[c][mb]
+        self._saved_moe[c][mb] = None
         if self._is_last_stage(c):
             if self.loss_fn is not None and label_mbs[mb] is not None:
                 loss = self.loss_fn(out, label_mbs[mb]) / self.m
                 losses.append(loss.detach())
+                if moe is not None and moe.requires_grad:
+                    loss = loss + moe / self.m
                 loss.backward()
         else:
             grad = self.p2p.recv_activation(tuple(out.shape), out.dtype,
                                             self.next_rank, tag=1)
-            torch.autograd.backward(out, grad_tensors=grad)
+            if moe is not None and moe.requires_grad:
+                torch.autograd.backward([out, moe / self.m],
+                                        grad_tensors=[grad, None])
+            else:
+                torch.autograd.backward(out, grad_tensors=grad)
         x = saved_in[c][mb]
         if not self._is_first_stage(c) and x is not None and x.grad is not None:
             pending.append(self.p2p.send_activation(x.grad, self.prev_rank,

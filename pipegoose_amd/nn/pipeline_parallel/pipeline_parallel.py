@@ -92,7 +92,8 @@ class PipelineParallel(Parallel):
             chunks = [stages[c * pp + rank] for c in range(v)]
             engine = InterleavedPipelineEngine(
                 chunks, self.parallel_context, self.n_microbatches,
-                loss_fn=self.loss_fn)
+                loss_fn=self.loss_fn, moe_aux_weight=self.moe_aux_weight,
+                moe_z_weight=self.moe_z_weight)
             wrapped = InterleavedStageModule(engine)
             self._save_metadata(wrapped, self.parallel_context)
             return wrapped

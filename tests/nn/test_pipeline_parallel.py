@@ -188,7 +188,7 @@ def test_1f1b_edge_schedules():
 
 # ------------------------------------------------------------------ MoE + PP
 
-def run_pp_moe(rank, world_size, port):
+def run_pp_moe(rank, world_size, port, schedule="1f1b", v=1):
     """MoE layers on BOTH stages: the engine must fold each stage's router
     aux/z losses into that microbatch's backward (ExpertLoss on the last
     stage can never see stage-0's routers).  Oracle: per-microbatch loop on
@@ -227,6 +227,7 @@ def run_pp_moe(rank, world_size, port):
     ExpertContext.get_instance().pop_all_z_loss()
 
     pp = PipelineParallel(model, ctx, n_microbatches=m, loss_fn=_loss_fn,
+                          schedule=schedule, virtual_stages=v,
                           moe_aux_weight=AUX_W, moe_z_weight=Z_W).parallelize()
     loss = pp(x, target)
     # engine must leave the context drained
@@ -247,11 +248,17 @@ def run_pp_moe(rank, world_size, port):
         ref_loss = torch.stack(ref_losses).sum()
         assert torch.allclose(loss, ref_loss, atol=1e-6), (loss, ref_loss)
 
-    stages = UniformPartitioner(ref, ctx).split(world_size)
+    n_parts = world_size * v
+    stages = UniformPartitioner(ref, ctx).split(n_parts)
     own_params = dict(pp.named_parameters())
     router_grads = 0
-    for name, p_ref in stages[rank].named_parameters():
-        p = own_params["stage." + name]
+    pairs = []
+    for c in range(v):
+        prefix = f"chunks.{c}." if v > 1 else "stage."
+        idx = c * world_size + rank if v > 1 else rank
+        pairs += [(prefix + n, pr) for n, pr in stages[idx].named_parameters()]
+    for name, p_ref in pairs:
+        p = own_params[name]
         if p_ref.grad is None:  # expert that received zero tokens
             assert p.grad is None or p.grad.abs().max() == 0, name
             continue
@@ -261,12 +268,18 @@ def run_pp_moe(rank, world_size, port):
         if "router.gate" in name:
             router_grads += 1
             assert p.grad.abs().sum() > 0, f"router grad zero: {name}"
-    assert router_grads > 0, f"stage {rank} holds no router (bad split)"
+    if v == 1:  # with v>1 chunks a rank may legitimately hold no router;
+        # the full-grad parity loop above already covers router params
+        assert router_grads > 0, f"stage {rank} holds no router (bad split)"
     ctx.destroy()
 
 
 def test_pp2_moe_aux_losses_reach_both_stages():
     spawn(run_pp_moe, world_size=2)
+
+
+def test_pp2_moe_interleaved():
+    spawn(run_pp_moe, world_size=2, schedule="interleaved", v=2)
 
 
 # ---------------------------------------------------- interleaved 1F1B (v>1)
