@@ -113,3 +113,38 @@ def run_tp_pp_sp(rank, world_size, port):
 
 def test_tp2_pp2_sequence_parallel_matches():
     spawn(run_tp_pp_sp, world_size=4)
+
+
+def run_dp_interleaved(rank, world_size, port):
+    """Interleaved PP2(v2) x DP2: deferred DP sync at the engine tail keeps
+    replicas identical after a step (mirror of run_dp_pp)."""
+    import dataclasses
+    from pipegoose_amd.distributed.parallel_mode import ParallelMode
+
+    ctx = init_parallel_context(rank, world_size, port, pipeline_parallel_size=2)
+    cfg = dataclasses.replace(bloom_tiny(), n_layer=4)
+    torch.manual_seed(12)
+    model = BloomForCausalLM(cfg, ctx)
+    pp = PipelineParallel(model, ctx, n_microbatches=2,
+                          schedule="interleaved", virtual_stages=2,
+                          loss_fn=lm_loss).parallelize()
+    pp = DataParallel(pp, ctx).parallelize()
+    optim = DistributedOptimizer(torch.optim.Adam(pp.parameters(), lr=1e-3), ctx)
+
+    dp_rank = ctx.get_local_rank(ParallelMode.DATA)
+    for step in range(2):
+        torch.manual_seed(910 + step * 10 + dp_rank)
+        ids = torch.randint(0, 256, (4, 16))
+        optim.zero_grad()
+        pp(ids, ids)
+        optim.step()
+
+    flat = torch.cat([p.detach().reshape(-1) for p in pp.parameters()])
+    peers = [torch.empty_like(flat) for _ in range(2)]
+    torch.distributed.all_gather(peers, flat, group=ctx.get_group(ParallelMode.DATA))
+    assert torch.allclose(peers[0], peers[1], atol=1e-6)
+    ctx.destroy()
+
+
+def test_interleaved_pp2_dp2_replicas_stay_synced():
+    spawn(run_dp_interleaved, world_size=4)
