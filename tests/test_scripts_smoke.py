@@ -31,3 +31,8 @@ def test_decode_bench_tool():
                 "--batch", "2", "--prompt-len", "16", "--new-tokens", "4",
                 "--graph"])
     assert "graph decode" in out
+
+
+def test_moe_training_example():
+    out = _run(["examples/moe_training.py", "--tiny", "--steps", "2"])
+    assert out.strip(), "no output"
