@@ -443,3 +443,16 @@ def run_interleaved_bloom(rank, world_size, port):
 
 def test_interleaved_pp2_v2_bloom():
     spawn(run_interleaved_bloom, world_size=2)
+
+
+def test_interleaved_guard_rails():
+    """Config errors fail fast with clear messages."""
+    from pipegoose_amd.nn.pipeline_parallel.interleaved import _fwd_seq
+
+    with pytest.raises(AssertionError):
+        _fwd_seq(2, 3, 2)  # m % p != 0
+
+    from pipegoose_amd.nn.pipeline_parallel import PipelineParallel
+    with pytest.raises(AssertionError):
+        PipelineParallel(torch.nn.Linear(2, 2), None,
+                         schedule="interleaved", virtual_stages=1)
