@@ -90,3 +90,48 @@ def _run_tp_dp_zero(rank, world_size, port):
 
 def test_tp2_dp2_zero1_convergence_and_replica_sync():
     spawn(_run_tp_dp_zero, world_size=4)
+
+
+def _run_moe_overfit(rank, world_size, port):
+    """MoE-ified BLOOM (the reference's run_ep.py convergence scenario,
+    tests/convergence/run_ep.py): ExpertParallel surgery + ExpertLoss must
+    train — loss halves on a fixed batch and the router gates move."""
+    from pipegoose_amd.nn import ExpertParallel
+    from pipegoose_amd.nn.expert_parallel import (ExpertLoss,
+                                                  SwitchNoisePolicy,
+                                                  Top1Router)
+
+    ctx = init_parallel_context(rank, world_size, port)
+    cfg = bloom_tiny()
+    torch.manual_seed(0)
+    model = BloomForCausalLM(cfg, ctx)
+    model = ExpertParallel(
+        model, 4,
+        router=Top1Router(SwitchNoisePolicy(), 4, cfg.hidden_size),
+        parallel_context=ctx).parallelize()
+    opt = torch.optim.Adam(model.parameters(), lr=3e-3)
+    loss_fn = ExpertLoss(lm_loss)
+    torch.manual_seed(1)
+    ids = torch.randint(0, 256, (4, 16))
+
+    gate0 = [m.router.gate.weight.detach().clone()
+             for m in model.modules() if hasattr(m, "router")]
+    first = last = None
+    for _ in range(25):
+        opt.zero_grad()
+        loss = loss_fn(model(ids), ids)
+        if first is None:
+            first = loss.item()
+        loss.backward()
+        opt.step()
+        last = loss.item()
+    assert last < 0.5 * first, (first, last)
+    gate1 = [m.router.gate.weight.detach()
+             for m in model.modules() if hasattr(m, "router")]
+    assert any(not torch.equal(a, b) for a, b in zip(gate0, gate1)), \
+        "router gates never updated"
+    ctx.destroy()
+
+
+def test_moe_overfit():
+    spawn(_run_moe_overfit, world_size=1)
