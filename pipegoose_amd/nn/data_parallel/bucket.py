@@ -89,6 +89,11 @@ class BucketManager:
         if bucket.is_full_with(numel):
             self._flush_key(key)
             bucket = self._get_bucket(key, param.grad.device, numel)
+            if bucket.size < numel:
+                # a param bigger than the bucket (e.g. a vocab embedding's
+                # grad is hundreds of MB) gets a dedicated full-size buffer
+                bucket = self.buckets[key] = Bucket(numel, key[1],
+                                                    param.grad.device)
         bucket.add_grad(param)
 
     def _flush_key(self, key):

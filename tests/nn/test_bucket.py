@@ -90,3 +90,31 @@ def _run_overflow_flush(rank, world_size, port):
 
 def test_bucket_manager_overflow_flush_dp2():
     spawn(_run_overflow_flush, world_size=2)
+
+
+def _run_param_bigger_than_bucket(rank, world_size, port):
+    """A single grad LARGER than the bucket itself (bloom-560m's embedding
+    grad is ~500 MB vs the 25 MB bucket) must get a dedicated buffer —
+    previously the post-flush bucket kept the old size and asserted."""
+    ctx = init_parallel_context(rank, world_size, port, data_parallel_size=2)
+    mgr = BucketManager(ctx, bucket_size_mb=1)  # 262144 fp32 elements
+    torch.manual_seed(30 + rank)
+    small = _param_with_grad((64,))
+    huge = _param_with_grad((400_000,))  # > bucket capacity
+    import torch.distributed as dist
+    expected = []
+    for p in (small, huge):
+        g = p.grad.clone()
+        dist.all_reduce(g)
+        expected.append(g / world_size)
+    mgr.add_param(small, ParallelMode.DATA)
+    mgr.add_param(huge, ParallelMode.DATA)
+    mgr.flush()
+    mgr.wait_all()
+    for p, e in zip((small, huge), expected):
+        assert torch.allclose(p.grad, e, atol=1e-6)
+    ctx.destroy()
+
+
+def test_bucket_param_bigger_than_bucket_dp2():
+    spawn(_run_param_bigger_than_bucket, world_size=2)
