@@ -67,3 +67,27 @@ def test_trainer_arms_watchdog_during_fit():
     tr.fit(batches, epochs=2)
     assert tr._watchdog is None  # stopped and cleared after fit
     assert tr.state.global_step == 2
+
+
+def test_throughput_meter_math():
+    import time as _time
+    from pipegoose_amd.trainer.logger import ThroughputMeter
+    m = ThroughputMeter(window=3)
+    assert m.tokens_per_sec != m.tokens_per_sec  # nan before 2 samples
+    for _ in range(4):
+        m.update(100)
+        _time.sleep(0.01)
+    # window of 3 stamps -> 2 intervals of 100 tokens over >= 0.02 s
+    tps = m.tokens_per_sec
+    assert 0 < tps < 200 / 0.02
+    assert len(m._stamps) == 3
+
+
+def test_distributed_logger_rank_prefix(capsys):
+    import logging
+    from pipegoose_amd.trainer.logger import DistributedLogger
+    lg = DistributedLogger(name="pg_test_logger", parallel_context=None,
+                          level=logging.INFO, rank_zero_only=False)
+    lg.info("hello")
+    lg.error("bad")
+    # with no context the prefix is empty but logging must not crash
