@@ -105,6 +105,57 @@ def from_pretrained(
     return module
 
 
+def consolidate_checkpoint(
+    ckpt_dir: str,
+    full_shapes: dict,
+    tp: int,
+    pp: int,
+) -> dict:
+    """Merge the per-(tp, pp) shard files into ONE full state_dict (CPU).
+
+    Beyond-reference capability: the reference's ``from_pretrained`` only
+    reloads into the exact topology that saved (nn/utils.py:31-50); this
+    reconstructs the unsharded weights so a checkpoint can be exported, or
+    loaded at a different parallelism degree.
+
+    ``full_shapes``: {param_name: torch.Size} of the UNSHARDED model —
+    ``{k: v.shape for k, v in reference_model.state_dict().items()}`` from a
+    tp=1/pp=1 instance.  The shard dim per tensor is inferred by shape
+    comparison (the dim where shard * tp == full); equal shapes mean the
+    tensor is replicated and rank 0's copy is taken.  pp shards partition
+    NAMES, so files are unioned across pp ranks.
+    """
+    # name -> [tensor per tp_rank] (pp files partition names: union)
+    collected: dict = {}
+    for pp_rank in range(pp):
+        for tp_rank in range(tp):
+            path = os.path.join(
+                ckpt_dir, CHECKPOINT_WEIGHTS_NAME.format(tp_rank, pp_rank))
+            state = torch.load(path, map_location="cpu", weights_only=True)
+            for name, t in state.items():
+                collected.setdefault(name, [None] * tp)[tp_rank] = t
+
+    out = {}
+    for name, shards in collected.items():
+        assert name in full_shapes, f"unexpected param in checkpoint: {name}"
+        full = tuple(full_shapes[name])
+        first = shards[0]
+        if tuple(first.shape) == full:
+            out[name] = first  # replicated (e.g. LayerNorm, row-linear bias)
+            continue
+        dims = [d for d in range(first.dim())
+                if first.shape[d] * tp == full[d]
+                and all(first.shape[i] == full[i]
+                        for i in range(first.dim()) if i != d)]
+        assert len(dims) == 1, \
+            f"{name}: cannot infer shard dim ({tuple(first.shape)} vs {full})"
+        assert all(s is not None for s in shards), f"{name}: missing tp shard"
+        out[name] = torch.cat(shards, dim=dims[0])
+    missing = set(full_shapes) - set(out)
+    assert not missing, f"params absent from checkpoint: {sorted(missing)[:5]}"
+    return out
+
+
 def save_training_state(
     optim,
     ckpt_dir: str = "./",
