@@ -156,6 +156,39 @@ def consolidate_checkpoint(
     return out
 
 
+def load_full_state(
+    module: nn.Module,
+    full_state: dict,
+    parallel_context: Optional[ParallelContext] = None,
+) -> nn.Module:
+    """Load an UNSHARDED state dict into an already-parallelized module:
+    each param takes its tp slice (shard dim inferred by shape — the dim
+    where local * tp == full); under pp the module naturally requests only
+    its own stage's names.  Inverse of ``consolidate_checkpoint`` — enables
+    tp1 → tpN warm starts the reference had no path for."""
+    ctx = parallel_context or ParallelContext.get_context()
+    assert ctx is not None, "load_full_state needs a ParallelContext"
+    tp = ctx.get_world_size(ParallelMode.TENSOR)
+    rank = ctx.get_local_rank(ParallelMode.TENSOR)
+    new_state = {}
+    for name, local in module.state_dict().items():
+        assert name in full_state, f"missing from full state: {name}"
+        full = full_state[name]
+        if tuple(local.shape) == tuple(full.shape):
+            new_state[name] = full
+            continue
+        dims = [d for d in range(local.dim())
+                if local.shape[d] * tp == full.shape[d]
+                and all(local.shape[i] == full.shape[i]
+                        for i in range(local.dim()) if i != d)]
+        assert len(dims) == 1, \
+            f"{name}: cannot infer shard dim ({tuple(local.shape)} vs " \
+            f"{tuple(full.shape)})"
+        new_state[name] = full.chunk(tp, dim=dims[0])[rank].contiguous()
+    module.load_state_dict(new_state)
+    return module
+
+
 def save_training_state(
     optim,
     ckpt_dir: str = "./",

@@ -145,3 +145,46 @@ def test_consolidate_tp2_checkpoint_into_full_model(tmp_path):
     model's logits at tp=1."""
     spawn(_run_save_tp2_phase, world_size=2, tmpdir=str(tmp_path))
     spawn(_run_consolidate_phase, world_size=1, tmpdir=str(tmp_path))
+
+
+def _run_export_full_phase(rank, world_size, port, tmpdir):
+    from pipegoose_amd.models.bloom import BloomForCausalLM, bloom_tiny
+
+    ctx = init_parallel_context(rank, world_size, port)  # tp=1
+    torch.manual_seed(88)
+    model = BloomForCausalLM(bloom_tiny(), ctx).eval()
+    torch.manual_seed(89)
+    ids = torch.randint(0, 256, (2, 10))
+    with torch.no_grad():
+        logits = model(ids)
+    torch.save({"state": model.state_dict(), "ids": ids, "logits": logits},
+               os.path.join(tmpdir, "full.pt"))
+    ctx.destroy()
+
+
+def _run_load_sharded_phase(rank, world_size, port, tmpdir):
+    import torch.distributed as dist
+    from pipegoose_amd.distributed.parallel_mode import ParallelMode
+    from pipegoose_amd.models.bloom import BloomForCausalLM, bloom_tiny
+    from pipegoose_amd.nn.utils import load_full_state
+
+    ctx = init_parallel_context(rank, world_size, port, tensor_parallel_size=2)
+    model = BloomForCausalLM(bloom_tiny(), ctx).eval()  # random init
+    blob = torch.load(os.path.join(tmpdir, "full.pt"), weights_only=True)
+    load_full_state(model, blob["state"], parallel_context=ctx)
+    with torch.no_grad():
+        local = model(blob["ids"])
+    shards = [torch.empty_like(local) for _ in range(2)]
+    dist.all_gather(shards, local.contiguous(),
+                    group=ctx.get_group(ParallelMode.TENSOR))
+    full = torch.cat(shards, dim=-1)
+    assert torch.allclose(full, blob["logits"], atol=1e-5), \
+        (full - blob["logits"]).abs().max()
+    ctx.destroy()
+
+
+def test_load_full_state_into_tp2(tmp_path):
+    """tp1 -> tp2 warm start: a full state dict slices into the sharded
+    model and reproduces the original logits."""
+    spawn(_run_export_full_phase, world_size=1, tmpdir=str(tmp_path))
+    spawn(_run_load_sharded_phase, world_size=2, tmpdir=str(tmp_path))
