@@ -331,3 +331,47 @@ def test_hf_bloom_weights_into_native_tp1():
 
 def test_hf_bloom_weights_into_native_tp2():
     spawn(_run_hf_weights_into_native, world_size=2)
+
+
+def _run_hf_llama_weights_into_native(rank, world_size, port):
+    """HF LlamaForCausalLM weights load unmodified into the native llama
+    (names, shapes, half-split RoPE convention all align) — incl. GQA."""
+    import dataclasses
+    from transformers import LlamaConfig as HFConfig
+    from transformers import LlamaForCausalLM as HFLlama
+    from pipegoose_amd.models.llama import LlamaConfig, LlamaForCausalLM
+    from pipegoose_amd.nn.utils import load_full_state
+
+    ctx = init_parallel_context(rank, world_size, port,
+                                tensor_parallel_size=world_size)
+    torch.manual_seed(5)
+    hf = HFLlama(HFConfig(vocab_size=512, hidden_size=64,
+                          intermediate_size=128, num_hidden_layers=2,
+                          num_attention_heads=4, num_key_value_heads=2,
+                          attn_implementation="eager")).eval()
+    native = LlamaForCausalLM(
+        LlamaConfig(vocab_size=512, hidden_size=64, intermediate_size=128,
+                    n_layer=2, n_head=4, n_kv_head=2), ctx).eval()
+    load_full_state(native, hf.state_dict(), parallel_context=ctx)
+    torch.manual_seed(6)
+    ids = torch.randint(0, 512, (2, 10))
+    with torch.no_grad():
+        ref = hf(ids).logits
+        out = native(ids)
+    if world_size > 1:
+        import torch.distributed as dist
+        from pipegoose_amd.distributed.parallel_mode import ParallelMode
+        shards = [torch.empty_like(out) for _ in range(world_size)]
+        dist.all_gather(shards, out.contiguous(),
+                        group=ctx.get_group(ParallelMode.TENSOR))
+        out = torch.cat(shards, dim=-1)
+    assert torch.allclose(out, ref, atol=1e-4), (out - ref).abs().max()
+    ctx.destroy()
+
+
+def test_hf_llama_weights_into_native_tp1():
+    spawn(_run_hf_llama_weights_into_native, world_size=1)
+
+
+def test_hf_llama_weights_into_native_tp2():
+    spawn(_run_hf_llama_weights_into_native, world_size=2)
