@@ -7,8 +7,8 @@ RCCL; this script reads RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from the env.
 Weak scaling: per-GPU microbatch fixed as N grows.
 
 Measures BASELINE.json's metric: tokens/sec (whole node), BLOOM-560M TP2xDP2
-(and BLOOM-7B1 TP2xPP2xDP2 via --model bloom-7b1 once PP lands in the bench
-path), bf16, synthetic data, random-init weights.
+(and BLOOM-7B1 TP2xPP2xDP2 via --model bloom-7b1, 1F1B pipeline + ZeRO-1),
+bf16, synthetic data, random-init weights.
 """
 import argparse
 import json

@@ -92,7 +92,7 @@ class LlamaAttention(nn.Module):
         key = (S, device, dtype)
         if key not in self._mask_cache:
             m = torch.triu(torch.full((S, S), float("-inf"), device=device), 1)
-            self._mask_cache = {key: m.to(dtype)[None, None]}
+            self._mask_cache[key] = m.to(dtype)[None, None]
         return self._mask_cache[key]
 
     def forward(self, hidden: torch.Tensor, past_kv=None,

@@ -7,9 +7,13 @@ all-reduce; expert params reduce over EXPERT_DATA).  MI355X redesign:
     with compute on its own HIP stream;
   - post-divide (sum, then /dp) instead of the reference's bf16-lossy
     pre-divide;
-  - completion is counter-triggered: when the last grad hook fires, the tail
-    buckets flush and all works are waited on, so ``loss.backward()`` returns
-    with gradients synchronized (no explicit user call needed).
+  - completion rides autograd's end-of-backward callback queue (the DDP
+    reducer idiom): the first grad hook of a backward pass queues a callback
+    that flushes the tail buckets and waits on all works, so
+    ``loss.backward()`` returns with gradients synchronized even when some
+    hooked params produced no grad this pass (e.g. a conditionally-executed
+    branch).  A fired-hook counter would go stale in that case and either
+    skip the flush or fire it mid-backward next step.
 """
 import torch
 from torch import nn
@@ -25,7 +29,7 @@ class DataParallel(Parallel):
         super().__init__(module, parallel_context)
         self._bucket_manager = BucketManager(parallel_context)
         self._hooked_params = []
-        self._fired = 0
+        self._callback_queued = False
         # False during pipeline microbatch accumulation; engine calls
         # sync_now() after the last microbatch.
         self.sync_enabled = True
@@ -63,11 +67,15 @@ class DataParallel(Parallel):
         mode = ParallelMode.EXPERT_DATA if getattr(param, "is_expert", False) \
             else ParallelMode.DATA
         self._bucket_manager.add_param(param, mode)
-        self._fired += 1
-        if self._fired == len(self._hooked_params):
-            self.finish_gradient_sync()
+        if not self._callback_queued:
+            self._callback_queued = True
+            torch.autograd.Variable._execution_engine.queue_callback(
+                self._end_of_backward)
+
+    def _end_of_backward(self):
+        self._callback_queued = False
+        self.finish_gradient_sync()
 
     def finish_gradient_sync(self):
         self._bucket_manager.flush()
         self._bucket_manager.wait_all()
-        self._fired = 0

@@ -75,14 +75,17 @@ class Experts(nn.Module):
         for local_idx, expert in enumerate(self.experts):
             global_idx = self.expert_offset + local_idx
             token_mask = dispatch_order.reshape(-1) == global_idx
-            if token_mask.any():
-                selected = flat[token_mask]
-                # experts receive ONLY their tokens; block-level extras such
-                # as HF Bloom's residual are handled by ExpertLayer
-                expert_out = expert(selected)
-                if isinstance(expert_out, tuple):
-                    expert_out = expert_out[0]
-                outputs[token_mask] = expert_out.to(outputs.dtype)
+            # A zero-token expert still runs (on a 0-row batch) so every
+            # expert param gets a grad every step: DP ranks route different
+            # tokens, and a rank that skipped an expert its EXPERT_DATA peer
+            # ran would launch a different bucket-reduce sequence → hang.
+            selected = flat[token_mask]
+            # experts receive ONLY their tokens; block-level extras such
+            # as HF Bloom's residual are handled by ExpertLayer
+            expert_out = expert(selected)
+            if isinstance(expert_out, tuple):
+                expert_out = expert_out[0]
+            outputs[token_mask] = expert_out.to(outputs.dtype)
         if self.enable_tensor_parallel:
             outputs = _AllReduceCombine.apply(outputs, self.parallel_context)
         return outputs.reshape(shape)

@@ -44,15 +44,28 @@ def _detach_to_cpu(state_dict: dict) -> dict:
     return out
 
 
+def _atomic_save(obj, path: str):
+    """Write to a sibling temp file and rename: a reader never observes a
+    half-written shard, and a crash mid-save leaves the old file intact."""
+    tmp = path + ".tmp"
+    torch.save(obj, tmp)
+    os.replace(tmp, path)
+
+
 class _AsyncWriter:
-    """One writer thread per process; serializes queued torch.save calls."""
+    """One writer thread per process; serializes queued torch.save calls.
+    Writes are atomic (temp + rename) and joined at interpreter exit so a
+    process that never calls ``wait_for_async_saves`` still can't leave a
+    truncated shard behind."""
 
     def __init__(self):
         self._thread: Optional[threading.Thread] = None
+        import atexit
+        atexit.register(self.wait)
 
     def submit(self, obj, path: str):
         self.wait()
-        t = threading.Thread(target=torch.save, args=(obj, path), daemon=True)
+        t = threading.Thread(target=_atomic_save, args=(obj, path), daemon=False)
         t.start()
         self._thread = t
 
@@ -139,6 +152,9 @@ def consolidate_checkpoint(
     for name, shards in collected.items():
         assert name in full_shapes, f"unexpected param in checkpoint: {name}"
         full = tuple(full_shapes[name])
+        assert all(s is not None for s in shards), \
+            f"{name}: missing tp shard(s) " \
+            f"{[i for i, s in enumerate(shards) if s is None]}"
         first = shards[0]
         if tuple(first.shape) == full:
             out[name] = first  # replicated (e.g. LayerNorm, row-linear bias)
@@ -149,7 +165,6 @@ def consolidate_checkpoint(
                         for i in range(first.dim()) if i != d)]
         assert len(dims) == 1, \
             f"{name}: cannot infer shard dim ({tuple(first.shape)} vs {full})"
-        assert all(s is not None for s in shards), f"{name}: missing tp shard"
         out[name] = torch.cat(shards, dim=dims[0])
     missing = set(full_shapes) - set(out)
     assert not missing, f"params absent from checkpoint: {sorted(missing)[:5]}"
@@ -227,8 +242,11 @@ def load_training_state(
     bookkeeping dict ({step, extra, topology})."""
     ctx = parallel_context or ParallelContext.get_context()
     assert ctx is not None
+    # weights_only=True everywhere: optimizer/scheduler state_dicts are plain
+    # tensors/dicts/primitives, so resuming from an untrusted directory can't
+    # execute code (ADVICE r1).
     payload = torch.load(_optim_path(ckpt_dir, ctx), map_location="cpu",
-                         weights_only=False)
+                         weights_only=True)
     topo = payload["topology"]
     assert topo["tp"] == ctx.get_world_size(ParallelMode.TENSOR), \
         f"checkpoint tp={topo['tp']} != runtime tp"
