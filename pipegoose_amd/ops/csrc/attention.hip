@@ -273,6 +273,335 @@ __global__ void mfma_probe_kernel(const bf16* __restrict__ A,
 
 }  // namespace
 
+// ======================================================================
+// Forward v2 — the round-2 speed-of-light ladder (docs/KERNEL_PLAN_R2.md):
+//   * 8 waves x 32 q rows, v_mfma_f32_32x32x16_bf16 (guide §B 8-warp ladder)
+//   * swapped QK^T (St = K·Q^T): the P row for q-row lane&31 is lane-local
+//     (32 f32 regs), so softmax is in-register — no p_lds bounce, no
+//     __shfl tree (T12: cross-half combine = one permlane32_swap)
+//   * swapped PV too (O^T = V^T·P^T): the O accumulator is ALSO per-lane
+//     q-row lane&31, so the online-softmax rescale is a scalar broadcast
+//     multiply — no cross-lane redistribution of alpha
+//   * K and V^T LDS images on 256-B rows with the T2 XOR swizzle
+//     (byte ^= (row&15)<<4): the b128 fragment reads are conflict-free
+//     (v1's padded rows measured 33-63% LDS bank-conflict cycles)
+//   * double-buffered tiles with the T14 split: global loads for tile n+1
+//     issue before tile n's QK^T, the LDS write lands after softmax, so HBM
+//     latency hides under MFMA; one barrier per tile
+__device__ __forceinline__ float u2f(unsigned u) { return __uint_as_float(u); }
+
+namespace {
+
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+
+__device__ __forceinline__ f32x16 MFMA_32x32x16(frag_ab a, frag_ab b, f32x16 c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+}
+
+// Element index into a swizzled [row][128] bf16 image (256-B rows): the
+// byte offset within the row is XORed with (row&15)<<4, spreading a b128
+// lane group over all 16 slots of the 256-B bank row (T2 / G4).  The XOR
+// touches byte-address bits 4-7 only, so 16-B and 4-B alignment survive.
+__device__ __forceinline__ int swz128(int row, int elem_col) {
+    return row * 128 + ((((elem_col) << 1) ^ ((row & 15) << 4)) >> 1);
+}
+
+// C/D layout of mfma_f32_32x32x16_bf16: col = lane&31, row = (r&3) +
+// 8*(r>>2) + 4*(lane>>5), r in [0,16).  A/B fragments: i/j = lane&31,
+// k = 8*(lane>>5) + elem (verified by mfma_probe32).
+template <int D, int WAVES>
+__global__ __launch_bounds__(WAVES * WAVE_SIZE)
+void attn_fwd_v2_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
+                        const bf16* __restrict__ v,
+                        const float* __restrict__ slopes, float scale,
+                        bf16* __restrict__ o, float* __restrict__ lse,
+                        int B, int H, int S, int kv_off, int kv_group,
+                        int64_t qb, int64_t qh, int64_t qs,
+                        int64_t kb, int64_t kh, int64_t ks,
+                        int64_t vb, int64_t vh, int64_t vs,
+                        int64_t ob, int64_t oh, int64_t os) {
+    constexpr int KVB = 64;                 // kv rows per tile
+    constexpr int NT = WAVES * WAVE_SIZE;
+    constexpr int ROWS = WAVES * 32;        // q rows per workgroup
+    constexpr int DCH = D / 16;             // QK^T k-chunks (K=16 per mfma)
+    constexpr int DSUB = D / 32;            // O row subtiles (O^T layout)
+    constexpr int NPK = KVB * (D / 8);      // K tile 16-B packets
+    constexpr int NPV = (KVB / 2) * (D / 8);  // V tile row-pair packets
+    constexpr int KPL = (NPK + NT - 1) / NT;
+    constexpr int VPL = (NPV + NT - 1) / NT;
+
+    __shared__ __attribute__((aligned(16))) bf16 k_lds[2][KVB * 128];
+    __shared__ __attribute__((aligned(16))) bf16 vt_lds[2][D * 128];
+
+    const int qblock = blockIdx.x;
+    const int h = blockIdx.y;
+    const int b = blockIdx.z;
+    const int tid = threadIdx.x;
+    const int wave = tid / WAVE_SIZE;
+    const int lane = tid % WAVE_SIZE;
+    const int l31 = lane & 31;
+    const int hi = lane >> 5;
+
+    const int64_t bh_off = ((int64_t)b * H + h) * S;
+    const int hk = h / kv_group;
+    const bf16* qp = q + b * qb + h * qh;
+    const bf16* kp = k + b * kb + hk * kh;
+    const bf16* vp = v + b * vb + hk * vh;
+    bf16* op = o + b * ob + h * oh;
+    const float slope = slopes[h];
+
+    const int qr0 = qblock * ROWS + wave * 32;
+    const int iq = qr0 + l31;
+
+    frag_ab bQ[DCH];
+#pragma unroll
+    for (int c = 0; c < DCH; ++c) {
+        bQ[c] = *reinterpret_cast<const frag_ab*>(
+            qp + (int64_t)iq * qs + c * 16 + 8 * hi);
+    }
+
+    f32x16 accO[DSUB];
+#pragma unroll
+    for (int dsb = 0; dsb < DSUB; ++dsb) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) accO[dsb][r] = 0.f;
+    }
+    float m_run = NEG_INF, l_run = 0.f;
+
+    const bool full_vis = (kv_off <= -S);
+    int nkv = full_vis ? (S / KVB)
+        : (qblock * ROWS + ROWS - kv_off + KVB - 1) / KVB;
+    nkv = min(max(nkv, 0), S / KVB);
+
+    frag_ab kreg[KPL];
+    frag_ab vreg[VPL][2];
+
+    auto stage_load = [&](int nb) {
+        const int kvrow0 = nb * KVB;
+#pragma unroll
+        for (int i = 0; i < KPL; ++i) {
+            const int p = tid + i * NT;
+            if (p < NPK) {
+                const int row = p / (D / 8);
+                const int col = (p % (D / 8)) * 8;
+                kreg[i] = *reinterpret_cast<const frag_ab*>(
+                    kp + (int64_t)(kvrow0 + row) * ks + col);
+            }
+        }
+#pragma unroll
+        for (int i = 0; i < VPL; ++i) {
+            const int p = tid + i * NT;
+            if (p < NPV) {
+                const int row = (p / (D / 8)) * 2;
+                const int col = (p % (D / 8)) * 8;
+                vreg[i][0] = *reinterpret_cast<const frag_ab*>(
+                    vp + (int64_t)(kvrow0 + row) * vs + col);
+                vreg[i][1] = *reinterpret_cast<const frag_ab*>(
+                    vp + (int64_t)(kvrow0 + row + 1) * vs + col);
+            }
+        }
+    };
+    auto stage_write = [&](int buf) {
+#pragma unroll
+        for (int i = 0; i < KPL; ++i) {
+            const int p = tid + i * NT;
+            if (p < NPK) {
+                const int row = p / (D / 8);
+                const int col = (p % (D / 8)) * 8;
+                *reinterpret_cast<frag_ab*>(&k_lds[buf][swz128(row, col)]) =
+                    kreg[i];
+            }
+        }
+#pragma unroll
+        for (int i = 0; i < VPL; ++i) {
+            const int p = tid + i * NT;
+            if (p < NPV) {
+                const int row = (p / (D / 8)) * 2;
+                const int col = (p % (D / 8)) * 8;
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    union { __bf16 h2[2]; uint32_t u; } w;
+                    w.h2[0] = vreg[i][0][j];
+                    w.h2[1] = vreg[i][1][j];
+                    *reinterpret_cast<uint32_t*>(
+                        &vt_lds[buf][swz128(col + j, row)]) = w.u;
+                }
+            }
+        }
+    };
+
+    if (nkv > 0) {
+        stage_load(0);
+        stage_write(0);
+    }
+    __syncthreads();
+
+    for (int nb = 0; nb < nkv; ++nb) {
+        const int cur = nb & 1;
+        const int kvrow0 = nb * KVB;
+        const bool last = (nb + 1 == nkv);
+        if (!last) stage_load(nb + 1);   // T14 issue-early
+
+        // a tile entirely above this wave's causal diagonal is skipped
+        // (wave-uniform branch; staging stays cooperative)
+        const bool active = full_vis || (kvrow0 + kv_off <= qr0 + 31);
+
+        f32x16 st[2];
+        frag_ab pfrag[4];
+        if (active) {
+#pragma unroll
+            for (int ns = 0; ns < 2; ++ns) {
+#pragma unroll
+                for (int r = 0; r < 16; ++r) st[ns][r] = 0.f;
+            }
+#pragma unroll
+            for (int ns = 0; ns < 2; ++ns) {
+                __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                for (int c = 0; c < DCH; ++c) {
+                    frag_ab aK = *reinterpret_cast<const frag_ab*>(
+                        &k_lds[cur][swz128(ns * 32 + l31, c * 16 + 8 * hi)]);
+                    st[ns] = MFMA_32x32x16(aK, bQ[c], st[ns]);
+                }
+                __builtin_amdgcn_s_setprio(0);
+            }
+
+            // scale + ALiBi + causal mask; row max fully in-register
+            float mx = NEG_INF;
+#pragma unroll
+            for (int ns = 0; ns < 2; ++ns) {
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const int jk = kvrow0 + ns * 32 + (r & 3) + 8 * (r >> 2) +
+                                   4 * hi + kv_off;
+                    float x = st[ns][r] * scale + slope * (float)(jk - iq);
+                    x = (jk <= iq) ? x : NEG_INF;
+                    st[ns][r] = x;
+                    mx = fmaxf(mx, x);
+                }
+            }
+            {   // lanes l and l+32 hold the same q row's two kv halves
+                auto sw = __builtin_amdgcn_permlane32_swap(
+                    __float_as_uint(mx), __float_as_uint(mx), false, false);
+                mx = fmaxf(u2f(sw[0]), u2f(sw[1]));
+            }
+            const float m_new = fmaxf(m_run, mx);
+            const float alpha = __expf(m_run - m_new);
+            m_run = m_new;
+            l_run *= alpha;
+#pragma unroll
+            for (int dsb = 0; dsb < DSUB; ++dsb) {
+#pragma unroll
+                for (int r = 0; r < 16; ++r) accO[dsb][r] *= alpha;
+            }
+            float ls = 0.f;
+#pragma unroll
+            for (int ns = 0; ns < 2; ++ns) {
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const float pv = __expf(st[ns][r] - m_new);
+                    st[ns][r] = pv;
+                    ls += pv;
+                }
+            }
+            {
+                auto sw = __builtin_amdgcn_permlane32_swap(
+                    __float_as_uint(ls), __float_as_uint(ls), false, false);
+                ls = u2f(sw[0]) + u2f(sw[1]);
+            }
+            l_run += ls;
+
+            // pack P into PV B-fragments: lane needs kv = 16*ks + 8*hi + e.
+            // Own regs hold kv%8<4 (lower half-wave) / >=4 (upper); one
+            // permlane32_swap per word pair completes both fragments (T12).
+#pragma unroll
+            for (int ns = 0; ns < 2; ++ns) {
+#pragma unroll
+                for (int hf = 0; hf < 2; ++hf) {
+                    union { __bf16 h2[2]; uint32_t u; } wa, wb, wc, wd;
+                    wa.h2[0] = (__bf16)st[ns][8 * hf + 0];
+                    wa.h2[1] = (__bf16)st[ns][8 * hf + 1];
+                    wc.h2[0] = (__bf16)st[ns][8 * hf + 2];
+                    wc.h2[1] = (__bf16)st[ns][8 * hf + 3];
+                    wb.h2[0] = (__bf16)st[ns][8 * hf + 4];
+                    wb.h2[1] = (__bf16)st[ns][8 * hf + 5];
+                    wd.h2[0] = (__bf16)st[ns][8 * hf + 6];
+                    wd.h2[1] = (__bf16)st[ns][8 * hf + 7];
+                    auto s1 = __builtin_amdgcn_permlane32_swap(wa.u, wb.u,
+                                                              false, false);
+                    auto s2 = __builtin_amdgcn_permlane32_swap(wc.u, wd.u,
+                                                              false, false);
+                    union { frag_ab f; uint32_t u[4]; } out;
+                    out.u[0] = s1[0];
+                    out.u[1] = s2[0];
+                    out.u[2] = s1[1];
+                    out.u[3] = s2[1];
+                    pfrag[2 * ns + hf] = out.f;
+                }
+            }
+        }
+
+        if (!last) stage_write(cur ^ 1);  // T14 write-late
+
+        if (active) {
+            // O^T += V^T · P^T  (both operands lane-local / LDS-contiguous)
+#pragma unroll
+            for (int dsb = 0; dsb < DSUB; ++dsb) {
+                __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                for (int ks2 = 0; ks2 < 4; ++ks2) {
+                    frag_ab aV = *reinterpret_cast<const frag_ab*>(
+                        &vt_lds[cur][swz128(dsb * 32 + l31, ks2 * 16 + 8 * hi)]);
+                    accO[dsb] = MFMA_32x32x16(aV, pfrag[ks2], accO[dsb]);
+                }
+                __builtin_amdgcn_s_setprio(0);
+            }
+        }
+        __syncthreads();
+    }
+
+    // epilogue: lane owns q row iq; accO reg r is d = 8*(r>>2)+(r&3)+4*hi
+    const float inv_l = (l_run > 0.f) ? 1.0f / l_run : 0.f;
+    bf16* orow = op + (int64_t)iq * os;
+#pragma unroll
+    for (int dsb = 0; dsb < DSUB; ++dsb) {
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+            union { __bf16 h4[4]; uint2 u; } w;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                w.h4[r] = (__bf16)(accO[dsb][4 * g + r] * inv_l);
+            }
+            *reinterpret_cast<uint2*>(orow + dsb * 32 + 8 * g + 4 * hi) = w.u;
+        }
+    }
+    if (hi == 0) {
+        lse[bh_off + iq] = m_run + __logf(l_run);
+    }
+}
+
+// layout probe for the 32x32x16 fragment maps (see attn_fwd_v2_kernel)
+__global__ void mfma_probe32_kernel(const bf16* __restrict__ A,
+                                    const bf16* __restrict__ Bt,
+                                    float* __restrict__ C) {
+    const int lane = threadIdx.x;
+    const int l31 = lane & 31, hi = lane >> 5;
+    // A[i][k]: i = l31, k = 8*hi + e (row-major A: 32x16)
+    frag_ab a = *reinterpret_cast<const frag_ab*>(A + l31 * 16 + 8 * hi);
+    // B[k][j] with Bt stored [j][k] (row-major 32x16): j = l31, k = 8*hi + e
+    frag_ab b = *reinterpret_cast<const frag_ab*>(Bt + l31 * 16 + 8 * hi);
+    f32x16 c;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) c[r] = 0.f;
+    c = MFMA_32x32x16(a, b, c);
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+        C[((r & 3) + 8 * (r >> 2) + 4 * hi) * 32 + l31] = c[r];
+    }
+}
+
+}  // namespace
+
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor slopes,
                                     double scale, int64_t kv_off) {
@@ -294,6 +623,40 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
     auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
 
     auto stream = at::cuda::getCurrentCUDAStream();
+
+    // v2 (8-wave 32x32-MFMA ladder kernel) handles S % 128 == 0; v1 covers
+    // the rest.  PG_ATTN_V2=0 forces the v1 path for A/B comparisons.
+    static const int use_v2 = [] {
+        const char* e = getenv("PG_ATTN_V2");
+        return e ? atoi(e) : 1;
+    }();
+    if (use_v2 && S % 128 == 0) {
+#define LAUNCH_FWD2(DV, WV)                                                   \
+    do {                                                                      \
+        dim3 grid(S / (32 * WV), H, B);                                       \
+        hipLaunchKernelGGL((attn_fwd_v2_kernel<DV, WV>), grid,                \
+            dim3(WV * WAVE_SIZE), 0, stream,                                  \
+            reinterpret_cast<const bf16*>(q.data_ptr()),                      \
+            reinterpret_cast<const bf16*>(k.data_ptr()),                      \
+            reinterpret_cast<const bf16*>(v.data_ptr()),                      \
+            slopes.data_ptr<float>(), (float)scale,                           \
+            reinterpret_cast<bf16*>(o_phys.data_ptr()), lse.data_ptr<float>(),\
+            B, H, S, (int)kv_off, kv_group,                                   \
+            q.stride(0), q.stride(1), q.stride(2),                            \
+            k.stride(0), k.stride(1), k.stride(2),                            \
+            v.stride(0), v.stride(1), v.stride(2),                            \
+            o.stride(0), o.stride(1), o.stride(2));                           \
+    } while (0)
+        if (D == 64) {
+            if (S % 256 == 0) LAUNCH_FWD2(64, 8); else LAUNCH_FWD2(64, 4);
+        } else {
+            if (S % 256 == 0) LAUNCH_FWD2(128, 8); else LAUNCH_FWD2(128, 4);
+        }
+#undef LAUNCH_FWD2
+        HIP_CHECK_LAUNCH();
+        return {o, lse};
+    }
+
     // MT=2 (128 q rows / workgroup) when S allows: 2x the MFMA work per LDS
     // fragment read
     // D=64: 4 waves x 2 row-tiles.  D=128: default 8 waves x 1 tile
@@ -339,6 +702,20 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
 #undef LAUNCH_FWD
     HIP_CHECK_LAUNCH();
     return {o, lse};
+}
+
+torch::Tensor mfma_probe32(torch::Tensor A, torch::Tensor Bt) {
+    TORCH_CHECK(A.is_cuda() && A.scalar_type() == torch::kBFloat16);
+    TORCH_CHECK(A.sizes() == torch::IntArrayRef({32, 16}) &&
+                Bt.sizes() == torch::IntArrayRef({32, 16}));
+    auto C = torch::empty({32, 32}, A.options().dtype(torch::kFloat));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(mfma_probe32_kernel, dim3(1), dim3(64), 0, stream,
+        reinterpret_cast<const bf16*>(A.contiguous().data_ptr()),
+        reinterpret_cast<const bf16*>(Bt.contiguous().data_ptr()),
+        C.data_ptr<float>());
+    HIP_CHECK_LAUNCH();
+    return C;
 }
 
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor Bt) {

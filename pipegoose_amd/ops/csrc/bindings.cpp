@@ -33,6 +33,7 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
                    torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
                    int64_t kv_off);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor Bt);
+torch::Tensor mfma_probe32(torch::Tensor A, torch::Tensor Bt);
 std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w,
                                         double eps);
 std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
@@ -64,6 +65,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "(e.g. fused-qkv gradient slices)");
     m.def("mfma_probe", &mfma_probe,
           "16x16x32 bf16 MFMA fragment-layout probe");
+    m.def("mfma_probe32", &mfma_probe32,
+          "32x32x16 bf16 MFMA fragment-layout probe");
     m.def("rms_norm_fwd", &rms_norm_fwd, "fused RMSNorm forward (gfx950)");
     m.def("rms_norm_bwd", &rms_norm_bwd, "fused RMSNorm backward (gfx950)");
     m.def("rope_apply", &rope_apply,

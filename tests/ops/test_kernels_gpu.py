@@ -241,6 +241,20 @@ def test_mfma_probe_layout():
     assert torch.allclose(C, ref, atol=1e-2), (C - ref).abs().max()
 
 
+def test_mfma_probe32_layout():
+    """Verify the 32x32x16 bf16 MFMA fragment layouts used by the v2
+    attention forward (asymmetric operands, as the guide demands)."""
+    ext = _ext()
+    if not hasattr(ext, "mfma_probe32"):
+        pytest.skip("extension predates probe32")
+    torch.manual_seed(11)
+    A = torch.randn(32, 16, device="cuda").to(torch.bfloat16)
+    B = torch.randn(16, 32, device="cuda").to(torch.bfloat16)
+    C = ext.mfma_probe32(A, B.t().contiguous())
+    ref = A.float() @ B.float()
+    assert torch.allclose(C, ref, atol=1e-2), (C - ref).abs().max()
+
+
 def _attn_oracle(q, k, v, slopes, scale):
     S = q.size(-2)
     pos = torch.arange(S, device=q.device)
@@ -252,7 +266,14 @@ def _attn_oracle(q, k, v, slopes, scale):
         q.float(), k.float(), v.float(), attn_mask=bias[None], scale=scale)
 
 
-@pytest.mark.parametrize("shape", [(2, 4, 128, 64), (1, 3, 256, 128)])
+@pytest.mark.parametrize("shape", [
+    (2, 4, 128, 64),      # v2 4-wave
+    (1, 3, 256, 128),     # v2 8-wave
+    (2, 2, 256, 64),      # v2 8-wave D=64
+    (1, 2, 384, 128),     # v2 4-wave (384 % 256 != 0)
+    (1, 2, 192, 64),      # v1 fallback (192 % 128 != 0)
+    (1, 2, 512, 128),     # v2 8-wave, multi-tile causal path
+])
 def test_attn_fwd_kernel(shape):
     ext = _ext()
     torch.manual_seed(6)
@@ -450,11 +471,13 @@ def test_attn_kernel_kv_offset(kv_off):
         assert err / ref_scale < 5e-2, f"{name}: {err}"
 
 
-def test_attn_kernel_gqa():
-    """GQA: kernel with Hkv < H vs expanded-kv fp32 oracle (fwd + grads)."""
+@pytest.mark.parametrize("S", [128, 256])
+def test_attn_kernel_gqa(S):
+    """GQA: kernel with Hkv < H vs expanded-kv fp32 oracle (fwd + grads).
+    S=128 exercises the 4-wave v2 forward, S=256 the 8-wave one."""
     ext = _ext()
     torch.manual_seed(15)
-    B, H, Hkv, S, D = 2, 8, 2, 128, 64
+    B, H, Hkv, D = 2, 8, 2, 64
     group = H // Hkv
     q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16,
                     requires_grad=True)
