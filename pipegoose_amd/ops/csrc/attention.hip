@@ -306,6 +306,11 @@ __device__ __forceinline__ int swz128(int row, int elem_col) {
     return row * 128 + ((((elem_col) << 1) ^ ((row & 15) << 4)) >> 1);
 }
 
+// Same idea on 128-B rows (64 bf16): 8 slots per row, <=2-way residual.
+__device__ __forceinline__ int swz64(int row, int elem_col) {
+    return row * 64 + ((((elem_col) << 1) ^ ((row & 7) << 4)) >> 1);
+}
+
 // C/D layout of mfma_f32_32x32x16_bf16: col = lane&31, row = (r&3) +
 // 8*(r>>2) + 4*(lane>>5), r in [0,16).  A/B fragments: i/j = lane&31,
 // k = 8*(lane>>5) + elem (verified by mfma_probe32).
@@ -333,9 +338,23 @@ void attn_fwd_v2_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
     __shared__ __attribute__((aligned(16))) bf16 k_lds[2][KVB * 128];
     __shared__ __attribute__((aligned(16))) bf16 vt_lds[2][D * 128];
 
-    const int qblock = blockIdx.x;
-    const int h = blockIdx.y;
-    const int b = blockIdx.z;
+    // XCD-affinity remap (T1): the hardware places linear block id L on XCD
+    // L%8, so with qblock fastest-varying the q-blocks sharing one (b,h)'s
+    // K/V land on 8 different L2s.  When H*B % 8 == 0, re-deriving (qblock,
+    // h, b) with (h,b) fastest pins each (b,h) group to ONE XCD, making its
+    // K/V tile L2-resident across q-blocks.  Pure speed, never correctness.
+    int qblock = blockIdx.x, h = blockIdx.y, b = blockIdx.z;
+    {
+        const int ngrp = gridDim.y * gridDim.z;  // H * B
+        if ((ngrp & 7) == 0) {
+            const int flat = blockIdx.x +
+                gridDim.x * (blockIdx.y + gridDim.y * blockIdx.z);
+            qblock = flat / ngrp;
+            const int g = flat % ngrp;
+            h = g % gridDim.y;
+            b = g / gridDim.y;
+        }
+    }
     const int tid = threadIdx.x;
     const int wave = tid / WAVE_SIZE;
     const int lane = tid % WAVE_SIZE;
@@ -348,7 +367,11 @@ void attn_fwd_v2_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* kp = k + b * kb + hk * kh;
     const bf16* vp = v + b * vb + hk * vh;
     bf16* op = o + b * ob + h * oh;
-    const float slope = slopes[h];
+    // softmax runs in the exp2 domain (v_exp_f32 is natively exp2): fold
+    // log2(e) into scale and slope so no per-element multiply is added
+    const float LOG2E = 1.4426950408889634f;
+    const float scale2 = scale * LOG2E;
+    const float slope2 = slopes[h] * LOG2E;
 
     const int qr0 = qblock * ROWS + wave * 32;
     const int iq = qr0 + l31;
@@ -466,27 +489,35 @@ void attn_fwd_v2_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
                 __builtin_amdgcn_s_setprio(0);
             }
 
-            // scale + ALiBi + causal mask; row max fully in-register
-            float mx = NEG_INF;
+            // scale + ALiBi + causal mask (exp2 domain); row max in-register
+            // as a log-depth tree (a 32-deep serial fmax chain is ~128
+            // dependent cycles)
+            float mr[32];
 #pragma unroll
             for (int ns = 0; ns < 2; ++ns) {
 #pragma unroll
                 for (int r = 0; r < 16; ++r) {
                     const int jk = kvrow0 + ns * 32 + (r & 3) + 8 * (r >> 2) +
                                    4 * hi + kv_off;
-                    float x = st[ns][r] * scale + slope * (float)(jk - iq);
+                    float x = st[ns][r] * scale2 + slope2 * (float)(jk - iq);
                     x = (jk <= iq) ? x : NEG_INF;
                     st[ns][r] = x;
-                    mx = fmaxf(mx, x);
+                    mr[ns * 16 + r] = x;
                 }
             }
+#pragma unroll
+            for (int w = 16; w > 0; w >>= 1) {
+#pragma unroll
+                for (int i = 0; i < w; ++i) mr[i] = fmaxf(mr[i], mr[i + w]);
+            }
+            float mx = mr[0];
             {   // lanes l and l+32 hold the same q row's two kv halves
                 auto sw = __builtin_amdgcn_permlane32_swap(
                     __float_as_uint(mx), __float_as_uint(mx), false, false);
                 mx = fmaxf(u2f(sw[0]), u2f(sw[1]));
             }
             const float m_new = fmaxf(m_run, mx);
-            const float alpha = __expf(m_run - m_new);
+            const float alpha = exp2f(m_run - m_new);
             m_run = m_new;
             l_run *= alpha;
 #pragma unroll
@@ -494,16 +525,22 @@ void attn_fwd_v2_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
 #pragma unroll
                 for (int r = 0; r < 16; ++r) accO[dsb][r] *= alpha;
             }
-            float ls = 0.f;
+            float sr[32];
 #pragma unroll
             for (int ns = 0; ns < 2; ++ns) {
 #pragma unroll
                 for (int r = 0; r < 16; ++r) {
-                    const float pv = __expf(st[ns][r] - m_new);
+                    const float pv = exp2f(st[ns][r] - m_new);
                     st[ns][r] = pv;
-                    ls += pv;
+                    sr[ns * 16 + r] = pv;
                 }
             }
+#pragma unroll
+            for (int w = 16; w > 0; w >>= 1) {
+#pragma unroll
+                for (int i = 0; i < w; ++i) sr[i] += sr[i + w];
+            }
+            float ls = sr[0];
             {
                 auto sw = __builtin_amdgcn_permlane32_swap(
                     __float_as_uint(ls), __float_as_uint(ls), false, false);
@@ -576,7 +613,8 @@ void attn_fwd_v2_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
         }
     }
     if (hi == 0) {
-        lse[bh_off + iq] = m_run + __logf(l_run);
+        // stats were kept in the exp2 domain; lse is defined in natural log
+        lse[bh_off + iq] = 0.6931471805599453f * (m_run + __log2f(l_run));
     }
 }
 
@@ -789,17 +827,19 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
                           int64_t wb, int64_t wh, int64_t ws) {
     constexpr int DCH = D / 32;
     constexpr int DSUB = D / 16;
-    constexpr int RSTRIDE = D + PAD;   // row-major tiles [QR][D+PAD]
-    constexpr int TSTRIDE = QR + PAD;  // transposed tiles [D][QR+PAD]
     constexpr int NT = W * WAVE_SIZE;
     constexpr int KVROWS = W * 16;     // kv rows per workgroup
 
-    __shared__ bf16 do_lds[QR * RSTRIDE];
-    __shared__ bf16 q_lds[QR * RSTRIDE];
-    __shared__ bf16 qt_lds[D * TSTRIDE];
-    __shared__ bf16 dot_lds[D * TSTRIDE];
-    __shared__ bf16 pt_lds[W][16 * TSTRIDE];   // P^T  (n rows, m cols)
-    __shared__ bf16 dst_lds[W][16 * TSTRIDE];  // dS^T (n rows, m cols)
+    // round-2 retrofit: power-of-2 row strides with the T2 XOR swizzle
+    // (row-major tiles on 256-B rows via swz128, transposed tiles on 128-B
+    // rows via swz64) replace the +8-element padding — r1 PMC measured
+    // 33-63% of LDS cycles lost to bank conflicts on the padded layout.
+    __shared__ bf16 do_lds[QR * 128];
+    __shared__ bf16 q_lds[QR * 128];
+    __shared__ bf16 qt_lds[D * 64];
+    __shared__ bf16 dot_lds[D * 64];
+    __shared__ bf16 pt_lds[W][16 * 64];   // P^T  (n rows, m cols)
+    __shared__ bf16 dst_lds[W][16 * 64];  // dS^T (n rows, m cols)
     __shared__ float lse_lds[QR];
     __shared__ float delta_lds[QR];
 
@@ -841,6 +881,9 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
     const bf16* dop = dout + b * gb + hq * gh;
     const float slope = slopes[hq];
 
+    const float LOG2E = 1.4426950408889634f;
+    const float scale2 = scale * LOG2E;
+    const float slope2 = slope * LOG2E;
     const int q_start = (kv_off <= -S) ? 0 : nb * KVROWS + kv_off;
     for (int q0 = q_start < 0 ? 0 : q_start; q0 < S; q0 += QR) {
         __syncthreads();
@@ -857,21 +900,21 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
                     dop + (int64_t)(q0 + row) * gs + col);
                 frag_ab pd1 = *reinterpret_cast<const frag_ab*>(
                     dop + (int64_t)(q0 + row + 1) * gs + col);
-                *reinterpret_cast<frag_ab*>(&q_lds[row * RSTRIDE + col]) = pq0;
-                *reinterpret_cast<frag_ab*>(&q_lds[(row + 1) * RSTRIDE + col]) = pq1;
-                *reinterpret_cast<frag_ab*>(&do_lds[row * RSTRIDE + col]) = pd0;
-                *reinterpret_cast<frag_ab*>(&do_lds[(row + 1) * RSTRIDE + col]) = pd1;
+                *reinterpret_cast<frag_ab*>(&q_lds[swz128(row, col)]) = pq0;
+                *reinterpret_cast<frag_ab*>(&q_lds[swz128(row + 1, col)]) = pq1;
+                *reinterpret_cast<frag_ab*>(&do_lds[swz128(row, col)]) = pd0;
+                *reinterpret_cast<frag_ab*>(&do_lds[swz128(row + 1, col)]) = pd1;
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
                     union { __bf16 h[2]; uint32_t u; } a, b2;
                     a.h[0] = pq0[j]; a.h[1] = pq1[j];
                     b2.h[0] = pd0[j]; b2.h[1] = pd1[j];
-                    *reinterpret_cast<uint32_t*>(&qt_lds[(col + j) * TSTRIDE + row]) = a.u;
-                    *reinterpret_cast<uint32_t*>(&dot_lds[(col + j) * TSTRIDE + row]) = b2.u;
+                    *reinterpret_cast<uint32_t*>(&qt_lds[swz64(col + j, row)]) = a.u;
+                    *reinterpret_cast<uint32_t*>(&dot_lds[swz64(col + j, row)]) = b2.u;
                 }
             }
             for (int i = tid; i < QR; i += NT) {
-                lse_lds[i] = lse[bh_off + q0 + i];
+                lse_lds[i] = lse[bh_off + q0 + i] * LOG2E;
                 delta_lds[i] = delta[bh_off + q0 + i];
             }
         }
@@ -885,10 +928,10 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
 #pragma unroll
             for (int c = 0; c < DCH; ++c) {
                 frag_ab bQ = *reinterpret_cast<const frag_ab*>(
-                    &q_lds[(ms * 16 + lcol) * RSTRIDE + c * 32 + 8 * lgrp]);
+                    &q_lds[swz128(ms * 16 + lcol, c * 32 + 8 * lgrp)]);
                 st = MFMA_16x16x32(aK[c], bQ, st);          // S^T = K·Q^T
                 frag_ab bDO = *reinterpret_cast<const frag_ab*>(
-                    &do_lds[(ms * 16 + lcol) * RSTRIDE + c * 32 + 8 * lgrp]);
+                    &do_lds[swz128(ms * 16 + lcol, c * 32 + 8 * lgrp)]);
                 dpt = MFMA_16x16x32(aV[c], bDO, dpt);       // dP^T = V·dO^T
             }
 #pragma unroll
@@ -896,11 +939,11 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
                 const int n_glob = kv0 + 4 * lgrp + r + kv_off;
                 const int m_loc = ms * 16 + lcol;
                 const int m_glob = q0 + m_loc;
-                float z = st[r] * scale + slope * (float)(n_glob - m_glob);
-                float p = (n_glob <= m_glob) ? __expf(z - lse_lds[m_loc]) : 0.f;
+                float z = st[r] * scale2 + slope2 * (float)(n_glob - m_glob);
+                float p = (n_glob <= m_glob) ? exp2f(z - lse_lds[m_loc]) : 0.f;
                 float ds = p * (dpt[r] - delta_lds[m_loc]) * scale;
-                pt_lds[wave][(4 * lgrp + r) * TSTRIDE + m_loc] = (bf16)p;
-                dst_lds[wave][(4 * lgrp + r) * TSTRIDE + m_loc] = (bf16)ds;
+                pt_lds[wave][swz64(4 * lgrp + r, m_loc)] = (bf16)p;
+                dst_lds[wave][swz64(4 * lgrp + r, m_loc)] = (bf16)ds;
             }
         }
         __builtin_amdgcn_s_waitcnt(0);  // own-wave LDS writes visible
@@ -909,16 +952,16 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
 #pragma unroll
         for (int kc = 0; kc < QR / 32; ++kc) {
             frag_ab aPT = *reinterpret_cast<const frag_ab*>(
-                &pt_lds[wave][lcol * TSTRIDE + kc * 32 + 8 * lgrp]);
+                &pt_lds[wave][swz64(lcol, kc * 32 + 8 * lgrp)]);
             frag_ab aDST = *reinterpret_cast<const frag_ab*>(
-                &dst_lds[wave][lcol * TSTRIDE + kc * 32 + 8 * lgrp]);
+                &dst_lds[wave][swz64(lcol, kc * 32 + 8 * lgrp)]);
 #pragma unroll
             for (int ds = 0; ds < DSUB; ++ds) {
                 frag_ab bDOT = *reinterpret_cast<const frag_ab*>(
-                    &dot_lds[(ds * 16 + lcol) * TSTRIDE + kc * 32 + 8 * lgrp]);
+                    &dot_lds[swz64(ds * 16 + lcol, kc * 32 + 8 * lgrp)]);
                 accDV[ds] = MFMA_16x16x32(aPT, bDOT, accDV[ds]);
                 frag_ab bQT = *reinterpret_cast<const frag_ab*>(
-                    &qt_lds[(ds * 16 + lcol) * TSTRIDE + kc * 32 + 8 * lgrp]);
+                    &qt_lds[swz64(ds * 16 + lcol, kc * 32 + 8 * lgrp)]);
                 accDK[ds] = MFMA_16x16x32(aDST, bQT, accDK[ds]);
             }
         }
@@ -960,15 +1003,14 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
                         int64_t wb, int64_t wh, int64_t ws) {
     constexpr int DCH = D / 32;
     constexpr int DSUB = D / 16;
-    constexpr int RSTRIDE = D + PAD;
-    constexpr int TSTRIDE = KVR + PAD;
     constexpr int NT = W * WAVE_SIZE;
     constexpr int QROWS = W * 16;          // q rows per workgroup
 
-    __shared__ bf16 k_lds[KVR * RSTRIDE];   // row-major K (QK^T B-frags)
-    __shared__ bf16 v_lds[KVR * RSTRIDE];   // row-major V (dP B-frags)
-    __shared__ bf16 kt_lds[D * TSTRIDE];    // transposed K (dQ B-frags)
-    __shared__ bf16 ds_lds[W][16 * TSTRIDE];  // dS (m rows, n cols)
+    // swizzled strides, same scheme as dkdv (see comment there)
+    __shared__ bf16 k_lds[KVR * 128];       // row-major K (QK^T B-frags)
+    __shared__ bf16 v_lds[KVR * 128];       // row-major V (dP B-frags)
+    __shared__ bf16 kt_lds[D * 64];         // transposed K (dQ B-frags)
+    __shared__ bf16 ds_lds[W][16 * 64];     // dS (m rows, n cols)
 
     const int qb = blockIdx.x;
     const int h = blockIdx.y;
@@ -997,11 +1039,14 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
             aDO[c] = *reinterpret_cast<const frag_ab*>(dorow + c * 32 + 8 * lgrp);
         }
     }
+    const float LOG2E = 1.4426950408889634f;
+    const float scale2 = scale * LOG2E;
+    const float slope2 = slope * LOG2E;
     float lse_r[4], delta_r[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
         const int iq = qrow0 + 4 * lgrp + r;
-        lse_r[r] = lse[bh_off + iq];
+        lse_r[r] = lse[bh_off + iq] * LOG2E;
         delta_r[r] = delta[bh_off + iq];
     }
 
@@ -1026,15 +1071,15 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
                     vp + (int64_t)(kv0 + row) * vs2 + col);
                 frag_ab pv1 = *reinterpret_cast<const frag_ab*>(
                     vp + (int64_t)(kv0 + row + 1) * vs2 + col);
-                *reinterpret_cast<frag_ab*>(&k_lds[row * RSTRIDE + col]) = pk0;
-                *reinterpret_cast<frag_ab*>(&k_lds[(row + 1) * RSTRIDE + col]) = pk1;
-                *reinterpret_cast<frag_ab*>(&v_lds[row * RSTRIDE + col]) = pv0;
-                *reinterpret_cast<frag_ab*>(&v_lds[(row + 1) * RSTRIDE + col]) = pv1;
+                *reinterpret_cast<frag_ab*>(&k_lds[swz128(row, col)]) = pk0;
+                *reinterpret_cast<frag_ab*>(&k_lds[swz128(row + 1, col)]) = pk1;
+                *reinterpret_cast<frag_ab*>(&v_lds[swz128(row, col)]) = pv0;
+                *reinterpret_cast<frag_ab*>(&v_lds[swz128(row + 1, col)]) = pv1;
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
                     union { __bf16 h[2]; uint32_t u; } a;
                     a.h[0] = pk0[j]; a.h[1] = pk1[j];
-                    *reinterpret_cast<uint32_t*>(&kt_lds[(col + j) * TSTRIDE + row]) = a.u;
+                    *reinterpret_cast<uint32_t*>(&kt_lds[swz64(col + j, row)]) = a.u;
                 }
             }
         }
@@ -1048,22 +1093,21 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
 #pragma unroll
             for (int c = 0; c < DCH; ++c) {
                 frag_ab bK = *reinterpret_cast<const frag_ab*>(
-                    &k_lds[(ns * 16 + lcol) * RSTRIDE + c * 32 + 8 * lgrp]);
+                    &k_lds[swz128(ns * 16 + lcol, c * 32 + 8 * lgrp)]);
                 sacc = MFMA_16x16x32(aQ[c], bK, sacc);      // S = Q·K^T
                 frag_ab bV = *reinterpret_cast<const frag_ab*>(
-                    &v_lds[(ns * 16 + lcol) * RSTRIDE + c * 32 + 8 * lgrp]);
+                    &v_lds[swz128(ns * 16 + lcol, c * 32 + 8 * lgrp)]);
                 dpacc = MFMA_16x16x32(aDO[c], bV, dpacc);   // dP = dO·V^T
             }
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int iq = qrow0 + 4 * lgrp + r;
                 const int jk = kv0 + ns * 16 + lcol + kv_off;
-                float z = sacc[r] * scale + slope * (float)(jk - iq);
-                float p = (jk <= iq) ? __expf(z - lse_r[r]) : 0.f;
+                float z = sacc[r] * scale2 + slope2 * (float)(jk - iq);
+                float p = (jk <= iq) ? exp2f(z - lse_r[r]) : 0.f;
                 float ds = p * (dpacc[r] - delta_r[r]) * scale;
                 // store transposed to [m][n] so the dQ A-frag read is linear
-                ds_lds[wave][(4 * lgrp + r) * TSTRIDE + ns * 16 + lcol] =
-                    (bf16)ds;
+                ds_lds[wave][swz64(4 * lgrp + r, ns * 16 + lcol)] = (bf16)ds;
             }
         }
         __builtin_amdgcn_s_waitcnt(0);
@@ -1072,11 +1116,11 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
 #pragma unroll
         for (int kc = 0; kc < KVR / 32; ++kc) {
             frag_ab aDS = *reinterpret_cast<const frag_ab*>(
-                &ds_lds[wave][lcol * TSTRIDE + kc * 32 + 8 * lgrp]);
+                &ds_lds[wave][swz64(lcol, kc * 32 + 8 * lgrp)]);
 #pragma unroll
             for (int ds = 0; ds < DSUB; ++ds) {
                 frag_ab bKT = *reinterpret_cast<const frag_ab*>(
-                    &kt_lds[(ds * 16 + lcol) * TSTRIDE + kc * 32 + 8 * lgrp]);
+                    &kt_lds[swz64(ds * 16 + lcol, kc * 32 + 8 * lgrp)]);
                 accDQ[ds] = MFMA_16x16x32(aDS, bKT, accDQ[ds]);
             }
         }
