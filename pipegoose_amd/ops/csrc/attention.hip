@@ -293,6 +293,8 @@ __device__ __forceinline__ float u2f(unsigned u) { return __uint_as_float(u); }
 namespace {
 
 using f32x16 = __attribute__((ext_vector_type(16))) float;
+using bf16x4 = __attribute__((ext_vector_type(4))) __bf16;
+using lds_bf16x4_p = __attribute__((address_space(3))) bf16x4*;
 
 __device__ __forceinline__ f32x16 MFMA_32x32x16(frag_ab a, frag_ab b, f32x16 c) {
     return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
@@ -309,6 +311,11 @@ __device__ __forceinline__ int swz128(int row, int elem_col) {
 // Same idea on 128-B rows (64 bf16): 8 slots per row, <=2-way residual.
 __device__ __forceinline__ int swz64(int row, int elem_col) {
     return row * 64 + ((((elem_col) << 1) ^ ((row & 7) << 4)) >> 1);
+}
+
+template <int RS>
+__device__ __forceinline__ int swz_rm(int row, int elem_col) {
+    return RS == 64 ? swz64(row, elem_col) : swz128(row, elem_col);
 }
 
 // C/D layout of mfma_f32_32x32x16_bf16: col = lane&31, row = (r&3) +
@@ -334,9 +341,15 @@ void attn_fwd_v2_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
     constexpr int NPV = (KVB / 2) * (D / 8);  // V tile row-pair packets
     constexpr int KPL = (NPK + NT - 1) / NT;
     constexpr int VPL = (NPV + NT - 1) / NT;
+    // V image is [kv/4][d/16][kv%4][16]: each [4 kv][16 d] sub-tile is 128 B
+    // contiguous, which is exactly what ds_read_b64_tr_b16 gathers (T10) —
+    // the PV A-fragment (V^T) comes out of the hardware transpose read, so
+    // no scattered u32 transpose-writes (those were 8-way write conflicts =
+    // the 61.6% CONF of the r2 PMC) and half the LDS of the [D][128] image.
+    constexpr int VST = (D / 16) * 64;      // elements per 4-kv-row group
 
     __shared__ __attribute__((aligned(16))) bf16 k_lds[2][KVB * 128];
-    __shared__ __attribute__((aligned(16))) bf16 vt_lds[2][D * 128];
+    __shared__ __attribute__((aligned(16))) bf16 vs_lds[2][KVB * D];
 
     // XCD-affinity remap (T1): the hardware places linear block id L on XCD
     // L%8, so with qblock fastest-varying the q-blocks sharing one (b,h)'s
@@ -442,12 +455,12 @@ void attn_fwd_v2_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
                 const int row = (p / (D / 8)) * 2;
                 const int col = (p % (D / 8)) * 8;
 #pragma unroll
-                for (int j = 0; j < 8; ++j) {
-                    union { __bf16 h2[2]; uint32_t u; } w;
-                    w.h2[0] = vreg[i][0][j];
-                    w.h2[1] = vreg[i][1][j];
-                    *reinterpret_cast<uint32_t*>(
-                        &vt_lds[buf][swz128(col + j, row)]) = w.u;
+                for (int rr = 0; rr < 2; ++rr) {
+                    const int r = row + rr;
+                    const int idx = (r >> 2) * VST + (col >> 4) * 64 +
+                                    (r & 3) * 16 + (col & 15);
+                    *reinterpret_cast<frag_ab*>(&vs_lds[buf][idx]) =
+                        vreg[i][rr];
                 }
             }
         }
@@ -581,15 +594,26 @@ void attn_fwd_v2_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
         if (!last) stage_write(cur ^ 1);  // T14 write-late
 
         if (active) {
-            // O^T += V^T · P^T  (both operands lane-local / LDS-contiguous)
+            // O^T += V^T · P^T.  The V^T A-fragment comes from two hardware
+            // transpose reads per (dsb, ks): quarter-wave q reads the
+            // [4 kv][16 d] sub-tile for d-block 2*dsb+(q&1), kv group
+            // 4*ks + 2*hi (+1), and each lane receives its d-column.
+            const int lq = lane >> 4;           // quarter-wave index
+            const int dblk_off = (lq & 1) * 64; // within-group d-block elems
+            const int lel = (lane & 15) * 4;    // lane's element in the tile
 #pragma unroll
             for (int dsb = 0; dsb < DSUB; ++dsb) {
                 __builtin_amdgcn_s_setprio(1);
 #pragma unroll
                 for (int ks2 = 0; ks2 < 4; ++ks2) {
-                    frag_ab aV = *reinterpret_cast<const frag_ab*>(
-                        &vt_lds[cur][swz128(dsb * 32 + l31, ks2 * 16 + 8 * hi)]);
-                    accO[dsb] = MFMA_32x32x16(aV, pfrag[ks2], accO[dsb]);
+                    const int g0 = 4 * ks2 + 2 * hi;
+                    const int base = g0 * VST + (2 * dsb) * 64 + dblk_off + lel;
+                    union { frag_ab f; bf16x4 h[2]; } av;
+                    av.h[0] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                        (lds_bf16x4_p)&vs_lds[cur][base]);
+                    av.h[1] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                        (lds_bf16x4_p)&vs_lds[cur][base + VST]);
+                    accO[dsb] = MFMA_32x32x16(av.f, pfrag[ks2], accO[dsb]);
                 }
                 __builtin_amdgcn_s_setprio(0);
             }
@@ -615,6 +639,27 @@ void attn_fwd_v2_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
     if (hi == 0) {
         // stats were kept in the exp2 domain; lse is defined in natural log
         lse[bh_off + iq] = 0.6931471805599453f * (m_run + __log2f(l_run));
+    }
+}
+
+// semantics probe for ds_read_b64_tr_b16: stage a [4][16] bf16 tile into
+// LDS, read it with the transpose instruction the way the PV loop does,
+// and write out what each lane received (expected: lane l&15 gets column
+// l&15, elements j=0..3 = rows).
+__global__ void tr16_probe_kernel(const bf16* __restrict__ in,
+                                  float* __restrict__ out) {
+    __shared__ __attribute__((aligned(16))) bf16 lds[64];
+    const int lane = threadIdx.x;
+    if (lane < 8) {
+        *reinterpret_cast<frag_ab*>(&lds[lane * 8]) =
+            *reinterpret_cast<const frag_ab*>(in + lane * 8);
+    }
+    __syncthreads();
+    if (lane < 16) {
+        bf16x4 v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (lds_bf16x4_p)&lds[lane * 4]);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) out[lane * 4 + j] = (float)v[j];
     }
 }
 
@@ -742,6 +787,18 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
     return {o, lse};
 }
 
+torch::Tensor tr16_probe(torch::Tensor tile) {
+    TORCH_CHECK(tile.is_cuda() && tile.scalar_type() == torch::kBFloat16 &&
+                tile.numel() == 64);
+    auto out = torch::empty({16, 4}, tile.options().dtype(torch::kFloat));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0, stream,
+        reinterpret_cast<const bf16*>(tile.contiguous().data_ptr()),
+        out.data_ptr<float>());
+    HIP_CHECK_LAUNCH();
+    return out;
+}
+
 torch::Tensor mfma_probe32(torch::Tensor A, torch::Tensor Bt) {
     TORCH_CHECK(A.is_cuda() && A.scalar_type() == torch::kBFloat16);
     TORCH_CHECK(A.sizes() == torch::IntArrayRef({32, 16}) &&
@@ -831,11 +888,13 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
     constexpr int KVROWS = W * 16;     // kv rows per workgroup
 
     // round-2 retrofit: power-of-2 row strides with the T2 XOR swizzle
-    // (row-major tiles on 256-B rows via swz128, transposed tiles on 128-B
-    // rows via swz64) replace the +8-element padding — r1 PMC measured
-    // 33-63% of LDS cycles lost to bank conflicts on the padded layout.
-    __shared__ bf16 do_lds[QR * 128];
-    __shared__ bf16 q_lds[QR * 128];
+    // (row-major tiles on 256-B rows via swz128 for D=128 — 128-B rows via
+    // swz64 for D=64, keeping the LDS footprint and occupancy of r1 —
+    // transposed tiles on 128-B rows via swz64) replace the +8-element
+    // padding: r1 PMC measured 33-63% of LDS cycles in bank conflicts.
+    constexpr int RMS = (D == 64) ? 64 : 128;   // row-major tile stride
+    __shared__ bf16 do_lds[QR * RMS];
+    __shared__ bf16 q_lds[QR * RMS];
     __shared__ bf16 qt_lds[D * 64];
     __shared__ bf16 dot_lds[D * 64];
     __shared__ bf16 pt_lds[W][16 * 64];   // P^T  (n rows, m cols)
@@ -900,10 +959,10 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
                     dop + (int64_t)(q0 + row) * gs + col);
                 frag_ab pd1 = *reinterpret_cast<const frag_ab*>(
                     dop + (int64_t)(q0 + row + 1) * gs + col);
-                *reinterpret_cast<frag_ab*>(&q_lds[swz128(row, col)]) = pq0;
-                *reinterpret_cast<frag_ab*>(&q_lds[swz128(row + 1, col)]) = pq1;
-                *reinterpret_cast<frag_ab*>(&do_lds[swz128(row, col)]) = pd0;
-                *reinterpret_cast<frag_ab*>(&do_lds[swz128(row + 1, col)]) = pd1;
+                *reinterpret_cast<frag_ab*>(&q_lds[swz_rm<RMS>(row, col)]) = pq0;
+                *reinterpret_cast<frag_ab*>(&q_lds[swz_rm<RMS>(row + 1, col)]) = pq1;
+                *reinterpret_cast<frag_ab*>(&do_lds[swz_rm<RMS>(row, col)]) = pd0;
+                *reinterpret_cast<frag_ab*>(&do_lds[swz_rm<RMS>(row + 1, col)]) = pd1;
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
                     union { __bf16 h[2]; uint32_t u; } a, b2;
@@ -928,10 +987,10 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
 #pragma unroll
             for (int c = 0; c < DCH; ++c) {
                 frag_ab bQ = *reinterpret_cast<const frag_ab*>(
-                    &q_lds[swz128(ms * 16 + lcol, c * 32 + 8 * lgrp)]);
+                    &q_lds[swz_rm<RMS>(ms * 16 + lcol, c * 32 + 8 * lgrp)]);
                 st = MFMA_16x16x32(aK[c], bQ, st);          // S^T = K·Q^T
                 frag_ab bDO = *reinterpret_cast<const frag_ab*>(
-                    &do_lds[swz128(ms * 16 + lcol, c * 32 + 8 * lgrp)]);
+                    &do_lds[swz_rm<RMS>(ms * 16 + lcol, c * 32 + 8 * lgrp)]);
                 dpt = MFMA_16x16x32(aV[c], bDO, dpt);       // dP^T = V·dO^T
             }
 #pragma unroll
@@ -1007,8 +1066,9 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
     constexpr int QROWS = W * 16;          // q rows per workgroup
 
     // swizzled strides, same scheme as dkdv (see comment there)
-    __shared__ bf16 k_lds[KVR * 128];       // row-major K (QK^T B-frags)
-    __shared__ bf16 v_lds[KVR * 128];       // row-major V (dP B-frags)
+    constexpr int RMS = (D == 64) ? 64 : 128;
+    __shared__ bf16 k_lds[KVR * RMS];       // row-major K (QK^T B-frags)
+    __shared__ bf16 v_lds[KVR * RMS];       // row-major V (dP B-frags)
     __shared__ bf16 kt_lds[D * 64];         // transposed K (dQ B-frags)
     __shared__ bf16 ds_lds[W][16 * 64];     // dS (m rows, n cols)
 
@@ -1071,10 +1131,10 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
                     vp + (int64_t)(kv0 + row) * vs2 + col);
                 frag_ab pv1 = *reinterpret_cast<const frag_ab*>(
                     vp + (int64_t)(kv0 + row + 1) * vs2 + col);
-                *reinterpret_cast<frag_ab*>(&k_lds[swz128(row, col)]) = pk0;
-                *reinterpret_cast<frag_ab*>(&k_lds[swz128(row + 1, col)]) = pk1;
-                *reinterpret_cast<frag_ab*>(&v_lds[swz128(row, col)]) = pv0;
-                *reinterpret_cast<frag_ab*>(&v_lds[swz128(row + 1, col)]) = pv1;
+                *reinterpret_cast<frag_ab*>(&k_lds[swz_rm<RMS>(row, col)]) = pk0;
+                *reinterpret_cast<frag_ab*>(&k_lds[swz_rm<RMS>(row + 1, col)]) = pk1;
+                *reinterpret_cast<frag_ab*>(&v_lds[swz_rm<RMS>(row, col)]) = pv0;
+                *reinterpret_cast<frag_ab*>(&v_lds[swz_rm<RMS>(row + 1, col)]) = pv1;
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
                     union { __bf16 h[2]; uint32_t u; } a;
@@ -1093,10 +1153,10 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
 #pragma unroll
             for (int c = 0; c < DCH; ++c) {
                 frag_ab bK = *reinterpret_cast<const frag_ab*>(
-                    &k_lds[swz128(ns * 16 + lcol, c * 32 + 8 * lgrp)]);
+                    &k_lds[swz_rm<RMS>(ns * 16 + lcol, c * 32 + 8 * lgrp)]);
                 sacc = MFMA_16x16x32(aQ[c], bK, sacc);      // S = Q·K^T
                 frag_ab bV = *reinterpret_cast<const frag_ab*>(
-                    &v_lds[swz128(ns * 16 + lcol, c * 32 + 8 * lgrp)]);
+                    &v_lds[swz_rm<RMS>(ns * 16 + lcol, c * 32 + 8 * lgrp)]);
                 dpacc = MFMA_16x16x32(aDO[c], bV, dpacc);   // dP = dO·V^T
             }
 #pragma unroll

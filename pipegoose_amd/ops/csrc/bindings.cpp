@@ -34,6 +34,7 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
                    int64_t kv_off);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor Bt);
 torch::Tensor mfma_probe32(torch::Tensor A, torch::Tensor Bt);
+torch::Tensor tr16_probe(torch::Tensor tile);
 std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w,
                                         double eps);
 std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
@@ -67,6 +68,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "16x16x32 bf16 MFMA fragment-layout probe");
     m.def("mfma_probe32", &mfma_probe32,
           "32x32x16 bf16 MFMA fragment-layout probe");
+    m.def("tr16_probe", &tr16_probe,
+          "ds_read_b64_tr_b16 semantics probe ([4][16] tile -> per-lane)");
     m.def("rms_norm_fwd", &rms_norm_fwd, "fused RMSNorm forward (gfx950)");
     m.def("rms_norm_bwd", &rms_norm_bwd, "fused RMSNorm backward (gfx950)");
     m.def("rope_apply", &rope_apply,

@@ -255,6 +255,18 @@ def test_mfma_probe32_layout():
     assert torch.allclose(C, ref, atol=1e-2), (C - ref).abs().max()
 
 
+def test_tr16_probe_semantics():
+    """ds_read_b64_tr_b16: lane l of a quarter-wave must receive column l&15
+    of the [4][16] bf16 tile its quarter covers."""
+    ext = _ext()
+    if not hasattr(ext, "tr16_probe"):
+        pytest.skip("extension predates tr16 probe")
+    tile = torch.arange(64, device="cuda", dtype=torch.float32).reshape(4, 16)
+    got = ext.tr16_probe(tile.to(torch.bfloat16).flatten())
+    want = tile.t().contiguous()  # lane l -> column l, elems j -> rows
+    assert torch.equal(got, want), got
+
+
 def _attn_oracle(q, k, v, slopes, scale):
     S = q.size(-2)
     pos = torch.arange(S, device=q.device)
