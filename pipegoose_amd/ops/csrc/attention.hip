@@ -318,6 +318,17 @@ __device__ __forceinline__ int swz_rm(int row, int elem_col) {
     return RS == 64 ? swz64(row, elem_col) : swz128(row, elem_col);
 }
 
+// Backward-tile indexers.  Measured split (r2 A/B): for D=128 the 256-B
+// swizzled rows beat the padded layout (5797->5474us); for D=64 the padded
+// odd strides (72/40) are already conflict-free on the fragment reads and
+// swizzled power-of-2 rows REGRESSED (1206->1606us) - keep v1's layout.
+template <int D> __device__ __forceinline__ int bwd_rm(int row, int col) {
+    return D == 64 ? row * 72 + col : swz128(row, col);
+}
+template <int D> __device__ __forceinline__ int bwd_tr(int row, int col) {
+    return D == 64 ? row * 40 + col : swz64(row, col);
+}
+
 // C/D layout of mfma_f32_32x32x16_bf16: col = lane&31, row = (r&3) +
 // 8*(r>>2) + 4*(lane>>5), r in [0,16).  A/B fragments: i/j = lane&31,
 // k = 8*(lane>>5) + elem (verified by mfma_probe32).
@@ -892,13 +903,14 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
     // swz64 for D=64, keeping the LDS footprint and occupancy of r1 —
     // transposed tiles on 128-B rows via swz64) replace the +8-element
     // padding: r1 PMC measured 33-63% of LDS cycles in bank conflicts.
-    constexpr int RMS = (D == 64) ? 64 : 128;   // row-major tile stride
+    constexpr int RMS = (D == 64) ? 72 : 128;   // row-major tile stride
+    constexpr int TRS = (D == 64) ? 40 : 64;    // transposed tile stride
     __shared__ bf16 do_lds[QR * RMS];
     __shared__ bf16 q_lds[QR * RMS];
-    __shared__ bf16 qt_lds[D * 64];
-    __shared__ bf16 dot_lds[D * 64];
-    __shared__ bf16 pt_lds[W][16 * 64];   // P^T  (n rows, m cols)
-    __shared__ bf16 dst_lds[W][16 * 64];  // dS^T (n rows, m cols)
+    __shared__ bf16 qt_lds[D * TRS];
+    __shared__ bf16 dot_lds[D * TRS];
+    __shared__ bf16 pt_lds[W][16 * TRS];   // P^T  (n rows, m cols)
+    __shared__ bf16 dst_lds[W][16 * TRS];  // dS^T (n rows, m cols)
     __shared__ float lse_lds[QR];
     __shared__ float delta_lds[QR];
 
@@ -959,17 +971,17 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
                     dop + (int64_t)(q0 + row) * gs + col);
                 frag_ab pd1 = *reinterpret_cast<const frag_ab*>(
                     dop + (int64_t)(q0 + row + 1) * gs + col);
-                *reinterpret_cast<frag_ab*>(&q_lds[swz_rm<RMS>(row, col)]) = pq0;
-                *reinterpret_cast<frag_ab*>(&q_lds[swz_rm<RMS>(row + 1, col)]) = pq1;
-                *reinterpret_cast<frag_ab*>(&do_lds[swz_rm<RMS>(row, col)]) = pd0;
-                *reinterpret_cast<frag_ab*>(&do_lds[swz_rm<RMS>(row + 1, col)]) = pd1;
+                *reinterpret_cast<frag_ab*>(&q_lds[bwd_rm<D>(row, col)]) = pq0;
+                *reinterpret_cast<frag_ab*>(&q_lds[bwd_rm<D>(row + 1, col)]) = pq1;
+                *reinterpret_cast<frag_ab*>(&do_lds[bwd_rm<D>(row, col)]) = pd0;
+                *reinterpret_cast<frag_ab*>(&do_lds[bwd_rm<D>(row + 1, col)]) = pd1;
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
                     union { __bf16 h[2]; uint32_t u; } a, b2;
                     a.h[0] = pq0[j]; a.h[1] = pq1[j];
                     b2.h[0] = pd0[j]; b2.h[1] = pd1[j];
-                    *reinterpret_cast<uint32_t*>(&qt_lds[swz64(col + j, row)]) = a.u;
-                    *reinterpret_cast<uint32_t*>(&dot_lds[swz64(col + j, row)]) = b2.u;
+                    *reinterpret_cast<uint32_t*>(&qt_lds[bwd_tr<D>(col + j, row)]) = a.u;
+                    *reinterpret_cast<uint32_t*>(&dot_lds[bwd_tr<D>(col + j, row)]) = b2.u;
                 }
             }
             for (int i = tid; i < QR; i += NT) {
@@ -987,10 +999,10 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
 #pragma unroll
             for (int c = 0; c < DCH; ++c) {
                 frag_ab bQ = *reinterpret_cast<const frag_ab*>(
-                    &q_lds[swz_rm<RMS>(ms * 16 + lcol, c * 32 + 8 * lgrp)]);
+                    &q_lds[bwd_rm<D>(ms * 16 + lcol, c * 32 + 8 * lgrp)]);
                 st = MFMA_16x16x32(aK[c], bQ, st);          // S^T = K·Q^T
                 frag_ab bDO = *reinterpret_cast<const frag_ab*>(
-                    &do_lds[swz_rm<RMS>(ms * 16 + lcol, c * 32 + 8 * lgrp)]);
+                    &do_lds[bwd_rm<D>(ms * 16 + lcol, c * 32 + 8 * lgrp)]);
                 dpt = MFMA_16x16x32(aV[c], bDO, dpt);       // dP^T = V·dO^T
             }
 #pragma unroll
@@ -1001,8 +1013,8 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
                 float z = st[r] * scale2 + slope2 * (float)(n_glob - m_glob);
                 float p = (n_glob <= m_glob) ? exp2f(z - lse_lds[m_loc]) : 0.f;
                 float ds = p * (dpt[r] - delta_lds[m_loc]) * scale;
-                pt_lds[wave][swz64(4 * lgrp + r, m_loc)] = (bf16)p;
-                dst_lds[wave][swz64(4 * lgrp + r, m_loc)] = (bf16)ds;
+                pt_lds[wave][bwd_tr<D>(4 * lgrp + r, m_loc)] = (bf16)p;
+                dst_lds[wave][bwd_tr<D>(4 * lgrp + r, m_loc)] = (bf16)ds;
             }
         }
         __builtin_amdgcn_s_waitcnt(0);  // own-wave LDS writes visible
@@ -1011,16 +1023,16 @@ void attn_bwd_dkdv_kernel(const bf16* __restrict__ dout,
 #pragma unroll
         for (int kc = 0; kc < QR / 32; ++kc) {
             frag_ab aPT = *reinterpret_cast<const frag_ab*>(
-                &pt_lds[wave][swz64(lcol, kc * 32 + 8 * lgrp)]);
+                &pt_lds[wave][bwd_tr<D>(lcol, kc * 32 + 8 * lgrp)]);
             frag_ab aDST = *reinterpret_cast<const frag_ab*>(
-                &dst_lds[wave][swz64(lcol, kc * 32 + 8 * lgrp)]);
+                &dst_lds[wave][bwd_tr<D>(lcol, kc * 32 + 8 * lgrp)]);
 #pragma unroll
             for (int ds = 0; ds < DSUB; ++ds) {
                 frag_ab bDOT = *reinterpret_cast<const frag_ab*>(
-                    &dot_lds[swz64(ds * 16 + lcol, kc * 32 + 8 * lgrp)]);
+                    &dot_lds[bwd_tr<D>(ds * 16 + lcol, kc * 32 + 8 * lgrp)]);
                 accDV[ds] = MFMA_16x16x32(aPT, bDOT, accDV[ds]);
                 frag_ab bQT = *reinterpret_cast<const frag_ab*>(
-                    &qt_lds[swz64(ds * 16 + lcol, kc * 32 + 8 * lgrp)]);
+                    &qt_lds[bwd_tr<D>(ds * 16 + lcol, kc * 32 + 8 * lgrp)]);
                 accDK[ds] = MFMA_16x16x32(aDST, bQT, accDK[ds]);
             }
         }
@@ -1066,11 +1078,12 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
     constexpr int QROWS = W * 16;          // q rows per workgroup
 
     // swizzled strides, same scheme as dkdv (see comment there)
-    constexpr int RMS = (D == 64) ? 64 : 128;
+    constexpr int RMS = (D == 64) ? 72 : 128;
+    constexpr int TRS = (D == 64) ? 40 : 64;
     __shared__ bf16 k_lds[KVR * RMS];       // row-major K (QK^T B-frags)
     __shared__ bf16 v_lds[KVR * RMS];       // row-major V (dP B-frags)
-    __shared__ bf16 kt_lds[D * 64];         // transposed K (dQ B-frags)
-    __shared__ bf16 ds_lds[W][16 * 64];     // dS (m rows, n cols)
+    __shared__ bf16 kt_lds[D * TRS];        // transposed K (dQ B-frags)
+    __shared__ bf16 ds_lds[W][16 * TRS];    // dS (m rows, n cols)
 
     const int qb = blockIdx.x;
     const int h = blockIdx.y;
@@ -1131,15 +1144,15 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
                     vp + (int64_t)(kv0 + row) * vs2 + col);
                 frag_ab pv1 = *reinterpret_cast<const frag_ab*>(
                     vp + (int64_t)(kv0 + row + 1) * vs2 + col);
-                *reinterpret_cast<frag_ab*>(&k_lds[swz_rm<RMS>(row, col)]) = pk0;
-                *reinterpret_cast<frag_ab*>(&k_lds[swz_rm<RMS>(row + 1, col)]) = pk1;
-                *reinterpret_cast<frag_ab*>(&v_lds[swz_rm<RMS>(row, col)]) = pv0;
-                *reinterpret_cast<frag_ab*>(&v_lds[swz_rm<RMS>(row + 1, col)]) = pv1;
+                *reinterpret_cast<frag_ab*>(&k_lds[bwd_rm<D>(row, col)]) = pk0;
+                *reinterpret_cast<frag_ab*>(&k_lds[bwd_rm<D>(row + 1, col)]) = pk1;
+                *reinterpret_cast<frag_ab*>(&v_lds[bwd_rm<D>(row, col)]) = pv0;
+                *reinterpret_cast<frag_ab*>(&v_lds[bwd_rm<D>(row + 1, col)]) = pv1;
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
                     union { __bf16 h[2]; uint32_t u; } a;
                     a.h[0] = pk0[j]; a.h[1] = pk1[j];
-                    *reinterpret_cast<uint32_t*>(&kt_lds[swz64(col + j, row)]) = a.u;
+                    *reinterpret_cast<uint32_t*>(&kt_lds[bwd_tr<D>(col + j, row)]) = a.u;
                 }
             }
         }
@@ -1153,10 +1166,10 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
 #pragma unroll
             for (int c = 0; c < DCH; ++c) {
                 frag_ab bK = *reinterpret_cast<const frag_ab*>(
-                    &k_lds[swz_rm<RMS>(ns * 16 + lcol, c * 32 + 8 * lgrp)]);
+                    &k_lds[bwd_rm<D>(ns * 16 + lcol, c * 32 + 8 * lgrp)]);
                 sacc = MFMA_16x16x32(aQ[c], bK, sacc);      // S = Q·K^T
                 frag_ab bV = *reinterpret_cast<const frag_ab*>(
-                    &v_lds[swz_rm<RMS>(ns * 16 + lcol, c * 32 + 8 * lgrp)]);
+                    &v_lds[bwd_rm<D>(ns * 16 + lcol, c * 32 + 8 * lgrp)]);
                 dpacc = MFMA_16x16x32(aDO[c], bV, dpacc);   // dP = dO·V^T
             }
 #pragma unroll
@@ -1167,7 +1180,7 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
                 float p = (jk <= iq) ? exp2f(z - lse_r[r]) : 0.f;
                 float ds = p * (dpacc[r] - delta_r[r]) * scale;
                 // store transposed to [m][n] so the dQ A-frag read is linear
-                ds_lds[wave][swz64(4 * lgrp + r, ns * 16 + lcol)] = (bf16)ds;
+                ds_lds[wave][bwd_tr<D>(4 * lgrp + r, ns * 16 + lcol)] = (bf16)ds;
             }
         }
         __builtin_amdgcn_s_waitcnt(0);
@@ -1176,11 +1189,11 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
 #pragma unroll
         for (int kc = 0; kc < KVR / 32; ++kc) {
             frag_ab aDS = *reinterpret_cast<const frag_ab*>(
-                &ds_lds[wave][swz64(lcol, kc * 32 + 8 * lgrp)]);
+                &ds_lds[wave][bwd_tr<D>(lcol, kc * 32 + 8 * lgrp)]);
 #pragma unroll
             for (int ds = 0; ds < DSUB; ++ds) {
                 frag_ab bKT = *reinterpret_cast<const frag_ab*>(
-                    &kt_lds[swz64(ds * 16 + lcol, kc * 32 + 8 * lgrp)]);
+                    &kt_lds[bwd_tr<D>(ds * 16 + lcol, kc * 32 + 8 * lgrp)]);
                 accDQ[ds] = MFMA_16x16x32(aDS, bKT, accDQ[ds]);
             }
         }

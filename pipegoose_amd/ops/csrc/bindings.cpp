@@ -35,6 +35,15 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor Bt);
 torch::Tensor mfma_probe32(torch::Tensor A, torch::Tensor Bt);
 torch::Tensor tr16_probe(torch::Tensor tile);
+torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor B,
+                      c10::optional<torch::Tensor> bias, bool gelu);
+void adamw_fused_step(std::vector<torch::Tensor> params,
+                      std::vector<torch::Tensor> grads,
+                      std::vector<torch::Tensor> exp_avgs,
+                      std::vector<torch::Tensor> exp_avg_sqs,
+                      torch::Tensor meta_dev, int64_t n_slabs,
+                      double lr, double beta1, double beta2, double eps,
+                      double weight_decay, int64_t step);
 std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w,
                                         double eps);
 std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x,
@@ -70,6 +79,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "32x32x16 bf16 MFMA fragment-layout probe");
     m.def("tr16_probe", &tr16_probe,
           "ds_read_b64_tr_b16 semantics probe ([4][16] tile -> per-lane)");
+    m.def("gemm_bt", &gemm_bt,
+          "hand-written MFMA bf16 GEMM C = A @ B^T (+bias, +gelu) (gfx950)",
+          py::arg("A"), py::arg("B"), py::arg("bias") = c10::nullopt,
+          py::arg("gelu") = false);
+    m.def("adamw_fused_step", &adamw_fused_step,
+          "whole AdamW update, one chunked kernel per step (gfx950)");
     m.def("rms_norm_fwd", &rms_norm_fwd, "fused RMSNorm forward (gfx950)");
     m.def("rms_norm_bwd", &rms_norm_bwd, "fused RMSNorm backward (gfx950)");
     m.def("rope_apply", &rope_apply,

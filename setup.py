@@ -23,6 +23,8 @@ SRC = [
     "pipegoose_amd/ops/csrc/rope.hip",
     "pipegoose_amd/ops/csrc/router.hip",
     "pipegoose_amd/ops/csrc/silu_mul.hip",
+    "pipegoose_amd/ops/csrc/adamw.hip",
+    "pipegoose_amd/ops/csrc/gemm.hip",
 ]
 
 setup(

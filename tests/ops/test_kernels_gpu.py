@@ -255,6 +255,31 @@ def test_mfma_probe32_layout():
     assert torch.allclose(C, ref, atol=1e-2), (C - ref).abs().max()
 
 
+@pytest.mark.parametrize("shape", [(128, 128, 64), (256, 384, 192),
+                                   (512, 256, 1024)])
+@pytest.mark.parametrize("bias,gelu", [(False, False), (True, False),
+                                       (True, True)])
+def test_gemm_bt_kernel(shape, bias, gelu):
+    """Hand MFMA GEMM vs fp32 torch oracle (asymmetric random operands)."""
+    ext = _ext()
+    if not hasattr(ext, "gemm_bt"):
+        pytest.skip("extension predates gemm_bt")
+    torch.manual_seed(21)
+    M, N, K = shape
+    A = (torch.rand(M, K, device="cuda") * 2 - 1).bfloat16()
+    B = (torch.rand(N, K, device="cuda") * 2 - 1).bfloat16()
+    b = torch.randn(N, device="cuda") if bias else None
+    C = ext.gemm_bt(A, B, b, gelu)
+    ref = A.float() @ B.float().t()
+    if bias:
+        ref = ref + b.float()
+    if gelu:
+        ref = torch.nn.functional.gelu(ref, approximate="tanh")
+    err = (C.float() - ref).abs().max()
+    tol = 0.02 * max(1.0, ref.abs().max().item())
+    assert err < tol, (err, ref.abs().max())
+
+
 def test_tr16_probe_semantics():
     """ds_read_b64_tr_b16: lane l of a quarter-wave must receive column l&15
     of the [4][16] bf16 tile its quarter covers."""
