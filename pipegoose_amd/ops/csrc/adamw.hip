@@ -24,7 +24,7 @@ namespace {
 
 constexpr long SLAB = 65536;  // elements per workgroup
 
-template <typename T>
+template <typename T, typename TS>
 __global__ __launch_bounds__(256)
 void adamw_chunked_kernel(const unsigned long long* __restrict__ ptrs,
                           const int* __restrict__ slab_tensor,
@@ -39,8 +39,8 @@ void adamw_chunked_kernel(const unsigned long long* __restrict__ ptrs,
     const int t = slab_tensor[s];
     T* p = reinterpret_cast<T*>(ptrs[t]);
     const T* g = reinterpret_cast<const T*>(ptrs[n_tensors + t]);
-    float* m = reinterpret_cast<float*>(ptrs[2 * n_tensors + t]);
-    float* v = reinterpret_cast<float*>(ptrs[3 * n_tensors + t]);
+    TS* m = reinterpret_cast<TS*>(ptrs[2 * n_tensors + t]);
+    TS* v = reinterpret_cast<TS*>(ptrs[3 * n_tensors + t]);
     const long n = numels[t];
     const long base = (long)slab_idx[s] * SLAB;
     const long end = min(base + SLAB, n);
@@ -51,32 +51,31 @@ void adamw_chunked_kernel(const unsigned long long* __restrict__ ptrs,
     for (; i + 8 <= end; i += (long)blockDim.x * 8) {
         typename vec8<T>::type pv8 = *reinterpret_cast<const typename vec8<T>::type*>(p + i);
         typename vec8<T>::type gv8 = *reinterpret_cast<const typename vec8<T>::type*>(g + i);
-        f32x8 mv = *reinterpret_cast<const f32x8*>(m + i);
-        f32x8 vv = *reinterpret_cast<const f32x8*>(v + i);
-        // fp32 writes are 32 B/lane: split into two 16-B halves
+        typename vec8<TS>::type mv = *reinterpret_cast<const typename vec8<TS>::type*>(m + i);
+        typename vec8<TS>::type vv = *reinterpret_cast<const typename vec8<TS>::type*>(v + i);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
             const float gj = to_float(gv8.v[j]);
-            const float mj = beta1 * mv.v[j] + c1 * gj;
-            const float vj = beta2 * vv.v[j] + c2 * gj * gj;
-            mv.v[j] = mj;
-            vv.v[j] = vj;
+            const float mj = beta1 * to_float(mv.v[j]) + c1 * gj;
+            const float vj = beta2 * to_float(vv.v[j]) + c2 * gj * gj;
+            mv.v[j] = from_float<TS>(mj);
+            vv.v[j] = from_float<TS>(vj);
             const float denom = sqrtf(vj) * bc2_rsqrt + eps;
             pv8.v[j] = from_float<T>(
                 to_float(pv8.v[j]) * wd_factor - step_size * mj / denom);
         }
         *reinterpret_cast<typename vec8<T>::type*>(p + i) = pv8;
-        *reinterpret_cast<f32x8*>(m + i) = mv;
-        *reinterpret_cast<f32x8*>(v + i) = vv;
+        *reinterpret_cast<typename vec8<TS>::type*>(m + i) = mv;
+        *reinterpret_cast<typename vec8<TS>::type*>(v + i) = vv;
     }
     // handle the ragged tail with the whole block, element-wise
     const long vec_end = base + ((end - base) / 8) * 8;
     for (long k = vec_end + threadIdx.x; k < end; k += blockDim.x) {
         const float gj = to_float(g[k]);
-        const float mj = beta1 * m[k] + c1 * gj;
-        const float vj = beta2 * v[k] + c2 * gj * gj;
-        m[k] = mj;
-        v[k] = vj;
+        const float mj = beta1 * to_float(m[k]) + c1 * gj;
+        const float vj = beta2 * to_float(v[k]) + c2 * gj * gj;
+        m[k] = from_float<TS>(mj);
+        v[k] = from_float<TS>(vj);
         const float denom = sqrtf(vj) * bc2_rsqrt + eps;
         p[k] = from_float<T>(to_float(p[k]) * wd_factor - step_size * mj / denom);
     }
@@ -95,6 +94,7 @@ void adamw_fused_step(std::vector<torch::Tensor> params,
     TORCH_CHECK(!params.empty());
     const int T_ = (int)params.size();
     const auto dt = params[0].scalar_type();
+    const auto st_dt = exp_avgs[0].scalar_type();
     const float bc1 = 1.0f - powf((float)beta1, (float)step);
     const float bc2 = 1.0f - powf((float)beta2, (float)step);
     const float step_size = (float)lr / bc1;
@@ -111,18 +111,21 @@ void adamw_fused_step(std::vector<torch::Tensor> params,
 
     auto stream = at::cuda::getCurrentCUDAStream();
     dim3 grid((unsigned)n_slabs);
-    if (dt == torch::kBFloat16) {
-        hipLaunchKernelGGL((adamw_chunked_kernel<__hip_bfloat16>), grid,
-            dim3(256), 0, stream, ptrs, slab_tensor, slab_idx, numels, T_,
-            (float)lr, (float)beta1, (float)beta2, (float)eps,
-            wd_factor, step_size, bc2_rsqrt);
+#define LAUNCH_ADAMW(TP, TSP)                                                 \
+    hipLaunchKernelGGL((adamw_chunked_kernel<TP, TSP>), grid, dim3(256), 0,   \
+        stream, ptrs, slab_tensor, slab_idx, numels, T_, (float)lr,           \
+        (float)beta1, (float)beta2, (float)eps, wd_factor, step_size,         \
+        bc2_rsqrt)
+    if (dt == torch::kBFloat16 && st_dt == torch::kFloat) {
+        LAUNCH_ADAMW(__hip_bfloat16, float);
+    } else if (dt == torch::kBFloat16 && st_dt == torch::kBFloat16) {
+        LAUNCH_ADAMW(__hip_bfloat16, __hip_bfloat16);
     } else {
-        TORCH_CHECK(dt == torch::kFloat, "adamw_fused_step: bf16/f32 only");
-        hipLaunchKernelGGL((adamw_chunked_kernel<float>), grid,
-            dim3(256), 0, stream, ptrs, slab_tensor, slab_idx, numels, T_,
-            (float)lr, (float)beta1, (float)beta2, (float)eps,
-            wd_factor, step_size, bc2_rsqrt);
+        TORCH_CHECK(dt == torch::kFloat && st_dt == torch::kFloat,
+                    "adamw_fused_step: p bf16/f32, state p-dtype or f32");
+        LAUNCH_ADAMW(float, float);
     }
+#undef LAUNCH_ADAMW
     HIP_CHECK_LAUNCH();
 }
 

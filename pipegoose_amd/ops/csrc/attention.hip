@@ -1212,6 +1212,555 @@ void attn_bwd_dq_kernel(const bf16* __restrict__ dout,
 
 }  // namespace
 
+// ======================================================================
+// Backward v2 — same ladder as the forward (32x32 MFMA, lane-local
+// layouts, swizzled/subtiled LDS, exp2 domain).  Two kernels:
+//   dq_v2   (q-outer):  wave owns 32 q rows; K/V tiles staged; St^T and
+//           dP^T have q on the lane axis (lse/delta are per-lane scalars),
+//           dS packs into B-fragments and dQ^T accumulates lane-locally.
+//   dkdv_v2 (kv-outer): wave owns 32 kv rows (K/V B-fragments live in
+//           registers); Q/dO tiles staged in the sub-tiled image, which
+//           serves BOTH the row-major A-fragments (St, dP) and the
+//           hardware-transpose reads (dV^T = dO^T·P, dK^T = Q^T·dS).
+// The sub-tiled image is [rows/4][d/16][4][16] with the d-block position
+// XORed by the row-group (tiles stay 128-B contiguous for tr16 while the
+// row-major reads spread across banks).
+namespace {
+
+// element index into a sub-tiled [R][D] image; DBLK = D/16
+template <int D>
+__device__ __forceinline__ int sub_idx(int row, int elem_col) {
+    constexpr int DBLK = D / 16;
+    const int g = row >> 2;
+    const int dblk = (elem_col >> 4) ^ (g & (DBLK - 1));
+    return g * (DBLK * 64) + dblk * 64 + (row & 3) * 16 + (elem_col & 15);
+}
+
+// byte-base of the [4][16] tr16 tile for (row group g, logical d-block)
+template <int D>
+__device__ __forceinline__ int sub_tile_base(int g, int dblk_log) {
+    constexpr int DBLK = D / 16;
+    return g * (DBLK * 64) + (dblk_log ^ (g & (DBLK - 1))) * 64;
+}
+
+template <int D, int WAVES>
+__global__ __launch_bounds__(WAVES * WAVE_SIZE)
+void attn_bwd_dq_v2_kernel(const bf16* __restrict__ dout,
+                           const bf16* __restrict__ q, const bf16* __restrict__ k,
+                           const bf16* __restrict__ v,
+                           const float* __restrict__ lse,
+                           const float* __restrict__ delta,
+                           const float* __restrict__ slopes, float scale,
+                           bf16* __restrict__ dq,
+                           int B, int H, int S, int kv_off, int kv_group,
+                           int64_t gb, int64_t gh, int64_t gs,
+                           int64_t qb2, int64_t qh2, int64_t qs2,
+                           int64_t kb2, int64_t kh2, int64_t ks2,
+                           int64_t vb2, int64_t vh2, int64_t vs2,
+                           int64_t wb, int64_t wh, int64_t ws) {
+    constexpr int KVB = 64;
+    constexpr int NT = WAVES * WAVE_SIZE;
+    constexpr int ROWS = WAVES * 32;
+    constexpr int DCH = D / 16;
+    constexpr int DSUB = D / 32;
+    constexpr int VST = (D / 16) * 64;
+    constexpr int NP = (KVB / 2) * (D / 8);     // row-pair packets per tile
+    constexpr int PPL = (NP + NT - 1) / NT;
+
+    __shared__ __attribute__((aligned(16))) bf16 ks_lds[2][KVB * D];
+    __shared__ __attribute__((aligned(16))) bf16 v_lds[2][KVB * 128];
+
+    const int qblock = blockIdx.x;
+    const int h = blockIdx.y;
+    const int b = blockIdx.z;
+    const int tid = threadIdx.x;
+    const int wave = tid / WAVE_SIZE;
+    const int lane = tid % WAVE_SIZE;
+    const int l31 = lane & 31;
+    const int hi = lane >> 5;
+    const int lq = lane >> 4;
+
+    const int64_t bh_off = ((int64_t)b * H + h) * S;
+    const int hk = h / kv_group;
+    const bf16* kp = k + b * kb2 + hk * kh2;
+    const bf16* vp = v + b * vb2 + hk * vh2;
+    const float LOG2E = 1.4426950408889634f;
+    const float scale2 = scale * LOG2E;
+    const float slope2 = slopes[h] * LOG2E;
+
+    const int qr0 = qblock * ROWS + wave * 32;
+    const int iq = qr0 + l31;
+
+    frag_ab bQ[DCH], bDO[DCH];
+    {
+        const bf16* qrow = q + b * qb2 + h * qh2 + (int64_t)iq * qs2;
+        const bf16* dorow = dout + b * gb + h * gh + (int64_t)iq * gs;
+#pragma unroll
+        for (int c = 0; c < DCH; ++c) {
+            bQ[c] = *reinterpret_cast<const frag_ab*>(qrow + c * 16 + 8 * hi);
+            bDO[c] = *reinterpret_cast<const frag_ab*>(dorow + c * 16 + 8 * hi);
+        }
+    }
+    const float lse2 = lse[bh_off + iq] * LOG2E;
+    const float dlt = delta[bh_off + iq];
+
+    f32x16 accDQ[DSUB];
+#pragma unroll
+    for (int dsb = 0; dsb < DSUB; ++dsb) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) accDQ[dsb][r] = 0.f;
+    }
+
+    const bool full_vis = (kv_off <= -S);
+    const int kv_end = full_vis ? S
+        : min(S, qblock * ROWS + ROWS - kv_off);
+    const int nkv = max(0, (kv_end + KVB - 1) / KVB);
+
+    frag_ab kreg[PPL][2], vreg[PPL][2];
+    auto stage_load = [&](int nb) {
+        const int kvrow0 = nb * KVB;
+#pragma unroll
+        for (int i = 0; i < PPL; ++i) {
+            const int p = tid + i * NT;
+            if (p < NP) {
+                const int row = (p / (D / 8)) * 2;
+                const int col = (p % (D / 8)) * 8;
+                kreg[i][0] = *reinterpret_cast<const frag_ab*>(
+                    kp + (int64_t)(kvrow0 + row) * ks2 + col);
+                kreg[i][1] = *reinterpret_cast<const frag_ab*>(
+                    kp + (int64_t)(kvrow0 + row + 1) * ks2 + col);
+                vreg[i][0] = *reinterpret_cast<const frag_ab*>(
+                    vp + (int64_t)(kvrow0 + row) * vs2 + col);
+                vreg[i][1] = *reinterpret_cast<const frag_ab*>(
+                    vp + (int64_t)(kvrow0 + row + 1) * vs2 + col);
+            }
+        }
+    };
+    auto stage_write = [&](int buf) {
+#pragma unroll
+        for (int i = 0; i < PPL; ++i) {
+            const int p = tid + i * NT;
+            if (p < NP) {
+                const int row = (p / (D / 8)) * 2;
+                const int col = (p % (D / 8)) * 8;
+#pragma unroll
+                for (int rr = 0; rr < 2; ++rr) {
+                    *reinterpret_cast<frag_ab*>(
+                        &ks_lds[buf][sub_idx<D>(row + rr, col)]) = kreg[i][rr];
+                    *reinterpret_cast<frag_ab*>(
+                        &v_lds[buf][swz128(row + rr, col)]) = vreg[i][rr];
+                }
+            }
+        }
+    };
+
+    constexpr bool EARLY_LOAD = true;
+    if (nkv > 0) {
+        stage_load(0);
+        stage_write(0);
+    }
+    __syncthreads();
+
+    for (int nb = 0; nb < nkv; ++nb) {
+        const int cur = nb & 1;
+        const int kvrow0 = nb * KVB;
+        const bool last = (nb + 1 == nkv);
+        if (EARLY_LOAD && !last) stage_load(nb + 1);
+
+        const bool active = full_vis || (kvrow0 + kv_off <= qr0 + 31);
+        if (active) {
+#pragma unroll
+            for (int ns = 0; ns < 2; ++ns) {
+                frag_ab dsfrag[2];
+                f32x16 st, dp;
+#pragma unroll
+                for (int r = 0; r < 16; ++r) { st[r] = 0.f; dp[r] = 0.f; }
+                __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                for (int c = 0; c < DCH; ++c) {
+                    // St^T = K·Q^T: A = K row-major from the sub-tiled image
+                    frag_ab aK = *reinterpret_cast<const frag_ab*>(
+                        &ks_lds[cur][sub_idx<D>(ns * 32 + l31, c * 16 + 8 * hi)]);
+                    st = MFMA_32x32x16(aK, bQ[c], st);
+                    frag_ab aV = *reinterpret_cast<const frag_ab*>(
+                        &v_lds[cur][swz128(ns * 32 + l31, c * 16 + 8 * hi)]);
+                    dp = MFMA_32x32x16(aV, bDO[c], dp);  // dP^T = V·dO^T
+                }
+                __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const int jk = kvrow0 + ns * 32 + (r & 3) + 8 * (r >> 2) +
+                                   4 * hi + kv_off;
+                    const float z2 = st[r] * scale2 +
+                                     slope2 * (float)(jk - iq);
+                    const float p = (jk <= iq) ? exp2f(z2 - lse2) : 0.f;
+                    st[r] = p * (dp[r] - dlt) * scale;  // dS^T
+                }
+                // pack dS^T into B-fragments (k = kv), as in the forward
+#pragma unroll
+                for (int hf = 0; hf < 2; ++hf) {
+                    union { __bf16 h2[2]; uint32_t u; } wa, wb2, wc, wd;
+                    wa.h2[0] = (__bf16)st[8 * hf + 0];
+                    wa.h2[1] = (__bf16)st[8 * hf + 1];
+                    wc.h2[0] = (__bf16)st[8 * hf + 2];
+                    wc.h2[1] = (__bf16)st[8 * hf + 3];
+                    wb2.h2[0] = (__bf16)st[8 * hf + 4];
+                    wb2.h2[1] = (__bf16)st[8 * hf + 5];
+                    wd.h2[0] = (__bf16)st[8 * hf + 6];
+                    wd.h2[1] = (__bf16)st[8 * hf + 7];
+                    auto s1 = __builtin_amdgcn_permlane32_swap(wa.u, wb2.u,
+                                                              false, false);
+                    auto s2 = __builtin_amdgcn_permlane32_swap(wc.u, wd.u,
+                                                              false, false);
+                    union { frag_ab f; uint32_t u[4]; } out;
+                    out.u[0] = s1[0];
+                    out.u[1] = s2[0];
+                    out.u[2] = s1[1];
+                    out.u[3] = s2[1];
+                    dsfrag[hf] = out.f;
+                }
+                // dQ^T += K^T·dS^T for this ns's two kv chunks (keeps the
+                // fragment set at 2 so D=128 stays under the VGPR cap)
+#pragma unroll
+                for (int dsb = 0; dsb < DSUB; ++dsb) {
+                    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                    for (int ksq = 0; ksq < 2; ++ksq) {
+                        const int g0 = 8 * ns + 4 * ksq + 2 * hi;
+                        const int dblk = 2 * dsb + (lq & 1);
+                        union { frag_ab f; bf16x4 h[2]; } akt;
+                        akt.h[0] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                            (lds_bf16x4_p)&ks_lds[cur][
+                                sub_tile_base<D>(g0, dblk) + (lane & 15) * 4]);
+                        akt.h[1] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                            (lds_bf16x4_p)&ks_lds[cur][
+                                sub_tile_base<D>(g0 + 1, dblk) + (lane & 15) * 4]);
+                        accDQ[dsb] = MFMA_32x32x16(akt.f, dsfrag[ksq],
+                                                   accDQ[dsb]);
+                    }
+                    __builtin_amdgcn_s_setprio(0);
+                }
+            }
+        }
+        if (!last) {
+            if (!EARLY_LOAD) stage_load(nb + 1);
+            stage_write(cur ^ 1);
+        }
+        __syncthreads();
+    }
+
+    // epilogue: dq row iq; reg r is d = 8*(r>>2)+(r&3)+4*hi  (dS already
+    // carries the softmax-scale factor)
+    bf16* dqrow = dq + b * wb + h * wh + (int64_t)iq * ws;
+#pragma unroll
+    for (int dsb = 0; dsb < DSUB; ++dsb) {
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+            union { __bf16 h4[4]; uint2 u; } w;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) w.h4[r] = (__bf16)accDQ[dsb][4 * g + r];
+            *reinterpret_cast<uint2*>(dqrow + dsb * 32 + 8 * g + 4 * hi) = w.u;
+        }
+    }
+}
+
+template <int D, int WAVES, bool DO_DK, bool DO_DV>
+__global__ __launch_bounds__(WAVES * WAVE_SIZE)
+void attn_bwd_dkdv_v2_kernel(const bf16* __restrict__ dout,
+                             const bf16* __restrict__ q, const bf16* __restrict__ k,
+                             const bf16* __restrict__ v,
+                             const float* __restrict__ lse,
+                             const float* __restrict__ delta,
+                             const float* __restrict__ slopes, float scale,
+                             bf16* __restrict__ dk, bf16* __restrict__ dv,
+                             int B, int H, int S, int kv_off, int kv_group,
+                             int64_t gb, int64_t gh, int64_t gs,
+                             int64_t qb2, int64_t qh2, int64_t qs2,
+                             int64_t kb2, int64_t kh2, int64_t ks2,
+                             int64_t vb2, int64_t vh2, int64_t vs2,
+                             int64_t wb, int64_t wh, int64_t ws) {
+    constexpr int QT = 64;                   // q rows staged per tile
+    constexpr int NT = WAVES * WAVE_SIZE;
+    constexpr int ROWS = WAVES * 32;         // kv rows per workgroup
+    constexpr int DCH = D / 16;
+    constexpr int DSUB = D / 32;
+    constexpr int NP = (QT / 2) * (D / 8);
+    constexpr int PPL = (NP + NT - 1) / NT;
+
+    __shared__ __attribute__((aligned(16))) bf16 qs_lds[2][QT * D];
+    __shared__ __attribute__((aligned(16))) bf16 dos_lds[2][QT * D];
+    __shared__ float lse_lds[2][QT];
+    __shared__ float dlt_lds[2][QT];
+
+    const int nb = blockIdx.x;
+    const int h = blockIdx.y;   // KV head
+    const int b = blockIdx.z;
+    const int tid = threadIdx.x;
+    const int wave = tid / WAVE_SIZE;
+    const int lane = tid % WAVE_SIZE;
+    const int l31 = lane & 31;
+    const int hi = lane >> 5;
+    const int lq = lane >> 4;
+
+    const int kv0w = nb * ROWS + wave * 32;  // wave's kv rows
+    const int n_glob = kv0w + l31;
+    const float LOG2E = 1.4426950408889634f;
+    const float scale2 = scale * LOG2E;
+
+    frag_ab bK[DCH], bV[DO_DK ? DCH : 1];
+    {
+        const bf16* krow = k + b * kb2 + h * kh2 + (int64_t)n_glob * ks2;
+        const bf16* vrow = v + b * vb2 + h * vh2 + (int64_t)n_glob * vs2;
+#pragma unroll
+        for (int c = 0; c < DCH; ++c) {
+            bK[c] = *reinterpret_cast<const frag_ab*>(krow + c * 16 + 8 * hi);
+            if (DO_DK)
+                bV[c] = *reinterpret_cast<const frag_ab*>(vrow + c * 16 + 8 * hi);
+        }
+    }
+    f32x16 accDK[DO_DK ? DSUB : 1], accDV[DO_DV ? DSUB : 1];
+#pragma unroll
+    for (int dsb = 0; dsb < DSUB; ++dsb) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+            if (DO_DK) accDK[dsb][r] = 0.f;
+            if (DO_DV) accDV[dsb][r] = 0.f;
+        }
+    }
+
+    const bool full_vis = (kv_off <= -S);
+    const int q_first = full_vis ? 0
+        : max(0, (nb * ROWS + kv_off) & ~(QT - 1));
+
+    for (int gq = 0; gq < kv_group; ++gq) {
+        const int hq = h * kv_group + gq;
+        const int64_t bh_off = ((int64_t)b * H + hq) * S;
+        const bf16* qp = q + b * qb2 + hq * qh2;
+        const bf16* dop = dout + b * gb + hq * gh;
+        const float slope2 = slopes[hq] * LOG2E;
+
+        frag_ab qreg[PPL][2], doreg[PPL][2];
+        auto stage_load = [&](int q0) {
+#pragma unroll
+            for (int i = 0; i < PPL; ++i) {
+                const int p = tid + i * NT;
+                if (p < NP) {
+                    const int row = (p / (D / 8)) * 2;
+                    const int col = (p % (D / 8)) * 8;
+                    qreg[i][0] = *reinterpret_cast<const frag_ab*>(
+                        qp + (int64_t)(q0 + row) * qs2 + col);
+                    qreg[i][1] = *reinterpret_cast<const frag_ab*>(
+                        qp + (int64_t)(q0 + row + 1) * qs2 + col);
+                    doreg[i][0] = *reinterpret_cast<const frag_ab*>(
+                        dop + (int64_t)(q0 + row) * gs + col);
+                    doreg[i][1] = *reinterpret_cast<const frag_ab*>(
+                        dop + (int64_t)(q0 + row + 1) * gs + col);
+                }
+            }
+        };
+        auto stage_write = [&](int buf, int q0) {
+#pragma unroll
+            for (int i = 0; i < PPL; ++i) {
+                const int p = tid + i * NT;
+                if (p < NP) {
+                    const int row = (p / (D / 8)) * 2;
+                    const int col = (p % (D / 8)) * 8;
+#pragma unroll
+                    for (int rr = 0; rr < 2; ++rr) {
+                        *reinterpret_cast<frag_ab*>(
+                            &qs_lds[buf][sub_idx<D>(row + rr, col)]) =
+                            qreg[i][rr];
+                        *reinterpret_cast<frag_ab*>(
+                            &dos_lds[buf][sub_idx<D>(row + rr, col)]) =
+                            doreg[i][rr];
+                    }
+                }
+            }
+            for (int i = tid; i < QT; i += NT) {
+                lse_lds[buf][i] = lse[bh_off + q0 + i] * LOG2E;
+                dlt_lds[buf][i] = delta[bh_off + q0 + i];
+            }
+        };
+
+        const int n_qt = (S - q_first) / QT;
+        if (n_qt > 0) {
+            stage_load(q_first);
+            stage_write(0, q_first);
+        }
+        __syncthreads();
+
+        for (int it = 0; it < n_qt; ++it) {
+            const int q0 = q_first + it * QT;
+            const int cur = it & 1;
+            const bool last = (it + 1 == n_qt);
+            if (!last) stage_load(q0 + QT);
+
+            // wave-skip: a q-tile entirely above this wave's diagonal
+            const bool active = full_vis || (q0 + QT - 1 >= kv0w + kv_off);
+            if (active) {
+#pragma unroll
+                for (int qs2c = 0; qs2c < 2; ++qs2c) {
+                    f32x16 st, dp;
+#pragma unroll
+                    for (int r = 0; r < 16; ++r) { st[r] = 0.f; dp[r] = 0.f; }
+                    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                    for (int c = 0; c < DCH; ++c) {
+                        // St = Q·K^T: A = Q rows (row-major read), B = K regs
+                        frag_ab aQ = *reinterpret_cast<const frag_ab*>(
+                            &qs_lds[cur][sub_idx<D>(qs2c * 32 + l31,
+                                                    c * 16 + 8 * hi)]);
+                        st = MFMA_32x32x16(aQ, bK[c], st);
+                        if (DO_DK) {
+                            frag_ab aDO = *reinterpret_cast<const frag_ab*>(
+                                &dos_lds[cur][sub_idx<D>(qs2c * 32 + l31,
+                                                         c * 16 + 8 * hi)]);
+                            dp = MFMA_32x32x16(aDO, bV[c], dp);  // dO·V^T
+                        }
+                    }
+                    __builtin_amdgcn_s_setprio(0);
+                    const int jk = n_glob + kv_off;
+#pragma unroll
+                    for (int r = 0; r < 16; ++r) {
+                        const int q_loc = qs2c * 32 + (r & 3) + 8 * (r >> 2) +
+                                          4 * hi;
+                        const int iq = q0 + q_loc;
+                        const float z2 = st[r] * scale2 +
+                                         slope2 * (float)(jk - iq);
+                        const float p = (jk <= iq)
+                            ? exp2f(z2 - lse_lds[cur][q_loc]) : 0.f;
+                        st[r] = p;                                   // P
+                        if (DO_DK)
+                            dp[r] = p * (dp[r] - dlt_lds[cur][q_loc]) * scale;
+                    }
+                    // pack P and dS into B-fragments over the q axis
+                    frag_ab pfrag[2], dsfrag[2];
+#pragma unroll
+                    for (int hf = 0; hf < 2; ++hf) {
+                        union { __bf16 h2[2]; uint32_t u; } wa, wb2, wc, wd;
+                        if (!DO_DV) {
+                            wa.h2[0] = (__bf16)dp[8 * hf + 0];
+                            wa.h2[1] = (__bf16)dp[8 * hf + 1];
+                            wc.h2[0] = (__bf16)dp[8 * hf + 2];
+                            wc.h2[1] = (__bf16)dp[8 * hf + 3];
+                            wb2.h2[0] = (__bf16)dp[8 * hf + 4];
+                            wb2.h2[1] = (__bf16)dp[8 * hf + 5];
+                            wd.h2[0] = (__bf16)dp[8 * hf + 6];
+                            wd.h2[1] = (__bf16)dp[8 * hf + 7];
+                            auto t1 = __builtin_amdgcn_permlane32_swap(
+                                wa.u, wb2.u, false, false);
+                            auto t2 = __builtin_amdgcn_permlane32_swap(
+                                wc.u, wd.u, false, false);
+                            union { frag_ab f; uint32_t u[4]; } o2;
+                            o2.u[0] = t1[0];
+                            o2.u[1] = t2[0];
+                            o2.u[2] = t1[1];
+                            o2.u[3] = t2[1];
+                            dsfrag[hf] = o2.f;
+                            continue;
+                        }
+                        wa.h2[0] = (__bf16)st[8 * hf + 0];
+                        wa.h2[1] = (__bf16)st[8 * hf + 1];
+                        wc.h2[0] = (__bf16)st[8 * hf + 2];
+                        wc.h2[1] = (__bf16)st[8 * hf + 3];
+                        wb2.h2[0] = (__bf16)st[8 * hf + 4];
+                        wb2.h2[1] = (__bf16)st[8 * hf + 5];
+                        wd.h2[0] = (__bf16)st[8 * hf + 6];
+                        wd.h2[1] = (__bf16)st[8 * hf + 7];
+                        auto s1 = __builtin_amdgcn_permlane32_swap(
+                            wa.u, wb2.u, false, false);
+                        auto s2 = __builtin_amdgcn_permlane32_swap(
+                            wc.u, wd.u, false, false);
+                        union { frag_ab f; uint32_t u[4]; } out;
+                        out.u[0] = s1[0];
+                        out.u[1] = s2[0];
+                        out.u[2] = s1[1];
+                        out.u[3] = s2[1];
+                        pfrag[hf] = out.f;
+                        if (!DO_DK) continue;
+
+                        wa.h2[0] = (__bf16)dp[8 * hf + 0];
+                        wa.h2[1] = (__bf16)dp[8 * hf + 1];
+                        wc.h2[0] = (__bf16)dp[8 * hf + 2];
+                        wc.h2[1] = (__bf16)dp[8 * hf + 3];
+                        wb2.h2[0] = (__bf16)dp[8 * hf + 4];
+                        wb2.h2[1] = (__bf16)dp[8 * hf + 5];
+                        wd.h2[0] = (__bf16)dp[8 * hf + 6];
+                        wd.h2[1] = (__bf16)dp[8 * hf + 7];
+                        s1 = __builtin_amdgcn_permlane32_swap(
+                            wa.u, wb2.u, false, false);
+                        s2 = __builtin_amdgcn_permlane32_swap(
+                            wc.u, wd.u, false, false);
+                        out.u[0] = s1[0];
+                        out.u[1] = s2[0];
+                        out.u[2] = s1[1];
+                        out.u[3] = s2[1];
+                        dsfrag[hf] = out.f;
+                    }
+                    // dV^T += dO^T·P  and  dK^T += Q^T·dS (transpose reads)
+#pragma unroll
+                    for (int dsb = 0; dsb < DSUB; ++dsb) {
+                        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                        for (int ksq = 0; ksq < 2; ++ksq) {
+                            const int g0 = qs2c * 8 + 4 * ksq + 2 * hi;
+                            const int dblk = 2 * dsb + (lq & 1);
+                            if (DO_DV) {
+                                union { frag_ab f; bf16x4 h[2]; } adot;
+                                adot.h[0] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                                    (lds_bf16x4_p)&dos_lds[cur][
+                                        sub_tile_base<D>(g0, dblk) + (lane & 15) * 4]);
+                                adot.h[1] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                                    (lds_bf16x4_p)&dos_lds[cur][
+                                        sub_tile_base<D>(g0 + 1, dblk) + (lane & 15) * 4]);
+                                accDV[DO_DV ? dsb : 0] = MFMA_32x32x16(
+                                    adot.f, pfrag[ksq], accDV[DO_DV ? dsb : 0]);
+                            }
+                            if (DO_DK) {
+                                union { frag_ab f; bf16x4 h[2]; } aqt;
+                                aqt.h[0] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                                    (lds_bf16x4_p)&qs_lds[cur][
+                                        sub_tile_base<D>(g0, dblk) + (lane & 15) * 4]);
+                                aqt.h[1] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                                    (lds_bf16x4_p)&qs_lds[cur][
+                                        sub_tile_base<D>(g0 + 1, dblk) + (lane & 15) * 4]);
+                                accDK[DO_DK ? dsb : 0] = MFMA_32x32x16(
+                                    aqt.f, dsfrag[ksq], accDK[DO_DK ? dsb : 0]);
+                            }
+                        }
+                        __builtin_amdgcn_s_setprio(0);
+                    }
+                }
+            }
+            if (!last) stage_write(cur ^ 1, q0 + QT);
+            __syncthreads();
+        }
+        __syncthreads();  // gq rotation reuses the LDS buffers
+    }
+
+    // epilogue: lane owns kv row n_glob; reg r is d = 8*(r>>2)+(r&3)+4*hi
+    bf16* dkrow = dk + b * wb + h * wh + (int64_t)n_glob * ws;
+    bf16* dvrow = dv + b * wb + h * wh + (int64_t)n_glob * ws;
+#pragma unroll
+    for (int dsb = 0; dsb < DSUB; ++dsb) {
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+            union { __bf16 h4[4]; uint2 u; } wk2, wv2;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                if (DO_DK) wk2.h4[r] = (__bf16)accDK[dsb][4 * g + r];
+                if (DO_DV) wv2.h4[r] = (__bf16)accDV[dsb][4 * g + r];
+            }
+            if (DO_DK)
+                *reinterpret_cast<uint2*>(dkrow + dsb * 32 + 8 * g + 4 * hi) = wk2.u;
+            if (DO_DV)
+                *reinterpret_cast<uint2*>(dvrow + dsb * 32 + 8 * g + 4 * hi) = wv2.u;
+        }
+    }
+}
+
+}  // namespace
+
 void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
                    torch::Tensor k, torch::Tensor v,
                    torch::Tensor o, torch::Tensor lse,
@@ -1232,6 +1781,87 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
 
     const int64_t rows = (int64_t)B * H * S;
     auto stream = at::cuda::getCurrentCUDAStream();
+
+    // v2 path (32x32 MFMA, lane-local stats, tr16 transposed reads)
+    static const int use_bwd_v2 = [] {
+        const char* e = getenv("PG_ATTN_BWD_V2");
+        return e ? atoi(e) : 1;
+    }();
+    if (use_bwd_v2 && S % 128 == 0 && (D == 64 || D == 128)) {
+#define STRV(t) t.stride(0), t.stride(1), t.stride(2)
+#define BWD2_ARGS(W_OUT)                                                      \
+            reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
+            reinterpret_cast<const bf16*>(q.data_ptr()),                      \
+            reinterpret_cast<const bf16*>(k.data_ptr()),                      \
+            reinterpret_cast<const bf16*>(v.data_ptr()),                      \
+            lse.data_ptr<float>(), delta.data_ptr<float>(),                   \
+            slopes.data_ptr<float>(), (float)scale, W_OUT,                    \
+            B, H, S, (int)kv_off, kv_group,                                   \
+            STRV(dout), STRV(q), STRV(k), STRV(v), STRV(dq)
+#define LAUNCH_DELTA(DV)                                                      \
+        hipLaunchKernelGGL((attn_bwd_delta_kernel<DV>),                       \
+            dim3((rows + 3) / 4), dim3(256), 0, stream,                       \
+            reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
+            reinterpret_cast<const bf16*>(o.data_ptr()),                      \
+            delta.data_ptr<float>(), rows, H, S, STRV(dout), STRV(o))
+#define LAUNCH_DQ2(DV, WV)                                                    \
+    do {                                                                      \
+        dim3 grid_q(S / (32 * WV), H, B);                                     \
+        hipLaunchKernelGGL((attn_bwd_dq_v2_kernel<DV, WV>), grid_q,           \
+            dim3(WV * WAVE_SIZE), 0, stream,                                  \
+            BWD2_ARGS(reinterpret_cast<bf16*>(dq.data_ptr())));               \
+    } while (0)
+#define LAUNCH_BWD2(DV, WV, DKV, DVV)                                         \
+    do {                                                                      \
+        dim3 grid_kv(S / (32 * WV), Hkv, B);                                  \
+        hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<DV, WV, DKV, DVV>),       \
+            grid_kv,                                                          \
+            dim3(WV * WAVE_SIZE), 0, stream,                                  \
+            reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
+            reinterpret_cast<const bf16*>(q.data_ptr()),                      \
+            reinterpret_cast<const bf16*>(k.data_ptr()),                      \
+            reinterpret_cast<const bf16*>(v.data_ptr()),                      \
+            lse.data_ptr<float>(), delta.data_ptr<float>(),                   \
+            slopes.data_ptr<float>(), (float)scale,                           \
+            reinterpret_cast<bf16*>(dk.data_ptr()),                           \
+            reinterpret_cast<bf16*>(dv.data_ptr()),                           \
+            B, H, S, (int)kv_off, kv_group,                                   \
+            STRV(dout), STRV(q), STRV(k), STRV(v), STRV(dk));                 \
+    } while (0)
+        // D=64 fits the fused dk+dv kernel in 2 waves/SIMD; D=128 would
+        // spill, so it runs as a dv-only and a dk-only pass (St recomputed,
+        // +25% MFMA but no scratch traffic)
+        if (D == 64) {
+            LAUNCH_DELTA(64);
+            if (S % 256 == 0) {
+                LAUNCH_DQ2(64, 8);
+                LAUNCH_BWD2(64, 8, true, true);
+            } else {
+                LAUNCH_DQ2(64, 4);
+                LAUNCH_BWD2(64, 4, true, true);
+            }
+        } else {
+            LAUNCH_DELTA(128);
+            if (S % 256 == 0) {
+                LAUNCH_DQ2(128, 8);
+                LAUNCH_BWD2(128, 8, false, true);
+            } else {
+                LAUNCH_DQ2(128, 4);
+                LAUNCH_BWD2(128, 4, false, true);
+            }
+            // dk-only spills at 8 waves (338 regs wanted); 4-wave build is
+            // spill-free at 1 wave/SIMD
+            LAUNCH_BWD2(128, 4, true, false);
+        }
+#undef LAUNCH_BWD2
+#undef LAUNCH_DQ2
+#undef LAUNCH_DELTA
+#undef BWD2_ARGS
+#undef STRV
+        HIP_CHECK_LAUNCH();
+        return;
+    }
+
     const int dkdv_waves = (S % 128 == 0) ? 8 : 4;
     dim3 grid_a(S / (16 * dkdv_waves), Hkv, B);
     dim3 grid_b(S / (16 * dkdv_waves), H, B);

@@ -55,6 +55,21 @@ def test_fused_adamw_state_dict_roundtrip():
 
 
 @pytest.mark.gpu
+def test_fused_adamw_gpu_bf16_state_matches_torch():
+    """Default state dtype follows the param dtype (torch semantics): bf16
+    params -> bf16 moments; compare against torch.optim.AdamW directly."""
+    m1, m2 = _models(torch.bfloat16, "cuda")
+    o1 = torch.optim.AdamW(m1.parameters(), lr=1e-2, weight_decay=0.05,
+                           foreach=True)
+    o2 = FusedAdamW(m2.parameters(), lr=1e-2, weight_decay=0.05)
+    _run(m1, o1, "cuda", torch.bfloat16)
+    _run(m2, o2, "cuda", torch.bfloat16)
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1.float(), p2.float(), atol=3e-2), \
+            (p1.float() - p2.float()).abs().max()
+
+
+@pytest.mark.gpu
 def test_fused_adamw_gpu_matches_torch_fp32():
     m1, m2 = _models(torch.float32, "cuda")
     o1 = torch.optim.AdamW(m1.parameters(), lr=1e-2, weight_decay=0.05)
@@ -73,7 +88,8 @@ def test_fused_adamw_gpu_bf16_vs_fp32_state_reference():
     ps1 = [torch.randn(s, device="cuda").bfloat16().requires_grad_(True)
            for s in shapes]
     ps2 = [p.detach().clone().requires_grad_(True) for p in ps1]
-    o2 = FusedAdamW(ps2, lr=1e-2, weight_decay=0.03)
+    o2 = FusedAdamW(ps2, lr=1e-2, weight_decay=0.03,
+                    state_dtype=torch.float32)
     # reference: fp32-state AdamW applied manually to bf16 params
     ms = [torch.zeros(s, device="cuda") for s in shapes]
     vs = [torch.zeros(s, device="cuda") for s in shapes]
