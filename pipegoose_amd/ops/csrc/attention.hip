@@ -1787,7 +1787,11 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
         const char* e = getenv("PG_ATTN_BWD_V2");
         return e ? atoi(e) : 1;
     }();
-    if (use_bwd_v2 && S % 128 == 0 && (D == 64 || D == 128)) {
+    // measured (r2f A/B): v2 wins at D=64 (1264 -> 1071 us on the 560m
+    // shape); at D=128 the spilled dq + 4-wave dk pass lose to v1
+    // (5474 -> 5622 us), so D=128 stays on v1 unless forced with =2
+    const bool bwd_v2_ok = (D == 64) || use_bwd_v2 >= 2;
+    if (use_bwd_v2 && bwd_v2_ok && S % 128 == 0 && (D == 64 || D == 128)) {
 #define STRV(t) t.stride(0), t.stride(1), t.stride(2)
 #define BWD2_ARGS(W_OUT)                                                      \
             reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
