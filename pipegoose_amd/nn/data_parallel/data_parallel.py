@@ -29,6 +29,7 @@ class DataParallel(Parallel):
         super().__init__(module, parallel_context)
         self._bucket_manager = BucketManager(parallel_context)
         self._hooked_params = []
+        self._hook_handles = []
         self._callback_queued = False
         # False during pipeline microbatch accumulation; engine calls
         # sync_now() after the last microbatch.
@@ -55,11 +56,24 @@ class DataParallel(Parallel):
                 self._bucket_manager.add_param(p, mode)
         self.finish_gradient_sync()
 
+    def deparallelize(self) -> nn.Module:
+        """Remove the grad hooks and wrapper attributes (reference declares
+        this but leaves it unimplemented)."""
+        for h in self._hook_handles:
+            h.remove()
+        self._hook_handles.clear()
+        self._hooked_params.clear()
+        for attr in ("finish_gradient_sync", "_dp_wrapper"):
+            if hasattr(self.module, attr):
+                delattr(self.module, attr)
+        return self.module
+
     def _register_grad_hooks(self, module: nn.Module):
         for p in module.parameters():
             if p.requires_grad:
                 self._hooked_params.append(p)
-                p.register_post_accumulate_grad_hook(self._on_grad_ready)
+                self._hook_handles.append(
+                    p.register_post_accumulate_grad_hook(self._on_grad_ready))
 
     def _on_grad_ready(self, param: torch.nn.Parameter):
         if not self.sync_enabled:

@@ -54,6 +54,7 @@ class LinearParallelizer(ModuleParallelizer):
 
     def parallelize(self):
         module, pc = self.module, self.parallel_context
+        orig_cls = type(module)
         if TensorParallelMapping.is_column_parallel(self.module_name):
             module.__class__ = ColumnParallelLinear
             module.weight.data = get_partition(module.weight.data, dim=0, parallel_context=pc)
@@ -61,10 +62,12 @@ class LinearParallelizer(ModuleParallelizer):
                 module.bias.data = get_partition(module.bias.data, dim=0, parallel_context=pc)
             module.gather_output = False
             module.out_features = module.weight.shape[0]
+            module._tp_info = {"kind": "column", "orig_cls": orig_cls}
         else:
             module.__class__ = RowParallelLinear
             module.weight.data = get_partition(module.weight.data, dim=1, parallel_context=pc)
             module.in_features = module.weight.shape[1]
+            module._tp_info = {"kind": "row", "orig_cls": orig_cls}
         module.parallel_context = pc
         return module
 
@@ -89,6 +92,7 @@ class Conv1DParallelizer(ModuleParallelizer):
         w = module.weight.data.t().contiguous()   # -> [out, in]
         b = module.bias.data if getattr(module, "bias", None) is not None \
             else None
+        orig_cls = type(module)
         if TensorParallelMapping.is_fused_column(self.module_name):
             n = TensorParallelMapping.get_fused_count(self.module_name)
             out = w.shape[0]
@@ -103,6 +107,8 @@ class Conv1DParallelizer(ModuleParallelizer):
             module.__class__ = ColumnParallelLinear
             module.gather_output = False
             module.out_features = module.weight.shape[0]
+            module._tp_info = {"kind": "fused_column", "orig_cls": orig_cls,
+                               "conv1d": True, "fused_n": n}
         elif TensorParallelMapping.is_column_parallel(self.module_name):
             module.weight = nn.Parameter(get_partition(w, 0, pc))
             if b is not None:
@@ -110,12 +116,16 @@ class Conv1DParallelizer(ModuleParallelizer):
             module.__class__ = ColumnParallelLinear
             module.gather_output = False
             module.out_features = module.weight.shape[0]
+            module._tp_info = {"kind": "column", "orig_cls": orig_cls,
+                               "conv1d": True}
         else:  # row-parallel: split the input dim, bias kept whole
             module.weight = nn.Parameter(get_partition(w, 1, pc))
             if b is not None:
                 module.bias = nn.Parameter(b.clone())
             module.__class__ = RowParallelLinear
             module.in_features = module.weight.shape[1]
+            module._tp_info = {"kind": "row", "orig_cls": orig_cls,
+                               "conv1d": True}
         module.parallel_context = pc
         return module
 
@@ -134,6 +144,7 @@ class EmbeddingParallelizer(ModuleParallelizer):
             f"vocab size {vocab_size} must be divisible by tp={world}"
         )
         partition = vocab_size // world
+        orig_cls = type(module)
         module.weight.data = get_partition(module.weight.data, dim=0, parallel_context=pc)
         module.__class__ = ParallelEmbedding
         module.num_embeddings = vocab_size
@@ -141,6 +152,7 @@ class EmbeddingParallelizer(ModuleParallelizer):
         module.vocab_start_idx = rank * partition
         module.vocab_end_idx = (rank + 1) * partition
         module.parallel_context = pc
+        module._tp_info = {"kind": "embedding", "orig_cls": orig_cls}
         return module
 
 
@@ -153,10 +165,12 @@ class LayerNormParallelizer(ModuleParallelizer):
 
     def parallelize(self):
         module, pc = self.module, self.parallel_context
+        orig_cls = type(module)
         normalized_shape = tuple(module.normalized_shape)
         module.__class__ = LayerNorm
         module.normalized_shape = normalized_shape
         module.parallel_context = pc
+        module._tp_info = {"kind": "layer_norm", "orig_cls": orig_cls}
         return module
 
 
@@ -170,17 +184,19 @@ class LMHeadParallelizer(ModuleParallelizer):
 
     def parallelize(self):
         module, pc = self.module, self.parallel_context
+        orig_cls = type(module)
         module.__class__ = ColumnParallelLinear
         embed_weight = _get_tied_embedding_weight(self.model)
-        if embed_weight is not None and module.weight is embed_weight:
-            pass  # already sliced via the embedding parallelizer
-        else:
+        tied = embed_weight is not None and module.weight is embed_weight
+        if not tied:
             module.weight.data = get_partition(module.weight.data, dim=0, parallel_context=pc)
         if getattr(module, "bias", None) is not None:
             module.bias.data = get_partition(module.bias.data, dim=0, parallel_context=pc)
         module.gather_output = True  # logits gathered for the (parallel) loss
         module.out_features = module.weight.shape[0]
         module.parallel_context = pc
+        module._tp_info = {"kind": "column", "orig_cls": orig_cls,
+                           "tied_weight": tied}
         return module
 
 

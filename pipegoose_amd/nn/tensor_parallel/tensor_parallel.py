@@ -55,12 +55,17 @@ class TensorParallel(Parallel):
                           for c in mod.children())
             if not has_cpl:
                 continue
+            restored = {}
             for attr in ("num_heads", "num_attention_heads",
                          "num_key_value_heads", "split_size", "all_head_size"):
                 val = getattr(mod, attr, None)
                 if isinstance(val, int) and val % tp == 0 and val >= tp:
+                    restored[attr] = val
                     setattr(mod, attr, val // tp)
+            if restored:
+                mod._tp_head_attrs = restored
             if type(mod).__name__ == "BloomAttention":
+                mod._tp_orig_forward = mod.forward
                 _patch_bloom_attention_alibi(mod, rank, tp)
 
     @staticmethod
@@ -83,6 +88,89 @@ class TensorParallel(Parallel):
             if p.is_parallelizable(module_name, module):
                 return p
         return None
+
+    def deparallelize(self) -> nn.Module:
+        """Reverse the surgery: all-gather each sliced weight over the
+        TENSOR group, restore the original module classes and attention
+        head counts.  The reference declares this per-parallelizer but
+        every body is ``pass`` (parallelizer.py:57-228); here the model
+        round-trips to its full single-rank form (state_dict equal to the
+        pre-parallelize one up to collective ordering)."""
+        import torch.distributed as dist
+
+        module = self.module
+        pc = self.parallel_context
+        tp = pc.get_world_size(ParallelMode.TENSOR)
+        if tp == 1:
+            return module
+        group = pc.get_group(ParallelMode.TENSOR)
+
+        def gather(t: torch.Tensor, dim: int) -> torch.Tensor:
+            parts = [torch.empty_like(t) for _ in range(tp)]
+            dist.all_gather(parts, t.contiguous(), group=group)
+            return torch.cat(parts, dim=dim)
+
+        for mod in module.modules():
+            ha = getattr(mod, "_tp_head_attrs", None)
+            if ha:
+                for attr, val in ha.items():
+                    setattr(mod, attr, val)
+                del mod._tp_head_attrs
+            if hasattr(mod, "_tp_orig_forward"):
+                mod.forward = mod._tp_orig_forward
+                del mod._tp_orig_forward
+
+        for name, m in module.named_modules():
+            info = getattr(m, "_tp_info", None)
+            if info is None:
+                continue
+            kind = info["kind"]
+            conv1d = info.get("conv1d", False)
+            if kind == "column":
+                if not info.get("tied_weight", False):
+                    m.weight.data = gather(m.weight.data, 0)
+                if m.bias is not None:
+                    m.bias.data = gather(m.bias.data, 0)
+            elif kind == "fused_column":
+                n = info["fused_n"]
+                o_nt, in_f = m.weight.shape[0] // n, m.weight.shape[1]
+                parts = [torch.empty_like(m.weight.data) for _ in range(tp)]
+                dist.all_gather(parts, m.weight.data.contiguous(), group=group)
+                m.weight.data = torch.cat(
+                    [torch.cat([p.view(n, o_nt, in_f)[i] for p in parts], 0)
+                     for i in range(n)], 0)
+                if m.bias is not None:
+                    bparts = [torch.empty_like(m.bias.data) for _ in range(tp)]
+                    dist.all_gather(bparts, m.bias.data.contiguous(), group=group)
+                    m.bias.data = torch.cat(
+                        [torch.cat([p.view(n, o_nt)[i] for p in bparts], 0)
+                         for i in range(n)], 0)
+            elif kind == "row":
+                m.weight.data = gather(m.weight.data, 1)
+            elif kind == "embedding":
+                m.weight.data = gather(m.weight.data, 0)
+                for attr in ("partition_size", "vocab_start_idx",
+                             "vocab_end_idx"):
+                    if hasattr(m, attr):
+                        delattr(m, attr)
+            # layer_norm: class swap only
+
+            if conv1d and kind in ("column", "fused_column", "row"):
+                m.weight.data = m.weight.data.t().contiguous()
+            m.__class__ = info["orig_cls"]
+            if kind in ("column", "fused_column") and not conv1d:
+                m.out_features = m.weight.shape[0]
+            elif kind == "row" and not conv1d:
+                m.in_features = m.weight.shape[1]
+            for attr in ("gather_output", "sequence_parallel",
+                         "parallel_context"):
+                if hasattr(m, attr):
+                    try:
+                        delattr(m, attr)
+                    except AttributeError:
+                        pass
+            del m._tp_info
+        return module
 
 
 def _patch_bloom_attention_alibi(attn, tp_rank: int, tp_size: int):
