@@ -57,6 +57,22 @@ class P2P:
         payload = data.detach().contiguous().to(device)
         dist.send(payload, dst=dst, tag=tag)
 
+    def send_many(self, tensors, dst: int, tag: int = 0):
+        """Typed send of a tuple/list of tensors: one count preamble, then
+        the per-tensor typed codec (used by the pipeline engine's first
+        microbatch when a stage boundary carries multiple values)."""
+        device = self._comm_device()
+        n = torch.tensor([len(tensors)], dtype=torch.int64, device=device)
+        dist.send(n, dst=dst, tag=tag)
+        for t in tensors:
+            self.send(t, dst, tag=tag)
+
+    def recv_many(self, src: int, tag: int = 0):
+        device = self._comm_device()
+        n = torch.zeros(1, dtype=torch.int64, device=device)
+        dist.recv(n, src=src, tag=tag)
+        return [self.recv(src, tag=tag) for _ in range(int(n.item()))]
+
     def recv(self, src: int, tag: int = 0) -> torch.Tensor:
         device = self._comm_device()
         header = torch.zeros(_HEADER_NUMEL, dtype=torch.int64, device=device)

@@ -55,8 +55,7 @@ class PipelineEngine:
         self.prev_rank = parallel_context.get_prev_global_rank(ParallelMode.PIPELINE)
         self.next_rank = parallel_context.get_next_global_rank(ParallelMode.PIPELINE)
 
-        self._act_shape = None
-        self._act_dtype = None
+        self._act_meta = None  # [(shape, dtype)] per boundary element
 
     def _actions(self):
         if self.schedule_kind == "gpipe":
@@ -137,9 +136,13 @@ class PipelineEngine:
                 x = input_mbs[mb]
             else:
                 x = self._recv_forward(mb)
-                x.requires_grad_(x.is_floating_point())
+                for el in _as_tuple(x):
+                    el.requires_grad_(el.is_floating_point())
             saved_in[mb] = x
-            out = self.stage(x)
+            # fx-partitioned stages exchange a TUPLE of live values (skip
+            # connections etc. — partitioner.FxUniformPartitioner); plain
+            # stages keep the single-tensor fast path
+            out = self.stage(*x) if isinstance(x, tuple) else self.stage(x)
             saved_out[mb] = out
             # Snapshot MoE router aux/z losses pushed during THIS forward.
             # They share graph nodes with `out` (the gate reads the stage's
@@ -150,7 +153,7 @@ class PipelineEngine:
             # the weights for every stage instead (moe_aux_weight/moe_z_weight).
             self._saved_moe[mb] = self._pop_moe_losses()
             if not self.is_last:
-                pending.append(self._send_forward(out, mb))
+                pending.extend(self._send_forward(out, mb))
             elif not has_loss:
                 outputs.append(out.detach())
         else:  # BACKWARD
@@ -165,15 +168,30 @@ class PipelineEngine:
                     loss.backward()
                 # inference-only: nothing to do
             else:
-                grad = self._recv_backward(mb, out)
+                grads = self._recv_backward(mb, out)
+                # pair grads with the float elements; drop grads for
+                # elements that don't require grad (e.g. detached paths)
+                outs_t = _as_tuple(out)
+                bwd_outs, bwd_grads = [], []
+                gi = 0
+                for o in outs_t:
+                    if o.is_floating_point():
+                        g = grads[gi]
+                        gi += 1
+                        if o.requires_grad:
+                            bwd_outs.append(o)
+                            bwd_grads.append(g)
                 if moe is not None and moe.requires_grad:
-                    torch.autograd.backward([out, moe / m],
-                                            grad_tensors=[grad, None])
-                else:
-                    torch.autograd.backward(out, grad_tensors=grad)
+                    bwd_outs.append(moe / m)
+                    bwd_grads.append(None)
+                torch.autograd.backward(bwd_outs, grad_tensors=bwd_grads)
             x = saved_in[mb]
-            if not self.is_first and x is not None and x.grad is not None:
-                pending.append(self._send_backward(x.grad, mb))
+            if not self.is_first and x is not None:
+                gsend = [el.grad if el.grad is not None else
+                         torch.zeros_like(el)
+                         for el in _as_tuple(x) if el.is_floating_point()]
+                if gsend:
+                    pending.extend(self._send_backward(gsend, mb))
             saved_in[mb] = saved_out[mb] = None  # free activations
             self._saved_moe[mb] = None
 
@@ -195,59 +213,85 @@ class PipelineEngine:
         return total
 
     # ------------------------------------------------------------- transport
+    # A stage boundary is a tuple of tensors (singleton for plain stages).
+    # Microbatch 0 negotiates count/shape/dtype via the typed codec; every
+    # later microbatch moves raw payloads element by element on the
+    # per-direction streams.  Backward carries one grad per FLOAT forward
+    # element (zeros where autograd produced none).
 
     def _send_forward(self, out, mb):
+        outs = _as_tuple(out)
         if mb == 0:
-            self.codec.send(out, self.next_rank)
-            return _NullWork(), None
-        return self.p2p.send_activation(out, self.next_rank, channel=0)
+            self.codec.send_many(outs, self.next_rank)
+            return [(_NullWork(), None)]
+        return [self.p2p.send_activation(o, self.next_rank, channel=0)
+                for o in outs]
 
     def _recv_forward(self, mb):
         if mb == self._fwd_order[0]:
             # first forward: negotiate shapes, then prefetch every remaining
             # forward recv on the recv stream IN SCHEDULE ORDER (= the
             # sender's send order) — P2P latency overlaps compute
-            t = self.codec.recv(self.prev_rank)
-            self._act_shape = tuple(t.shape)
-            self._act_dtype = t.dtype
+            ts = self.codec.recv_many(self.prev_rank)
+            self._act_meta = [(tuple(t.shape), t.dtype) for t in ts]
             for nxt in self._fwd_order[1:]:
-                self._fwd_prefetch[nxt] = self.p2p.recv_activation_async(
-                    self._act_shape, self._act_dtype, self.prev_rank,
-                    channel=0)
-            return t
-        work, buf = self._fwd_prefetch.pop(mb)
-        work.wait()  # on CUDA: a stream dependency, not a host stall
-        if buf.is_cuda:
-            buf.record_stream(torch.cuda.current_stream())
-        buf.requires_grad_(buf.is_floating_point())
-        return buf
-
-    def _send_backward(self, grad, mb):
-        return self.p2p.send_activation(grad, self.prev_rank, channel=1)
-
-    def _recv_backward(self, mb, out):
-        if mb in self._bwd_prefetch:
-            work, buf = self._bwd_prefetch.pop(mb)
-            work.wait()
+                self._fwd_prefetch[nxt] = [
+                    self.p2p.recv_activation_async(sh, dt, self.prev_rank,
+                                                   channel=0)
+                    for sh, dt in self._act_meta]
+            return ts[0] if len(ts) == 1 else tuple(ts)
+        bufs = []
+        for work, buf in self._fwd_prefetch.pop(mb):
+            work.wait()  # on CUDA: a stream dependency, not a host stall
             if buf.is_cuda:
                 buf.record_stream(torch.cuda.current_stream())
-            return buf
+            buf.requires_grad_(buf.is_floating_point())
+            bufs.append(buf)
+        return bufs[0] if len(bufs) == 1 else tuple(bufs)
+
+    def _send_backward(self, grads, mb):
+        return [self.p2p.send_activation(g, self.prev_rank, channel=1)
+                for g in grads]
+
+    def _grad_meta(self, out):
+        return [(tuple(o.shape), o.dtype) for o in _as_tuple(out)
+                if o.is_floating_point()]
+
+    def _recv_backward(self, mb, out):
+        meta = self._grad_meta(out)
+        if mb in self._bwd_prefetch:
+            bufs = []
+            for work, buf in self._bwd_prefetch.pop(mb):
+                work.wait()
+                if buf.is_cuda:
+                    buf.record_stream(torch.cuda.current_stream())
+                bufs.append(buf)
+            return bufs
         # first backward: post the CURRENT recv first (the peer sends it
         # first), then prefetch the rest in schedule order
-        work, buf = self.p2p.recv_activation_async(
-            tuple(out.shape), out.dtype, self.next_rank, channel=1)
+        cur = [self.p2p.recv_activation_async(sh, dt, self.next_rank,
+                                              channel=1) for sh, dt in meta]
         started = False
         for nxt in self._bwd_order:
             if nxt == mb:
                 started = True
                 continue
             if started:
-                self._bwd_prefetch[nxt] = self.p2p.recv_activation_async(
-                    tuple(out.shape), out.dtype, self.next_rank, channel=1)
-        work.wait()
-        if buf.is_cuda:
-            buf.record_stream(torch.cuda.current_stream())
-        return buf
+                self._bwd_prefetch[nxt] = [
+                    self.p2p.recv_activation_async(sh, dt, self.next_rank,
+                                                   channel=1)
+                    for sh, dt in meta]
+        bufs = []
+        for work, buf in cur:
+            work.wait()
+            if buf.is_cuda:
+                buf.record_stream(torch.cuda.current_stream())
+            bufs.append(buf)
+        return bufs
+
+
+def _as_tuple(x):
+    return x if isinstance(x, tuple) else (x,)
 
 
 class _NullWork:
