@@ -65,16 +65,30 @@ class Experts(nn.Module):
         from pipegoose_amd.nn.expert_parallel.grouped import match_grouped_mlp
         self._grouped = match_grouped_mlp(self.experts)
 
-    def forward(self, inputs: torch.Tensor, dispatch_order: torch.Tensor, *args, **kwargs):
-        # inputs: [B, S, H]; dispatch_order: [B*S] global expert index (top-1)
+    def forward(self, inputs: torch.Tensor, dispatch_order: torch.Tensor,
+                weight: torch.Tensor = None, *args, **kwargs):
+        # inputs: [B, S, H]; dispatch_order: [N] (top-1) or [N, k] global
+        # expert ids; weight: optional [N, num_experts] routing weights
+        # (capacity-dropped entries are 0).  With a weight matrix the output
+        # is the gate-weighted combine (Switch semantics — this also gives
+        # the router a gradient through the main path, which the reference's
+        # unweighted scatter never did); without it, top-1 unweighted
+        # (reference parity).
         shape = inputs.shape
         flat = inputs.reshape(-1, shape[-1])
         if self.dispatch == "alltoall" and self.enable_tensor_parallel:
-            return self._forward_alltoall(flat, dispatch_order).reshape(shape)
+            return self._forward_alltoall(flat, dispatch_order,
+                                          weight).reshape(shape)
+        k = 1 if dispatch_order.dim() == 1 else dispatch_order.size(-1)
+        order2d = dispatch_order.reshape(-1, k)
         outputs = torch.zeros_like(flat)
         for local_idx, expert in enumerate(self.experts):
             global_idx = self.expert_offset + local_idx
-            token_mask = dispatch_order.reshape(-1) == global_idx
+            token_mask = (order2d == global_idx).any(dim=-1)
+            if weight is not None:
+                # capacity-dropped tokens (weight 0) are not computed
+                token_mask = token_mask & (
+                    weight.reshape(-1, self.num_experts)[:, global_idx] > 0)
             # A zero-token expert still runs (on a 0-row batch) so every
             # expert param gets a grad every step: DP ranks route different
             # tokens, and a rank that skipped an expert its EXPERT_DATA peer
@@ -85,12 +99,23 @@ class Experts(nn.Module):
             expert_out = expert(selected)
             if isinstance(expert_out, tuple):
                 expert_out = expert_out[0]
-            outputs[token_mask] = expert_out.to(outputs.dtype)
+            expert_out = expert_out.to(outputs.dtype)
+            if weight is not None:
+                w = weight.reshape(-1, self.num_experts)[token_mask,
+                                                         global_idx]
+                expert_out = expert_out * w.unsqueeze(-1).to(outputs.dtype)
+                idx = token_mask.nonzero(as_tuple=True)[0]
+                outputs = outputs.index_put((idx,), expert_out,
+                                            accumulate=True)  # top-2 sums
+            else:
+                outputs[token_mask] = expert_out
         if self.enable_tensor_parallel:
             outputs = _AllReduceCombine.apply(outputs, self.parallel_context)
         return outputs.reshape(shape)
 
-    def _forward_alltoall(self, flat: torch.Tensor, dispatch_order: torch.Tensor):
+    def _forward_alltoall(self, flat: torch.Tensor,
+                          dispatch_order: torch.Tensor,
+                          weight: torch.Tensor = None):
         """EP dispatch over xGMI.  The activation is replicated across the EP
         group, so first each rank takes its 1/ep chunk of the token set (no
         redundant expert FLOPs — the mask path recomputes every token on every
@@ -99,7 +124,13 @@ class Experts(nn.Module):
         output.  Backward mirrors: chunk the output grad, reverse exchanges,
         all-gather input grads — every rank ends with the FULL input gradient
         (the mask path leaves per-rank partials that are only correct after a
-        downstream TP sum)."""
+        downstream TP sum).
+
+        With a routing-weight matrix: capacity-dropped tokens are NOT put on
+        the wire (their output row stays 0 — the residual path carries them,
+        Switch semantics), top-2 tokens ship once per selected expert, and
+        the combine is gate-weighted (giving the router main-path gradient).
+        """
         from pipegoose_amd.nn.expert_parallel.dispatch import AllToAllDispatcher
         from pipegoose_amd.nn.tensor_parallel._functional import (
             _Gather, _Scatter)
@@ -110,9 +141,28 @@ class Experts(nn.Module):
         rank = self.parallel_context.get_local_rank(ParallelMode.TENSOR)
         assert flat.size(0) % ep == 0, \
             f"token count {flat.size(0)} not divisible by ep={ep}"
+        k = 1 if dispatch_order.dim() == 1 else dispatch_order.size(-1)
         chunk = _Scatter.apply(flat, 0, self.parallel_context)
-        route_chunk = dispatch_order.reshape(-1).chunk(ep, dim=0)[rank]
-        recv, local_idx, state = self._dispatcher.dispatch(chunk, route_chunk)
+        n = chunk.size(0)
+        route_chunk = dispatch_order.reshape(-1, k).chunk(ep, dim=0)[rank]
+
+        if weight is None and k == 1:
+            send_tokens, send_routes = chunk, route_chunk.reshape(-1)
+            tok_idx = None
+        else:
+            w_chunk = weight.reshape(-1, self.num_experts).chunk(ep, dim=0)[rank]
+            tok_rep = torch.arange(n, device=chunk.device) \
+                .repeat_interleave(k)
+            ridx = route_chunk.reshape(-1)
+            wvals = w_chunk[tok_rep, ridx]
+            keep = wvals > 0  # capacity-dropped entries never hit the wire
+            tok_idx = tok_rep[keep]
+            send_tokens = chunk[tok_idx]
+            send_routes = ridx[keep]
+            wkeep = wvals[keep]
+
+        recv, local_idx, state = self._dispatcher.dispatch(send_tokens,
+                                                           send_routes)
         counts = torch.bincount(local_idx, minlength=self.num_local_experts).tolist()
         if self._grouped is not None and recv.size(0) > 0:
             from pipegoose_amd.nn.expert_parallel.grouped import (
@@ -131,6 +181,11 @@ class Experts(nn.Module):
                 start += counts[i]
             expert_out = torch.cat(outs, dim=0) if outs else recv
         combined = self._dispatcher.combine(expert_out, state)
+        if tok_idx is not None:
+            combined = combined * wkeep.unsqueeze(-1).to(combined.dtype)
+            outc = torch.zeros_like(chunk)
+            outc = outc.index_put((tok_idx,), combined, accumulate=True)
+            combined = outc
         return _Gather.apply(combined, 0, self.parallel_context)
 
 

@@ -29,7 +29,8 @@ class ExpertLayer(nn.Module):
         expert_context = ExpertContext.get_instance()
         expert_context.push_aux_loss(router_output.aux_loss)
         expert_context.push_z_loss(router_output.z_loss)
-        outputs = self._experts(inputs, router_output.dispatch_order)
+        outputs = self._experts(inputs, router_output.dispatch_order,
+                                router_output.weight)
         # HF Bloom's MLP signature is (hidden, residual) with the residual
         # added inside the block; when this layer replaces such an MLP the
         # residual arrives as the 2nd positional arg — add it here.
