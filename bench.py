@@ -39,6 +39,10 @@ def parse_args():
                    help="pipeline microbatches (1F1B bubble = (pp-1)/(m+pp-1): "
                         "m=8 at pp=2 -> 11%% vs 20%% at m=4; per-rank "
                         "microbatch stays 2048 tokens)")
+    p.add_argument("--cp", type=int, default=1,
+                   help="context-parallel degree: sequences shard along S "
+                        "over the CONTEXT group (ring attention); seq-len "
+                        "is the FULL sequence")
     p.add_argument("--sp", action="store_true",
                    help="Megatron-style sequence parallelism over the TP group "
                         "(BASELINE config 5)")
@@ -75,6 +79,11 @@ def main():
     os.environ.setdefault("MASTER_PORT", "29571")
 
     tp, pp, dp = pick_parallelism(world_size, args.model, args.tp, args.pp, args.dp)
+    cp = max(1, args.cp)
+    if cp > 1:
+        assert world_size % (tp * pp * cp) == 0, "world != tp*pp*cp*dp"
+        # cp takes its degree out of dp in the auto layout
+        dp = world_size // (tp * pp * cp)
 
     from pipegoose_amd import ParallelContext, ParallelMode
     from pipegoose_amd.models.bloom import (
@@ -85,7 +94,8 @@ def main():
     from pipegoose_amd.optim import DistributedOptimizer
 
     ctx = ParallelContext.from_torch(
-        tensor_parallel_size=tp, pipeline_parallel_size=pp, data_parallel_size=dp)
+        tensor_parallel_size=tp, pipeline_parallel_size=pp,
+        data_parallel_size=dp, context_parallel_size=cp)
 
     use_gpu = torch.cuda.is_available()
     device = torch.device(args.device) if args.device else ctx.device
@@ -96,6 +106,9 @@ def main():
 
     if args.sp:
         cfg.sequence_parallel = True
+    if cp > 1:
+        cfg.context_parallel = True
+        assert args.seq_len % cp == 0
     torch.manual_seed(1234)
     model = BloomForCausalLM(cfg, ctx)
     moe_loss_wrap = None
@@ -124,6 +137,9 @@ def main():
     model = model.to(device=device, dtype=dtype)
     if dp > 1:
         model = DataParallel(model, ctx).parallelize()
+    if cp > 1:
+        # params replicate over CP: grads all-reduce over the CONTEXT group
+        model = DataParallel(model, ctx, mode=ParallelMode.CONTEXT).parallelize()
 
     # hipGraph capture: single-rank path only (no RCCL inside the graph);
     # kills per-kernel launch latency on the launch-bound small models.
@@ -152,6 +168,13 @@ def main():
     # synthetic data, fixed per rank (weak scaling: per-GPU work constant)
     g = torch.Generator().manual_seed(4242 + rank)
     input_ids = torch.randint(0, cfg.vocab_size, (B, S), generator=g).to(device)
+    if cp > 1:
+        # every CP rank sees the SAME full sequence, keeps its S/cp shard
+        g2 = torch.Generator().manual_seed(4242 + (rank // cp) * cp)
+        full = torch.randint(0, cfg.vocab_size, (B, S), generator=g2)
+        cpr = ctx.get_local_rank(ParallelMode.CONTEXT)
+        Sl = S // cp
+        input_ids = full[:, cpr * Sl:(cpr + 1) * Sl].to(device)
 
     def one_step(set_to_none: bool = True):
         optim.zero_grad(set_to_none=set_to_none)
@@ -225,7 +248,7 @@ def main():
                 "model": args.model,
                 "global_batch": global_batch,
                 "seq_len": S,
-                "parallelism": f"tp{tp}pp{pp}dp{dp}"
+                "parallelism": f"tp{tp}pp{pp}dp{dp}" + (f"cp{cp}" if cp > 1 else "")
                                + (f"mb{args.microbatches}" if pp > 1 else "")
                                + ("sp" if args.sp else "")
                                + (f"moe{args.moe}" if args.moe else ""),

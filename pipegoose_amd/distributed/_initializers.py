@@ -3,14 +3,19 @@
 Rank layout (identical math to the reference so checkpoints/group semantics
 match; see reference pipegoose/distributed/_initializers/initialize_*.py):
 
-    world is ordered [pipeline][data][tensor], tensor fastest:
-        global_rank = pp_rank * (dp_size * tp_size) + dp_rank * tp_size + tp_rank
+    world is ordered [pipeline][data][context][tensor], tensor fastest:
+        global_rank = pp*(dp_size*cp_size*tp_size) + dp*(cp_size*tp_size)
+                      + cp*tp_size + tp
 
   - TENSOR groups: contiguous blocks of ``tp_size`` ranks.
   - PIPELINE groups: strided, ``range(i, world, world // pp_size)``.
-  - DATA groups: within each pipeline block, ``range(start + tp_off, end, tp_size)``.
+  - CONTEXT groups: within each (pp, dp) block, same tp offset across the
+    cp blocks: ``range(base + tp_off, base + cp*tp, tp)``.
+  - DATA groups: within each pipeline block, same (cp, tp) coordinate across
+    dp: ``range(start + cp_off*tp + tp_off, end, cp*tp)``.
   - EXPERT_DATA groups: same rank layout as DATA groups (expert params are
     sharded over TENSOR, replicated over DATA).
+    With cp == 1 (the default) every layout reduces to the reference's.
 
 On MI355X a node has 8 GPUs fully connected over xGMI (7 point-to-point links
 per GPU), so unlike NVSwitch there is no switch hop to co-locate around; the
@@ -49,12 +54,14 @@ def _make_group(rank: int, ranks: List[int], mode: ParallelMode) -> ProcessGroup
 
 class ProcessGroupInitializer:
     def __init__(self, rank: int, world_size: int, tensor_parallel_size: int,
-                 pipeline_parallel_size: int, data_parallel_size: int):
+                 pipeline_parallel_size: int, data_parallel_size: int,
+                 context_parallel_size: int = 1):
         self.rank = rank
         self.world_size = world_size
         self.tensor_parallel_size = tensor_parallel_size
         self.pipeline_parallel_size = pipeline_parallel_size
         self.data_parallel_size = data_parallel_size
+        self.context_parallel_size = context_parallel_size
 
     def init_dist_group(self) -> ProcessGroupResult:
         raise NotImplementedError
@@ -87,14 +94,35 @@ class DataParallelGroupInitializer(ProcessGroupInitializer):
     def init_dist_group(self) -> ProcessGroupResult:
         tp = self.tensor_parallel_size
         pp = self.pipeline_parallel_size
-        block = self.world_size // pp  # dp_size * tp_size
+        cp = self.context_parallel_size
+        block = self.world_size // pp  # dp * cp * tp
         result = None
         for p in range(pp):
             start = p * block
-            for j in range(tp):
-                ranks = list(range(start + j, start + block, tp))
+            for j in range(cp * tp):
+                ranks = list(range(start + j, start + block, cp * tp))
                 r = _make_group(self.rank, ranks, ParallelMode.DATA)
                 result = r or result
+        return result
+
+
+class ContextParallelGroupInitializer(ProcessGroupInitializer):
+    """Sequence-block groups: rank order along the group == sequence order
+    (ring attention rotates KV around exactly this ring)."""
+
+    def init_dist_group(self) -> ProcessGroupResult:
+        tp = self.tensor_parallel_size
+        cp = self.context_parallel_size
+        pp = self.pipeline_parallel_size
+        block = self.world_size // pp
+        result = None
+        for p in range(pp):
+            for d in range(block // (cp * tp)):
+                base = p * block + d * cp * tp
+                for j in range(tp):
+                    ranks = list(range(base + j, base + cp * tp, tp))
+                    r = _make_group(self.rank, ranks, ParallelMode.CONTEXT)
+                    result = r or result
         return result
 
 

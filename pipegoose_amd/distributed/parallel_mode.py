@@ -17,3 +17,9 @@ class ParallelMode(Enum):
     # sharded over the TENSOR axis, so their gradients reduce over a group with
     # the same layout as the DATA axis restricted to matching expert shards.
     EXPERT_DATA = "expert_data"
+
+    # Context (sequence-block) parallelism: long sequences shard along S over
+    # this group; ring attention rotates KV blocks around it (absent in the
+    # reference — SURVEY.md §5 'Long-context').  Grid order is
+    # [pipeline][data][context][tensor], tensor fastest.
+    CONTEXT = "context"

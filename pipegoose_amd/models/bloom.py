@@ -41,6 +41,11 @@ class BloomConfig:
     # residuals run on [B, S/tp, H] shards; TP boundaries become
     # all-gather / reduce-scatter along S.  Requires S % tp == 0.
     sequence_parallel: bool = False
+    # Context parallelism: sequences shard along S over the CONTEXT group;
+    # attention runs as rotation-based ring attention (nn/ring_attention.py).
+    # Requires S % cp == 0; params are replicated over CP (sync grads with
+    # DataParallel(mode=ParallelMode.CONTEXT)).
+    context_parallel: bool = False
 
     @property
     def head_dim(self):
@@ -103,6 +108,8 @@ class BloomAttention(nn.Module):
         local = slopes.chunk(tp)[tp_rank].clone()
         self.register_buffer("alibi_slopes", local, persistent=False)
         self._bias_cache = {}
+        self.config = config
+        self.parallel_context = parallel_context
 
     def _alibi_bias(self, seq_len: int, device, dtype) -> torch.Tensor:
         key = (seq_len, device, dtype)
@@ -182,6 +189,18 @@ class BloomAttention(nn.Module):
             bias = self._alibi_bias_rect(S, k.size(2), q.device, q.dtype)
             out = TF.scaled_dot_product_attention(
                 q, k, v, attn_mask=bias, scale=self.inv_norm)
+        elif (getattr(self.config, "context_parallel", False)
+              and self.parallel_context.get_world_size(ParallelMode.CONTEXT) > 1
+              and past_kv is None and not use_cache):
+            # context parallelism: hidden is this rank's S-shard; ring
+            # attention rotates KV around the CONTEXT group (ALiBi offsets
+            # ride the blockwise kv_off)
+            from pipegoose_amd.nn.ring_attention import ring_attention_rotate
+            out = ring_attention_rotate(
+                q.contiguous(), k.contiguous(), v.contiguous(),
+                self.alibi_slopes.to(q.device, torch.float32), self.inv_norm,
+                parallel_context=self.parallel_context,
+                parallel_mode=ParallelMode.CONTEXT).to(q.dtype)
         elif past_kv is None and not use_cache and _kernel_supported(q):
             # training fast path: backward writes one d(fused) buffer
             out = alibi_attention_qkv(fused, self.alibi_slopes, self.inv_norm)

@@ -37,6 +37,9 @@ class LlamaConfig:
     rms_norm_eps: float = 1e-6
     initializer_range: float = 0.02
     sequence_parallel: bool = False
+    # sequences shard along S over the CONTEXT group (ring attention);
+    # params replicate over CP — see models/bloom.py BloomConfig
+    context_parallel: bool = False
 
     @property
     def head_dim(self):
@@ -87,6 +90,13 @@ class LlamaAttention(nn.Module):
         self.register_buffer("zero_slopes", torch.zeros(self.num_heads),
                              persistent=False)
         self._mask_cache = {}
+        self.config = config
+        self.parallel_context = parallel_context
+
+    def _cp_size(self) -> int:
+        if not getattr(self.config, "context_parallel", False):
+            return 1
+        return self.parallel_context.get_world_size(ParallelMode.CONTEXT)
 
     def _causal_mask(self, S: int, device, dtype):
         key = (S, device, dtype)
@@ -120,6 +130,10 @@ class LlamaAttention(nn.Module):
                 offset = past_kv.len
             else:
                 offset = past_kv[0].size(2)
+            if self._cp_size() > 1 and past_kv is None:
+                # context parallelism: positions are global, shards local
+                offset += self.parallel_context.get_local_rank(
+                    ParallelMode.CONTEXT) * S
             q = apply_rope(q, self.rope_theta, pos_offset=offset)
             k = apply_rope(k, self.rope_theta, pos_offset=offset)
         present = None
@@ -137,7 +151,18 @@ class LlamaAttention(nn.Module):
                                                  alibi_attention)
         kernel_ok = (_kernel_supported(q) and past_kv is None
                      and not use_cache)
-        if graph_mode:
+        if self._cp_size() > 1 and past_kv is None and not use_cache:
+            from pipegoose_amd.nn.ring_attention import ring_attention_rotate
+            ke = k.repeat_interleave(self.kv_group, dim=1) \
+                if self.kv_group > 1 else k
+            ve = v.repeat_interleave(self.kv_group, dim=1) \
+                if self.kv_group > 1 else v
+            out = ring_attention_rotate(
+                q.contiguous(), ke.contiguous(), ve.contiguous(),
+                torch.zeros(self.num_heads, device=q.device),
+                self.inv_norm, parallel_context=self.parallel_context,
+                parallel_mode=ParallelMode.CONTEXT).to(q.dtype)
+        elif graph_mode:
             # full-length masked attention over the static cache (constant
             # shapes; unfilled slots are zero + masked by -inf)
             ke = k.repeat_interleave(self.kv_group, dim=1) \

@@ -25,8 +25,12 @@ from pipegoose_amd.nn.parallel import Parallel
 
 
 class DataParallel(Parallel):
-    def __init__(self, module: nn.Module, parallel_context: ParallelContext):
+    def __init__(self, module: nn.Module, parallel_context: ParallelContext,
+                 mode: ParallelMode = ParallelMode.DATA):
+        """``mode`` selects the replication group: DATA (default) or CONTEXT
+        (context-parallel ranks replicate parameters the same way)."""
         super().__init__(module, parallel_context)
+        self._mode = mode
         self._bucket_manager = BucketManager(parallel_context)
         self._hooked_params = []
         self._hook_handles = []
@@ -37,7 +41,7 @@ class DataParallel(Parallel):
 
     def parallelize(self) -> nn.Module:
         module = self.module
-        if self.parallel_context.get_world_size(ParallelMode.DATA) > 1:
+        if self.parallel_context.get_world_size(self._mode) > 1:
             self._register_grad_hooks(module)
         self._save_metadata(module, self.parallel_context)
         # expose hooks for manual / pipeline-driven sync
@@ -47,12 +51,12 @@ class DataParallel(Parallel):
 
     def sync_now(self):
         """Bucket + all-reduce every accumulated grad (pipeline tail sync)."""
-        if self.parallel_context.get_world_size(ParallelMode.DATA) == 1:
+        if self.parallel_context.get_world_size(self._mode) == 1:
             return
         for p in self._hooked_params:
             if p.grad is not None:
                 mode = ParallelMode.EXPERT_DATA if getattr(p, "is_expert", False) \
-                    else ParallelMode.DATA
+                    else self._mode
                 self._bucket_manager.add_param(p, mode)
         self.finish_gradient_sync()
 
@@ -79,7 +83,7 @@ class DataParallel(Parallel):
         if not self.sync_enabled:
             return
         mode = ParallelMode.EXPERT_DATA if getattr(param, "is_expert", False) \
-            else ParallelMode.DATA
+            else self._mode
         self._bucket_manager.add_param(param, mode)
         if not self._callback_queued:
             self._callback_queued = True

@@ -52,6 +52,7 @@ def init_parallel_context(
     data_parallel_size: int = None,
     backend: str = "gloo",
     seed: int = 69,
+    context_parallel_size: int = 1,
 ) -> ParallelContext:
     return ParallelContext(
         rank=rank,
@@ -65,6 +66,7 @@ def init_parallel_context(
         tensor_parallel_size=tensor_parallel_size,
         pipeline_parallel_size=pipeline_parallel_size,
         data_parallel_size=data_parallel_size,
+        context_parallel_size=context_parallel_size,
     )
 
 

@@ -90,3 +90,52 @@ def _run_ring_cp4(rank, world_size, port):
 
 def test_ring_attention_cp4_matches_full():
     spawn(_run_ring_cp4, world_size=4)
+
+
+# ------------------------- rotation-based (memory-bounded) CP over CONTEXT
+
+def _run_rotate(rank, world_size, port, with_alibi):
+    ctx = init_parallel_context(rank, world_size, port,
+                                context_parallel_size=world_size)
+    from pipegoose_amd.distributed.parallel_mode import ParallelMode
+    from pipegoose_amd.nn.ring_attention import ring_attention_rotate
+    torch.manual_seed(11)
+    B, H, S, D = 2, 3, 16 * world_size, 8
+    qf = torch.randn(B, H, S, D)
+    kf = torch.randn(B, H, S, D)
+    vf = torch.randn(B, H, S, D)
+    slopes = (torch.rand(H) * 0.3) if with_alibi else torch.zeros(H)
+    scale = D ** -0.5
+
+    # full-sequence fp32 oracle with grads
+    q0 = qf.clone().requires_grad_(True)
+    k0 = kf.clone().requires_grad_(True)
+    v0 = vf.clone().requires_grad_(True)
+    ref, _ = _block_attention_ref(q0, k0, v0, slopes, scale, 0)
+    g = torch.randn_like(ref)
+    ref.backward(g)
+
+    Sl = S // world_size
+    sl = slice(rank * Sl, (rank + 1) * Sl)
+    q = qf[:, :, sl].clone().requires_grad_(True)
+    k = kf[:, :, sl].clone().requires_grad_(True)
+    v = vf[:, :, sl].clone().requires_grad_(True)
+    out = ring_attention_rotate(q, k, v, slopes, scale, parallel_context=ctx,
+                                parallel_mode=ParallelMode.CONTEXT)
+    assert torch.allclose(out, ref[:, :, sl].to(out.dtype), atol=1e-4), \
+        (out - ref[:, :, sl]).abs().max()
+    out.backward(g[:, :, sl].to(out.dtype))
+    for got, want, name in ((q.grad, q0.grad[:, :, sl], "dq"),
+                            (k.grad, k0.grad[:, :, sl], "dk"),
+                            (v.grad, v0.grad[:, :, sl], "dv")):
+        assert torch.allclose(got, want.to(got.dtype), atol=1e-4), \
+            (name, (got - want).abs().max())
+    ctx.destroy()
+
+
+def test_ring_rotate_cp2_fwd_bwd_parity():
+    spawn(_run_rotate, world_size=2, with_alibi=True)
+
+
+def test_ring_rotate_cp4_fwd_bwd_parity():
+    spawn(_run_rotate, world_size=4, with_alibi=False)
