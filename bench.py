@@ -101,6 +101,32 @@ def main():
     device = torch.device(args.device) if args.device else ctx.device
     dtype = torch.bfloat16 if use_gpu else torch.float32
 
+    # Multi-GPU de-risking: the first real N>1 run must fail LOUDLY, never
+    # hang to the harness timeout.  A per-step hang watchdog dumps every
+    # thread's stack and aborts; PG_BENCH_DEADLINE_S (or a default scaled
+    # to the step budget) bounds the WHOLE run the same way.
+    wd = None
+    if world_size > 1:
+        from pipegoose_amd.utils.watchdog import HangWatchdog
+        step_timeout = float(os.environ.get("PG_BENCH_STEP_TIMEOUT_S", 300))
+        wd = HangWatchdog(timeout_s=step_timeout)
+        wd.start()
+        deadline = float(os.environ.get(
+            "PG_BENCH_DEADLINE_S",
+            step_timeout + 12.0 * (args.steps + args.warmup)))
+        import faulthandler
+        import threading
+
+        def _deadline_abort():
+            faulthandler.dump_traceback()
+            print(f"[bench rank {rank}] deadline {deadline}s exceeded — "
+                  "aborting (suspect a desynced collective)", flush=True)
+            os._exit(124)
+
+        t = threading.Timer(deadline, _deadline_abort)
+        t.daemon = True
+        t.start()
+
     cfg = {"bloom-560m": bloom_560m, "bloom-1b7": bloom_1b7,
            "bloom-7b1": bloom_7b1, "bloom-tiny": bloom_tiny}[args.model]()
 
@@ -177,6 +203,8 @@ def main():
         input_ids = full[:, cpr * Sl:(cpr + 1) * Sl].to(device)
 
     def one_step(set_to_none: bool = True):
+        if wd is not None:
+            wd.tick()
         optim.zero_grad(set_to_none=set_to_none)
         if pp > 1:
             # the engine runs forward AND backward internally (1F1B)

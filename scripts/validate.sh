@@ -24,6 +24,14 @@ python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
     --master-addr 127.0.0.1 --master-port 29802 \
     bench.py --gpus 8 --steps 2 --warmup 1 --model bloom-tiny \
     --tp 2 --pp 2 --dp 2 --seq-len 128 --micro-batch 8
+python -m torch.distributed.run --nnodes=1 --nproc-per-node 2 \
+    --master-addr 127.0.0.1 --master-port 29804 \
+    bench.py --gpus 2 --steps 2 --warmup 1 --model bloom-tiny \
+    --cp 2 --seq-len 128 --micro-batch 4
+python -m torch.distributed.run --nnodes=1 --nproc-per-node 2 \
+    --master-addr 127.0.0.1 --master-port 29805 \
+    bench.py --gpus 2 --steps 2 --warmup 1 --model bloom-tiny \
+    --tp 2 --moe 4 --seq-len 128 --micro-batch 4
 
 # 4. GPU tier (on an MI355X box)
 if python -c "import torch; raise SystemExit(0 if torch.cuda.is_available() else 1)"; then
