@@ -1369,7 +1369,7 @@ void attn_bwd_dq_v2_kernel(const bf16* __restrict__ dout,
 
         const bool active = full_vis || (kvrow0 + kv_off <= qr0 + 31);
         if (active) {
-#pragma unroll
+#pragma unroll 1
             for (int ns = 0; ns < 2; ++ns) {
                 frag_ab dsfrag[2];
                 f32x16 st, dp;
@@ -1598,7 +1598,7 @@ void attn_bwd_dkdv_v2_kernel(const bf16* __restrict__ dout,
             // wave-skip: a q-tile entirely above this wave's diagonal
             const bool active = full_vis || (q0 + QT - 1 >= kv0w + kv_off);
             if (active) {
-#pragma unroll
+#pragma unroll 1
                 for (int qs2c = 0; qs2c < 2; ++qs2c) {
                     f32x16 st, dp;
 #pragma unroll
@@ -1845,6 +1845,9 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
                 LAUNCH_BWD2(64, 4, true, true);
             }
         } else {
+            // fused dk+dv spills at D=128 (256+144 regs wanted); run as a
+            // dv-only and a dk-only pass (St recomputed once more, but no
+            // scratch traffic and 2 waves/SIMD)
             LAUNCH_DELTA(128);
             if (S % 256 == 0) {
                 LAUNCH_DQ2(128, 8);
@@ -1853,8 +1856,8 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
                 LAUNCH_DQ2(128, 4);
                 LAUNCH_BWD2(128, 4, false, true);
             }
-            // dk-only spills at 8 waves (338 regs wanted); 4-wave build is
-            // spill-free at 1 wave/SIMD
+            // the dk pass holds bK+bV+dP and spills at 8 waves; the 4-wave
+            // build is spill-free (352 regs, 1 wave/SIMD)
             LAUNCH_BWD2(128, 4, true, false);
         }
 #undef LAUNCH_BWD2
