@@ -1787,11 +1787,10 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
         const char* e = getenv("PG_ATTN_BWD_V2");
         return e ? atoi(e) : 1;
     }();
-    // measured (r2f A/B): v2 wins at D=64 (1264 -> 1071 us on the 560m
-    // shape); at D=128 the spilled dq + 4-wave dk pass lose to v1
-    // (5474 -> 5622 us), so D=128 stays on v1 unless forced with =2
-    const bool bwd_v2_ok = (D == 64) || use_bwd_v2 >= 2;
-    if (use_bwd_v2 && bwd_v2_ok && S % 128 == 0 && (D == 64 || D == 128)) {
+    // measured (r2h A/B): v2 wins at both head dims once the sub-tile loops
+    // stopped unrolling (D=64: 1264 -> 1110 us; D=128: 5448 -> 5153 us on
+    // the bench shapes); PG_ATTN_BWD_V2=0 forces the v1 kernels
+    if (use_bwd_v2 && S % 128 == 0 && (D == 64 || D == 128)) {
 #define STRV(t) t.stride(0), t.stride(1), t.stride(2)
 #define BWD2_ARGS(W_OUT)                                                      \
             reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
