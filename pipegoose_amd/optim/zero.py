@@ -2,16 +2,20 @@
 
 Reference parity: optim/zero/optim.py:14-75 (greedy numel sharding over DP,
 local step on own shard, param re-sync).  MI355X redesign of the sync:
-instead of dp_size sequential broadcasts of flattened shards (reference
-optim.py:62-66), every rank launches async broadcasts of ONE flat buffer per
-owner rank, all in flight together over xGMI, then unpacks.  On RCCL these
-ride separate channels and saturate the 7 P2P links.
+
+- every rank launches async broadcasts of the owner shards all in flight
+  together over xGMI (the reference did dp_size sequential broadcasts,
+  optim.py:62-66);
+- each owner's parameters are RE-POINTED into one persistent flat buffer
+  per (owner, dtype) at setup, so the broadcast sends the live storage
+  directly — no per-step flatten copy and no unflatten copy-back at all
+  (the round-1 version re-flattened ~all params every step: for bloom-7b1
+  that was a full extra parameter copy per step per rank).
 """
 from typing import List
 
 import torch
 import torch.distributed as dist
-from torch._utils import _flatten_dense_tensors, _unflatten_dense_tensors
 
 from pipegoose_amd.distributed.parallel_context import ParallelContext
 from pipegoose_amd.distributed.parallel_mode import ParallelMode
@@ -58,6 +62,39 @@ class DistributedOptimizer(BaseDistributedOptimizer):
         self.optim.param_groups = []
         for group in partitions[rank]:
             self.optim.add_param_group(group)
+        self._flat_shards = None
+        self._build_flat_shards()
+
+    def _build_flat_shards(self):
+        """Re-point every param's storage into one flat buffer per
+        (owner, dtype): the post-step broadcast then ships the live
+        storage, with no flatten/unflatten copies."""
+        self._flat_shards = []
+        for params in self._rank_params:
+            by_dtype = {}
+            for p in params:
+                by_dtype.setdefault((p.dtype, p.device), []).append(p)
+            flats = []
+            for (dt, dev), ps in by_dtype.items():
+                total = sum(p.numel() for p in ps)
+                flat = torch.empty(total, dtype=dt, device=dev)
+                off = 0
+                for p in ps:
+                    n = p.numel()
+                    flat[off:off + n].copy_(p.data.reshape(-1))
+                    p.data = flat[off:off + n].view_as(p.data)
+                    off += n
+                flats.append((flat, ps))
+            self._flat_shards.append(flats)
+
+    def _flat_valid(self) -> bool:
+        for flats in self._flat_shards:
+            for flat, ps in flats:
+                for p in ps:
+                    if (p.data.untyped_storage().data_ptr()
+                            != flat.untyped_storage().data_ptr()):
+                        return False  # model was moved/re-allocated
+        return True
 
     # ------------------------------------------------------------------- api
 
@@ -77,22 +114,18 @@ class DistributedOptimizer(BaseDistributedOptimizer):
         pc = self.parallel_context
         group = pc.get_group(self.parallel_mode)
         ranks = pc.get_ranks_in_group(self.parallel_mode)
-        works, flats = [], []
-        for owner, params in enumerate(self._rank_params):
-            if not params:
-                flats.append(None)
-                continue
-            flat = _flatten_dense_tensors([p.detach() for p in params])
-            works.append(dist.broadcast(flat, src=ranks[owner], group=group, async_op=True))
-            flats.append(flat)
+        if not self._flat_valid():
+            # the model was moved (.to()) after setup: param storages were
+            # re-allocated off the flat buffers — rebuild them once
+            self._build_flat_shards()
+        works = []
+        for owner, flats in enumerate(self._flat_shards):
+            for flat, _ps in flats:
+                works.append(dist.broadcast(flat, src=ranks[owner],
+                                            group=group, async_op=True))
         for w in works:
             w.wait()
-        rank = pc.get_local_rank(self.parallel_mode)
-        for owner, params in enumerate(self._rank_params):
-            if owner == rank or not params:
-                continue  # own shard already up to date
-            for p, synced in zip(params, _unflatten_dense_tensors(flats[owner], params)):
-                p.data.copy_(synced)
+        # nothing to copy back: every param IS a view of a broadcast buffer
 
     @torch.no_grad()
     def clip_grad_norm_(self, max_norm: float) -> torch.Tensor:

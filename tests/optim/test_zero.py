@@ -105,3 +105,41 @@ def _run_shard_reduce_parity(rank, world_size, port):
 def test_zero_shard_grad_reduce_matches_replicate_dp2():
     from pipegoose_amd.testing.utils import spawn
     spawn(_run_shard_reduce_parity, world_size=2)
+
+
+def _run_flat_shard_rebuild(rank, world_size, port):
+    """Params are views into persistent flat shard buffers; a model move
+    (.to()) re-allocates storages and must trigger a transparent rebuild."""
+    import torch
+    from torch import nn
+    from pipegoose_amd.optim import DistributedOptimizer
+    from pipegoose_amd.testing.utils import init_parallel_context
+
+    ctx = init_parallel_context(rank, world_size, port, data_parallel_size=2)
+    torch.manual_seed(4)
+    model = nn.Sequential(nn.Linear(16, 16), nn.Linear(16, 16))
+    optim = DistributedOptimizer(torch.optim.SGD(model.parameters(), lr=0.1),
+                                 ctx)
+    assert optim._flat_valid()
+    # simulate a post-setup move: re-allocate every param storage
+    for p in model.parameters():
+        p.data = p.data.clone()
+    assert not optim._flat_valid()
+
+    x = torch.randn(4, 16)
+    model(x).sum().backward()
+    for p in model.parameters():  # emulate DP grad sync (identical data)
+        torch.distributed.all_reduce(p.grad)
+    optim.step()
+    assert optim._flat_valid()  # rebuilt
+    # replicas identical after the in-place broadcast
+    for p in model.parameters():
+        peers = [torch.empty_like(p.data) for _ in range(world_size)]
+        torch.distributed.all_gather(peers, p.data)
+        assert torch.equal(peers[0], peers[1])
+    ctx.destroy()
+
+
+def test_zero_flat_shard_views_and_rebuild():
+    from pipegoose_amd.testing.utils import spawn
+    spawn(_run_flat_shard_rebuild, world_size=2)
