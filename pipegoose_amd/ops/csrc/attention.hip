@@ -1369,8 +1369,9 @@ void attn_bwd_dq_v2_kernel(const bf16* __restrict__ dout,
 
         const bool active = full_vis || (kvrow0 + kv_off <= qr0 + 31);
         if (active) {
-#pragma unroll 1
-            for (int ns = 0; ns < 2; ++ns) {
+            // D=128 must NOT unroll (doubled live f32x16 pairs spill);
+            // D=64 has register headroom and full unroll measured faster
+            auto ns_body = [&](int ns) {
                 frag_ab dsfrag[2];
                 f32x16 st, dp;
 #pragma unroll
@@ -1440,6 +1441,13 @@ void attn_bwd_dq_v2_kernel(const bf16* __restrict__ dout,
                     }
                     __builtin_amdgcn_s_setprio(0);
                 }
+            };
+            if (D == 128) {
+#pragma unroll 1
+                for (int ns = 0; ns < 2; ++ns) ns_body(ns);
+            } else {
+                ns_body(0);
+                ns_body(1);
             }
         }
         if (!last) {
@@ -1598,8 +1606,7 @@ void attn_bwd_dkdv_v2_kernel(const bf16* __restrict__ dout,
             // wave-skip: a q-tile entirely above this wave's diagonal
             const bool active = full_vis || (q0 + QT - 1 >= kv0w + kv_off);
             if (active) {
-#pragma unroll 1
-                for (int qs2c = 0; qs2c < 2; ++qs2c) {
+                auto qs_body = [&](int qs2c) {
                     f32x16 st, dp;
 #pragma unroll
                     for (int r = 0; r < 16; ++r) { st[r] = 0.f; dp[r] = 0.f; }
@@ -1730,6 +1737,13 @@ void attn_bwd_dkdv_v2_kernel(const bf16* __restrict__ dout,
                         }
                         __builtin_amdgcn_s_setprio(0);
                     }
+                };
+                if (D == 128) {
+#pragma unroll 1
+                    for (int qs2c = 0; qs2c < 2; ++qs2c) qs_body(qs2c);
+                } else {
+                    qs_body(0);
+                    qs_body(1);
                 }
             }
             if (!last) stage_write(cur ^ 1, q0 + QT);
