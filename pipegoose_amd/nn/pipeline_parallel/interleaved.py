@@ -69,9 +69,14 @@ def interleaved_schedule(rank: int, p: int, v: int, m: int
 
 
 class InterleavedPipelineEngine:
-    """1F1B over v local chunks per rank.  Blocking recvs in schedule order
-    (no prefetch — keeps per-pair FIFO matching trivially correct; overlap
-    tuning is a round-2 item)."""
+    """1F1B over v local chunks per rank.
+
+    Recvs are PREFETCHED on the per-direction recv streams (r2): every
+    activation recv of a run has the same (shape, dtype) and per-(src,dst)
+    FIFO order equals the schedule order on both endpoints, so after the
+    first (codec-negotiated) recv the engine posts every remaining forward
+    irecv at once, and the first grad recv posts the remaining grad
+    irecvs — P2P latency overlaps compute exactly as in PipelineEngine."""
 
     def __init__(self, chunks: List[torch.nn.Module],
                  parallel_context: ParallelContext, n_microbatches: int,
@@ -97,6 +102,8 @@ class InterleavedPipelineEngine:
         self._recv_negotiated = False
         self._shape = None
         self._dtype = None
+        self._fwd_q = None   # prefetched (work, buf) deques, schedule order
+        self._bwd_q = None
 
     def _pop_moe_losses(self):
         from pipegoose_amd.nn.expert_parallel import ExpertContext
@@ -130,6 +137,15 @@ class InterleavedPipelineEngine:
         label_mbs = microbatch.split(labels, m) \
             if (self.rank == self.p - 1 and labels is not None) else [None] * m
 
+        from collections import deque
+        sched = interleaved_schedule(self.rank, self.p, v, m)
+        self._n_fwd_recv = sum(1 for kind, c, _ in sched
+                               if kind == "F" and not self._is_first_stage(c))
+        self._n_bwd_recv = sum(1 for kind, c, _ in sched
+                               if kind == "B" and not self._is_last_stage(c))
+        self._fwd_q = deque()
+        self._bwd_q = deque()
+
         saved_in = [[None] * m for _ in range(v)]
         saved_out = [[None] * m for _ in range(v)]
         # MoE router aux/z losses share graph nodes with the chunk output —
@@ -142,7 +158,7 @@ class InterleavedPipelineEngine:
         if dp is not None:
             dp.sync_enabled = False
 
-        for kind, c, mb in interleaved_schedule(self.rank, self.p, v, m):
+        for kind, c, mb in sched:
             if kind == "F":
                 self._forward(c, mb, input_mbs, saved_in, saved_out,
                               label_mbs, losses, outputs, pending)
@@ -171,9 +187,16 @@ class InterleavedPipelineEngine:
             t = self.codec.recv(self.prev_rank)
             self._shape, self._dtype = tuple(t.shape), t.dtype
             self._recv_negotiated = True
+            # prefetch every remaining forward recv of this run
+            for _ in range(self._n_fwd_recv - 1):
+                self._fwd_q.append(self.p2p.recv_activation_async(
+                    self._shape, self._dtype, self.prev_rank, channel=0))
             return t
-        return self.p2p.recv_activation(self._shape, self._dtype,
-                                        self.prev_rank)
+        work, buf = self._fwd_q.popleft()
+        work.wait()
+        if buf.is_cuda:
+            buf.record_stream(torch.cuda.current_stream())
+        return buf
 
     def _send_act(self, out, pending):
         if not self._send_negotiated:
@@ -212,8 +235,17 @@ class InterleavedPipelineEngine:
                     loss = loss + moe / self.m
                 loss.backward()
         else:
-            grad = self.p2p.recv_activation(tuple(out.shape), out.dtype,
-                                            self.next_rank, tag=1)
+            if not self._bwd_q:
+                # first grad recv: post the current + every remaining one on
+                # the grad recv stream (all grads share the chunk-out shape)
+                for _ in range(self._n_bwd_recv):
+                    self._bwd_q.append(self.p2p.recv_activation_async(
+                        tuple(out.shape), out.dtype, self.next_rank,
+                        tag=1, channel=1))
+            work, grad = self._bwd_q.popleft()
+            work.wait()
+            if grad.is_cuda:
+                grad.record_stream(torch.cuda.current_stream())
             if moe is not None and moe.requires_grad:
                 torch.autograd.backward([out, moe / self.m],
                                         grad_tensors=[grad, None])
