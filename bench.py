@@ -106,10 +106,15 @@ def main():
     # thread's stack and aborts; PG_BENCH_DEADLINE_S (or a default scaled
     # to the step budget) bounds the WHOLE run the same way.
     wd = None
+    hb = None
     if world_size > 1:
+        from pipegoose_amd.utils.failure import HeartbeatMonitor
         from pipegoose_amd.utils.watchdog import HangWatchdog
+        hb = HeartbeatMonitor(interval_s=5.0).start()
         step_timeout = float(os.environ.get("PG_BENCH_STEP_TIMEOUT_S", 300))
-        wd = HangWatchdog(timeout_s=step_timeout)
+        # on hang: name the laggard rank(s) before dumping stacks/aborting
+        wd = HangWatchdog(timeout_s=step_timeout,
+                          on_hang=lambda: print(hb.report(), flush=True))
         wd.start()
         deadline = float(os.environ.get(
             "PG_BENCH_DEADLINE_S",
@@ -205,6 +210,7 @@ def main():
     def one_step(set_to_none: bool = True):
         if wd is not None:
             wd.tick()
+            hb.tick()
         optim.zero_grad(set_to_none=set_to_none)
         if pp > 1:
             # the engine runs forward AND backward internally (1F1B)

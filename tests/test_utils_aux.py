@@ -4,6 +4,7 @@ import time
 
 import torch
 
+from pipegoose_amd.testing.utils import init_parallel_context, spawn
 from pipegoose_amd.utils.tracing import mark, trace_range
 from pipegoose_amd.utils.watchdog import HangWatchdog
 
@@ -91,3 +92,27 @@ def test_distributed_logger_rank_prefix(capsys):
     lg.info("hello")
     lg.error("bad")
     # with no context the prefix is empty but logging must not crash
+
+
+def _run_heartbeat(rank, world_size, port):
+    import time
+    from pipegoose_amd.utils.failure import HeartbeatMonitor
+    ctx = init_parallel_context(rank, world_size, port)
+    hb = HeartbeatMonitor(interval_s=0.1).start()
+    # rank 0 "stalls": stops ticking after 2 steps; rank 1 keeps going
+    for step in range(8):
+        if rank != 0 or step < 2:
+            hb.tick()
+        time.sleep(0.08)
+    time.sleep(0.4)  # let a few gathers land
+    rep = hb.report()
+    assert hb.laggards() == [0], (rank, hb._peer_counts)
+    assert "rank(s) [0]" in rep, rep
+    hb.stop()
+    ctx.destroy()
+
+
+def test_heartbeat_rank_attribution():
+    """Failure attribution: the heartbeat side channel names
+    which rank stopped making progress."""
+    spawn(_run_heartbeat, world_size=2)
