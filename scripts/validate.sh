@@ -41,5 +41,7 @@ if python -c "import torch; raise SystemExit(0 if torch.cuda.is_available() else
     python bench.py --model bloom-7b1 --steps 5 --warmup 2
     python tools/attn_bench.py
     python tools/decode_bench.py
+    # race-detection lane (serialized kernels/copies; see the script)
+    bash scripts/race_check.sh
 fi
 echo "ALL VALIDATION PASSED"
