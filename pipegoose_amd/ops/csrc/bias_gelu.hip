@@ -145,7 +145,13 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
     const int64_t N = x.numel() / C;
     constexpr int BLOCK = 256;
     const int CV = C / 8;
-    dim3 grid((CV + BLOCK - 1) / BLOCK, (int)std::min<int64_t>(N, 512));
+    // grid.y trades db atomic count (CV*8*grid.y adds) against per-thread
+    // row work; PG_BGELU_ROWS for A/B (default 512)
+    static const int rows = [] {
+        const char* e = getenv("PG_BGELU_ROWS");
+        return e ? atoi(e) : 512;
+    }();
+    dim3 grid((CV + BLOCK - 1) / BLOCK, (int)std::min<int64_t>(N, rows));
     auto stream = at::cuda::getCurrentCUDAStream();
     if (x.scalar_type() == torch::kBFloat16) {
         hipLaunchKernelGGL((bias_gelu_bwd_fused_kernel<__hip_bfloat16, BLOCK>),
