@@ -1472,7 +1472,11 @@ void attn_bwd_dq_v2_kernel(const bf16* __restrict__ dout,
     }
 }
 
-template <int D, int WAVES, bool DO_DK, bool DO_DV>
+// DKPART: -1 = accumulate dK over all of D; 0/1 = only that d-half (the
+// dk pass at D=128 exceeds the 8-wave register cap with a full-width
+// accumulator — two half-width launches recompute St/dP but run at
+// 2 waves/SIMD instead of the 4-wave build's 1)
+template <int D, int WAVES, bool DO_DK, bool DO_DV, int DKPART = -1>
 __global__ __launch_bounds__(WAVES * WAVE_SIZE)
 void attn_bwd_dkdv_v2_kernel(const bf16* __restrict__ dout,
                              const bf16* __restrict__ q, const bf16* __restrict__ k,
@@ -1515,6 +1519,8 @@ void attn_bwd_dkdv_v2_kernel(const bf16* __restrict__ dout,
     const float LOG2E = 1.4426950408889634f;
     const float scale2 = scale * LOG2E;
 
+    constexpr int DK0 = (DKPART < 0) ? 0 : DKPART * (DSUB / 2);
+    constexpr int NDSB = (DKPART < 0) ? DSUB : DSUB / 2;
     frag_ab bK[DCH], bV[DO_DK ? DCH : 1];
     {
         const bf16* krow = k + b * kb2 + h * kh2 + (int64_t)n_glob * ks2;
@@ -1526,12 +1532,12 @@ void attn_bwd_dkdv_v2_kernel(const bf16* __restrict__ dout,
                 bV[c] = *reinterpret_cast<const frag_ab*>(vrow + c * 16 + 8 * hi);
         }
     }
-    f32x16 accDK[DO_DK ? DSUB : 1], accDV[DO_DV ? DSUB : 1];
+    f32x16 accDK[DO_DK ? NDSB : 1], accDV[DO_DV ? DSUB : 1];
 #pragma unroll
     for (int dsb = 0; dsb < DSUB; ++dsb) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-            if (DO_DK) accDK[dsb][r] = 0.f;
+            if (DO_DK && dsb < NDSB) accDK[dsb][r] = 0.f;
             if (DO_DV) accDV[dsb][r] = 0.f;
         }
     }
@@ -1590,6 +1596,9 @@ void attn_bwd_dkdv_v2_kernel(const bf16* __restrict__ dout,
             }
         };
 
+        // dk-only at D=128 sits at the register cap: staging loads move to
+        // the write point so their registers don't live across the compute
+        constexpr bool EARLY = !(DO_DK && !DO_DV && D == 128);
         const int n_qt = (S - q_first) / QT;
         if (n_qt > 0) {
             stage_load(q_first);
@@ -1601,7 +1610,7 @@ void attn_bwd_dkdv_v2_kernel(const bf16* __restrict__ dout,
             const int q0 = q_first + it * QT;
             const int cur = it & 1;
             const bool last = (it + 1 == n_qt);
-            if (!last) stage_load(q0 + QT);
+            if (EARLY && !last) stage_load(q0 + QT);
 
             // wave-skip: a q-tile entirely above this wave's diagonal
             const bool active = full_vis || (q0 + QT - 1 >= kv0w + kv_off);
@@ -1723,16 +1732,18 @@ void attn_bwd_dkdv_v2_kernel(const bf16* __restrict__ dout,
                                 accDV[DO_DV ? dsb : 0] = MFMA_32x32x16(
                                     adot.f, pfrag[ksq], accDV[DO_DV ? dsb : 0]);
                             }
-                            if (DO_DK) {
+                            if (DO_DK && dsb >= DK0 && dsb < DK0 + NDSB) {
+                                const int dblk_k = 2 * dsb + (lq & 1);
                                 union { frag_ab f; bf16x4 h[2]; } aqt;
                                 aqt.h[0] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
                                     (lds_bf16x4_p)&qs_lds[cur][
-                                        sub_tile_base<D>(g0, dblk) + (lane & 15) * 4]);
+                                        sub_tile_base<D>(g0, dblk_k) + (lane & 15) * 4]);
                                 aqt.h[1] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
                                     (lds_bf16x4_p)&qs_lds[cur][
-                                        sub_tile_base<D>(g0 + 1, dblk) + (lane & 15) * 4]);
-                                accDK[DO_DK ? dsb : 0] = MFMA_32x32x16(
-                                    aqt.f, dsfrag[ksq], accDK[DO_DK ? dsb : 0]);
+                                        sub_tile_base<D>(g0 + 1, dblk_k) + (lane & 15) * 4]);
+                                const int ai = DO_DK ? dsb - DK0 : 0;
+                                accDK[ai] = MFMA_32x32x16(
+                                    aqt.f, dsfrag[ksq], accDK[ai]);
                             }
                         }
                         __builtin_amdgcn_s_setprio(0);
@@ -1746,7 +1757,10 @@ void attn_bwd_dkdv_v2_kernel(const bf16* __restrict__ dout,
                     qs_body(1);
                 }
             }
-            if (!last) stage_write(cur ^ 1, q0 + QT);
+            if (!last) {
+                if (!EARLY) stage_load(q0 + QT);
+                stage_write(cur ^ 1, q0 + QT);
+            }
             __syncthreads();
         }
         __syncthreads();  // gq rotation reuses the LDS buffers
@@ -1762,10 +1776,11 @@ void attn_bwd_dkdv_v2_kernel(const bf16* __restrict__ dout,
             union { __bf16 h4[4]; uint2 u; } wk2, wv2;
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                if (DO_DK) wk2.h4[r] = (__bf16)accDK[dsb][4 * g + r];
+                if (DO_DK && dsb >= DK0 && dsb < DK0 + NDSB)
+                    wk2.h4[r] = (__bf16)accDK[dsb - DK0][4 * g + r];
                 if (DO_DV) wv2.h4[r] = (__bf16)accDV[dsb][4 * g + r];
             }
-            if (DO_DK)
+            if (DO_DK && dsb >= DK0 && dsb < DK0 + NDSB)
                 *reinterpret_cast<uint2*>(dkrow + dsb * 32 + 8 * g + 4 * hi) = wk2.u;
             if (DO_DV)
                 *reinterpret_cast<uint2*>(dvrow + dsb * 32 + 8 * g + 4 * hi) = wv2.u;
@@ -1828,10 +1843,10 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
             dim3(WV * WAVE_SIZE), 0, stream,                                  \
             BWD2_ARGS(reinterpret_cast<bf16*>(dq.data_ptr())));               \
     } while (0)
-#define LAUNCH_BWD2(DV, WV, DKV, DVV)                                         \
+#define LAUNCH_BWD2(DV, WV, DKV, DVV, DKP)                                         \
     do {                                                                      \
         dim3 grid_kv(S / (32 * WV), Hkv, B);                                  \
-        hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<DV, WV, DKV, DVV>),       \
+        hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<DV, WV, DKV, DVV, DKP>), \
             grid_kv,                                                          \
             dim3(WV * WAVE_SIZE), 0, stream,                                  \
             reinterpret_cast<const bf16*>(dout.data_ptr()),                   \
@@ -1852,26 +1867,26 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
             LAUNCH_DELTA(64);
             if (S % 256 == 0) {
                 LAUNCH_DQ2(64, 8);
-                LAUNCH_BWD2(64, 8, true, true);
+                LAUNCH_BWD2(64, 8, true, true, -1);
             } else {
                 LAUNCH_DQ2(64, 4);
-                LAUNCH_BWD2(64, 4, true, true);
+                LAUNCH_BWD2(64, 4, true, true, -1);
             }
         } else {
             // fused dk+dv spills at D=128 (256+144 regs wanted); run as a
-            // dv-only and a dk-only pass (St recomputed once more, but no
-            // scratch traffic and 2 waves/SIMD)
+            // dv-only pass plus two half-width dk passes, all 8-wave
             LAUNCH_DELTA(128);
             if (S % 256 == 0) {
                 LAUNCH_DQ2(128, 8);
-                LAUNCH_BWD2(128, 8, false, true);
+                LAUNCH_BWD2(128, 8, false, true, -1);
+                LAUNCH_BWD2(128, 8, true, false, 0);
+                LAUNCH_BWD2(128, 8, true, false, 1);
             } else {
                 LAUNCH_DQ2(128, 4);
-                LAUNCH_BWD2(128, 4, false, true);
+                LAUNCH_BWD2(128, 4, false, true, -1);
+                LAUNCH_BWD2(128, 4, true, false, 0);
+                LAUNCH_BWD2(128, 4, true, false, 1);
             }
-            // the dk pass holds bK+bV+dP and spills at 8 waves; the 4-wave
-            // build is spill-free (352 regs, 1 wave/SIMD)
-            LAUNCH_BWD2(128, 4, true, false);
         }
 #undef LAUNCH_BWD2
 #undef LAUNCH_DQ2
