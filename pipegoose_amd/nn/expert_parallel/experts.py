@@ -81,6 +81,15 @@ class Experts(nn.Module):
                                           weight).reshape(shape)
         k = 1 if dispatch_order.dim() == 1 else dispatch_order.size(-1)
         order2d = dispatch_order.reshape(-1, k)
+        if self._grouped is not None:
+            # grouped fast path: E separate per-expert GEMMs underfill the
+            # 256-CU chip; sort this rank's routed tokens by local expert
+            # and run the whole bank as one batched MFMA GEMM
+            outputs = self._forward_local_grouped(flat, order2d, weight)
+            if self.enable_tensor_parallel:
+                outputs = _AllReduceCombine.apply(outputs,
+                                                  self.parallel_context)
+            return outputs.reshape(shape)
         outputs = torch.zeros_like(flat)
         for local_idx, expert in enumerate(self.experts):
             global_idx = self.expert_offset + local_idx
@@ -112,6 +121,36 @@ class Experts(nn.Module):
         if self.enable_tensor_parallel:
             outputs = _AllReduceCombine.apply(outputs, self.parallel_context)
         return outputs.reshape(shape)
+
+    def _forward_local_grouped(self, flat: torch.Tensor,
+                               order2d: torch.Tensor,
+                               weight: torch.Tensor = None) -> torch.Tensor:
+        """Mask-dispatch semantics via one grouped GEMM over the tokens
+        routed to THIS rank's experts (all of them when EP is off)."""
+        from pipegoose_amd.nn.expert_parallel.grouped import (
+            grouped_mlp_forward)
+        N, k = order2d.shape
+        dev = flat.device
+        tok = torch.arange(N, device=dev).repeat_interleave(k)
+        ridx = order2d.reshape(-1)
+        keep = (ridx >= self.expert_offset) & \
+               (ridx < self.expert_offset + self.num_local_experts)
+        if weight is not None:
+            wvals = weight.reshape(N, self.num_experts)[tok, ridx]
+            keep = keep & (wvals > 0)
+        tok, ridx = tok[keep], ridx[keep] - self.expert_offset
+        perm = torch.argsort(ridx, stable=True)
+        tok, ridx = tok[perm], ridx[perm]
+        counts = torch.bincount(ridx,
+                                minlength=self.num_local_experts).tolist()
+        w1, b1, w2, b2, act = self._grouped
+        expert_out = grouped_mlp_forward(flat[tok], counts, w1, b1, w2, b2,
+                                         act).to(flat.dtype)
+        if weight is not None:
+            expert_out = expert_out * wvals[keep][perm].unsqueeze(-1) \
+                .to(flat.dtype)
+        outputs = torch.zeros_like(flat)
+        return outputs.index_put((tok,), expert_out, accumulate=True)
 
     def _forward_alltoall(self, flat: torch.Tensor,
                           dispatch_order: torch.Tensor,

@@ -58,7 +58,11 @@ class FusedAdamW(torch.optim.Optimizer):
             self._static_meta = (shape_key, tail, len(st))
             self._ptr_meta = None
         _, tail, n_slabs = self._static_meta
-        ptr_key = tuple(t.data_ptr() for t in grads)
+        # key over grads AND params: grads reallocate every step under
+        # set_to_none=True, while params can move under a ZeRO flat-shard
+        # rebuild with grads untouched (set_to_none=False)
+        ptr_key = tuple(t.data_ptr() for t in grads) + \
+            tuple(p.data_ptr() for p in params)
         if self._ptr_meta is None or self._ptr_meta[0] != ptr_key:
             ptrs = torch.empty(4 * T, dtype=torch.int64)
             for i, (p, g, m, v) in enumerate(zip(params, grads, ms, vs)):
