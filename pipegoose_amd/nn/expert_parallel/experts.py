@@ -149,8 +149,16 @@ class Experts(nn.Module):
         if weight is not None:
             expert_out = expert_out * wvals[keep][perm].unsqueeze(-1) \
                 .to(flat.dtype)
-        outputs = torch.zeros_like(flat)
-        return outputs.index_put((tok,), expert_out, accumulate=True)
+        if k == 1:
+            # top-1: each token appears at most once — plain scatter
+            outputs = torch.zeros_like(flat)
+            return outputs.index_put((tok,), expert_out)
+        # top-k: a token can receive k contributions; accumulate in fp32
+        # (bf16 scattered atomics are neither precise nor reliable here)
+        out32 = torch.zeros(flat.shape, device=flat.device,
+                            dtype=torch.float32)
+        out32 = out32.index_put((tok,), expert_out.float(), accumulate=True)
+        return out32.to(flat.dtype)
 
     def _forward_alltoall(self, flat: torch.Tensor,
                           dispatch_order: torch.Tensor,
@@ -222,9 +230,15 @@ class Experts(nn.Module):
         combined = self._dispatcher.combine(expert_out, state)
         if tok_idx is not None:
             combined = combined * wkeep.unsqueeze(-1).to(combined.dtype)
-            outc = torch.zeros_like(chunk)
-            outc = outc.index_put((tok_idx,), combined, accumulate=True)
-            combined = outc
+            if k == 1:  # no per-token overlap: plain scatter
+                outc = torch.zeros_like(chunk)
+                combined = outc.index_put((tok_idx,), combined)
+            else:       # top-k sums accumulate in fp32 (see grouped path)
+                out32 = torch.zeros(chunk.shape, device=chunk.device,
+                                    dtype=torch.float32)
+                out32 = out32.index_put((tok_idx,), combined.float(),
+                                        accumulate=True)
+                combined = out32.to(chunk.dtype)
         return _Gather.apply(combined, 0, self.parallel_context)
 
 
