@@ -61,13 +61,17 @@ def grouped_mlp_forward(tokens: torch.Tensor, counts: List[int],
             padded[i, :c] = tokens[start:start + c]
         start += c
 
-    W1 = torch.stack(w1).to(dt)          # [E, I, H]
-    W2 = torch.stack(w2).to(dt)          # [E, H_out, I]
-    h = torch.bmm(padded, W1.transpose(1, 2))
+    # the batch-strided transposed-B bmm MEM-FAULTS in the ROCm GEMM
+    # backend at E=8, m~2k, k=2048, n=8192 (reproduced + bisected,
+    # tools/moe_repro.py r2p: transpose view faults, contiguous passes) —
+    # materialize [E, K, N] operands explicitly
+    W1 = torch.stack(w1).to(dt).transpose(1, 2).contiguous()  # [E, H, I]
+    W2 = torch.stack(w2).to(dt).transpose(1, 2).contiguous()  # [E, I, H_out]
+    h = torch.bmm(padded, W1)
     if b1[0] is not None:
         h = h + torch.stack(b1).to(dt).unsqueeze(1)
     h = act(h)
-    out = torch.bmm(h, W2.transpose(1, 2))
+    out = torch.bmm(h, W2)
     if b2[0] is not None:
         out = out + torch.stack(b2).to(dt).unsqueeze(1)
 
