@@ -44,6 +44,30 @@ def match_grouped_mlp(experts: nn.ModuleList):
     return w1, b1, w2, b2, act
 
 
+class _GroupedLinear(torch.autograd.Function):
+    """Batched y = x @ W^T with CONTIGUOUS-operand bmms on both passes.
+
+    torch.bmm with a transposed (batch-strided) operand mem-faults in the
+    ROCm GEMM backend at the 1b7-MoE shapes (bisected: tools/moe_repro.py)
+    — and autograd's built-in bmm backward issues exactly such transposed
+    views, so the workaround must own the backward too: dx = dy·W and
+    dW = dy^T·x with explicit contiguous copies.
+    """
+
+    @staticmethod
+    def forward(ctx, x, W):              # x [E, M, K], W [E, N, K]
+        ctx.save_for_backward(x, W)
+        return torch.bmm(x.contiguous(), W.transpose(1, 2).contiguous())
+
+    @staticmethod
+    def backward(ctx, dy):               # dy [E, M, N]
+        x, W = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = torch.bmm(dy, W.contiguous())                    # [E, M, K]
+        dW = torch.bmm(dy.transpose(1, 2).contiguous(), x.contiguous())
+        return dx, dW
+
+
 def grouped_mlp_forward(tokens: torch.Tensor, counts: List[int],
                         w1: List[torch.Tensor], b1, w2: List[torch.Tensor],
                         b2, act: nn.Module) -> torch.Tensor:
@@ -61,17 +85,13 @@ def grouped_mlp_forward(tokens: torch.Tensor, counts: List[int],
             padded[i, :c] = tokens[start:start + c]
         start += c
 
-    # the batch-strided transposed-B bmm MEM-FAULTS in the ROCm GEMM
-    # backend at E=8, m~2k, k=2048, n=8192 (reproduced + bisected,
-    # tools/moe_repro.py r2p: transpose view faults, contiguous passes) —
-    # materialize [E, K, N] operands explicitly
-    W1 = torch.stack(w1).to(dt).transpose(1, 2).contiguous()  # [E, H, I]
-    W2 = torch.stack(w2).to(dt).transpose(1, 2).contiguous()  # [E, I, H_out]
-    h = torch.bmm(padded, W1)
+    W1 = torch.stack(w1).to(dt)          # [E, I, H]
+    W2 = torch.stack(w2).to(dt)          # [E, H_out, I]
+    h = _GroupedLinear.apply(padded, W1)
     if b1[0] is not None:
         h = h + torch.stack(b1).to(dt).unsqueeze(1)
     h = act(h)
-    out = torch.bmm(h, W2)
+    out = _GroupedLinear.apply(h, W2)
     if b2[0] is not None:
         out = out + torch.stack(b2).to(dt).unsqueeze(1)
 
