@@ -81,10 +81,14 @@ class Experts(nn.Module):
                                           weight).reshape(shape)
         k = 1 if dispatch_order.dim() == 1 else dispatch_order.size(-1)
         order2d = dispatch_order.reshape(-1, k)
-        if self._grouped is not None:
-            # grouped fast path: E separate per-expert GEMMs underfill the
-            # 256-CU chip; sort this rank's routed tokens by local expert
-            # and run the whole bank as one batched MFMA GEMM
+        import os
+        if self._grouped is not None and os.environ.get("PG_MOE_GROUPED") == "1":
+            # measured SLOWER than the per-expert loop on the 1b7-MoE bench
+            # (16.5k vs 41.3k tok/s): the ROCm transposed-bmm fault forces
+            # contiguous weight copies every call (grouped.py), which
+            # dominate.  Kept opt-in for re-evaluation when the backend bug
+            # is fixed; the a2a path still uses the grouped bank (tokens/ep
+            # per rank makes the copies proportionally cheaper there).
             outputs = self._forward_local_grouped(flat, order2d, weight)
             if self.enable_tensor_parallel:
                 outputs = _AllReduceCombine.apply(outputs,
