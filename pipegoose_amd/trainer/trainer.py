@@ -130,7 +130,18 @@ class Trainer:
         self.callbacks.fire("on_fit_start", self)
         if self.hang_timeout_s is not None:
             from pipegoose_amd.utils.watchdog import HangWatchdog
-            self._watchdog = HangWatchdog(self.hang_timeout_s).start()
+            import torch.distributed as dist
+            self._heartbeat = None
+            if dist.is_initialized() and dist.get_world_size() > 1:
+                # name the stalled rank before aborting (utils/failure.py)
+                from pipegoose_amd.utils.failure import HeartbeatMonitor
+                self._heartbeat = HeartbeatMonitor().start()
+                hb = self._heartbeat
+                self._watchdog = HangWatchdog(
+                    self.hang_timeout_s,
+                    on_hang=lambda: print(hb.report(), flush=True)).start()
+            else:
+                self._watchdog = HangWatchdog(self.hang_timeout_s).start()
         try:
             for epoch in range(epochs):
                 self.state.epoch = epoch
@@ -139,6 +150,8 @@ class Trainer:
                     self.train(batch)
                     if self._watchdog is not None:
                         self._watchdog.tick()
+                        if self._heartbeat is not None:
+                            self._heartbeat.tick()
                     if max_steps is not None and self.state.global_step >= max_steps:
                         break
                 self.callbacks.fire("on_epoch_end", self)
@@ -148,6 +161,9 @@ class Trainer:
             if self._watchdog is not None:
                 self._watchdog.stop()
                 self._watchdog = None
+            if getattr(self, "_heartbeat", None) is not None:
+                self._heartbeat.stop()
+                self._heartbeat = None
         self.state.stage = TrainerStage.FINISHED
         self.callbacks.fire("on_fit_end", self)
         return self.state

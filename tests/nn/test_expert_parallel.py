@@ -347,3 +347,44 @@ def _run_top2_single_rank_oracle(rank, world_size, port):
 
 def test_top2_weighted_combine_oracle():
     spawn(_run_top2_single_rank_oracle, world_size=1)
+
+
+def _run_local_grouped_parity(rank, world_size, port):
+    """PG_MOE_GROUPED=1 routes the mask path through the batched expert
+    bank; outputs/grads must match the per-expert loop exactly."""
+    import os
+    os.environ["PG_MOE_GROUPED"] = "1"
+    try:
+        ctx = init_parallel_context(rank, world_size, port)
+        torch.manual_seed(8)
+        grouped = Experts(NUM_EXPERTS, _mlp(), enable_tensor_parallel=False,
+                          parallel_context=ctx)
+        torch.manual_seed(8)
+        loop = Experts(NUM_EXPERTS, _mlp(), enable_tensor_parallel=False,
+                       parallel_context=ctx)
+        loop._grouped = None  # force the per-expert loop
+        N = 24
+        x1 = torch.randn(2, N // 2, H, requires_grad=True)
+        x2 = x1.detach().clone().requires_grad_(True)
+        route = torch.stack([torch.randperm(NUM_EXPERTS)[:2]
+                             for _ in range(N)])
+        weight = torch.zeros(N, NUM_EXPERTS)
+        weight.scatter_(1, route, torch.rand(N, 2) + 0.1)
+        o1 = grouped(x1, route, weight)
+        o2 = loop(x2, route, weight)
+        assert torch.allclose(o1, o2, atol=1e-5), (o1 - o2).abs().max()
+        g = torch.randn_like(o1)
+        o1.backward(g)
+        o2.backward(g)
+        assert torch.allclose(x1.grad, x2.grad, atol=1e-5)
+        for p1, p2 in zip(grouped.parameters(), loop.parameters()):
+            if p2.grad is None:
+                continue
+            assert torch.allclose(p1.grad, p2.grad, atol=1e-5)
+        ctx.destroy()
+    finally:
+        os.environ.pop("PG_MOE_GROUPED", None)
+
+
+def test_local_grouped_matches_loop():
+    spawn(_run_local_grouped_parity, world_size=1)

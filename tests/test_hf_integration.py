@@ -375,3 +375,42 @@ def test_hf_llama_weights_into_native_tp1():
 
 def test_hf_llama_weights_into_native_tp2():
     spawn(_run_hf_llama_weights_into_native, world_size=2)
+
+
+def _run_tp2_hf_qwen2(rank, world_size, port):
+    """Qwen2 also uses the llama module names but adds BIASES on q/k/v —
+    TP2 logits parity proves the column-parallel bias slicing covers it."""
+    from transformers import Qwen2Config, Qwen2ForCausalLM
+
+    ctx = init_parallel_context(rank, world_size, port, tensor_parallel_size=2)
+
+    def make():
+        cfg = Qwen2Config(vocab_size=256, hidden_size=64,
+                          intermediate_size=128, num_hidden_layers=2,
+                          num_attention_heads=4, num_key_value_heads=2,
+                          max_position_embeddings=64)
+        torch.manual_seed(101)
+        return Qwen2ForCausalLM(cfg).eval()
+
+    model, ref = make(), make()
+    torch.manual_seed(102)
+    ids = torch.randint(0, 256, (2, 12))
+    with torch.no_grad():
+        ref_logits = ref(ids).logits
+    model = TensorParallel(model, ctx).parallelize()
+    with torch.no_grad():
+        out = model(ids).logits
+    if out.size(-1) == ref_logits.size(-1) // 2:
+        import torch.distributed as dist
+        from pipegoose_amd.distributed.parallel_mode import ParallelMode
+        shards = [torch.empty_like(out) for _ in range(2)]
+        dist.all_gather(shards, out.contiguous(),
+                        group=ctx.get_group(ParallelMode.TENSOR))
+        out = torch.cat(shards, dim=-1)
+    assert torch.allclose(out, ref_logits, atol=1e-4), \
+        (out - ref_logits).abs().max()
+    ctx.destroy()
+
+
+def test_tensor_parallel_hf_qwen2_tp2():
+    spawn(_run_tp2_hf_qwen2, world_size=2)
