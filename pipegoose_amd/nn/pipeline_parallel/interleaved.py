@@ -145,6 +145,11 @@ class InterleavedPipelineEngine:
                                if kind == "B" and not self._is_last_stage(c))
         self._fwd_q = deque()
         self._bwd_q = deque()
+        if self._recv_negotiated:
+            # later runs: shapes known — post every forward recv upfront
+            for _ in range(self._n_fwd_recv):
+                self._fwd_q.append(self.p2p.recv_activation_async(
+                    self._shape, self._dtype, self.prev_rank, channel=0))
 
         saved_in = [[None] * m for _ in range(v)]
         saved_out = [[None] * m for _ in range(v)]
