@@ -125,7 +125,12 @@ class PipelineEngine:
                 return torch.stack(losses).sum()
             return None
         if self.is_last:
-            return torch.cat(outputs, dim=0) if outputs else None
+            if not outputs:
+                return None
+            if isinstance(outputs[0], tuple):  # fx stage with tuple output
+                return tuple(torch.cat(parts, dim=0)
+                             for parts in zip(*outputs))
+            return torch.cat(outputs, dim=0)
         return None
 
     def _run_task(self, task, mb, input_mbs, label_mbs, saved_in, saved_out,
@@ -155,7 +160,8 @@ class PipelineEngine:
             if not self.is_last:
                 pending.extend(self._send_forward(out, mb))
             elif not has_loss:
-                outputs.append(out.detach())
+                outputs.append(tuple(o.detach() for o in out)
+                               if isinstance(out, tuple) else out.detach())
         else:  # BACKWARD
             out = saved_out[mb]
             moe = self._saved_moe[mb]

@@ -35,14 +35,18 @@ class FusedAdamW(torch.optim.Optimizer):
                         weight_decay=weight_decay)
         super().__init__(params, defaults)
         self.state_dtype = state_dtype  # None = follow param dtype (torch)
-        self._static_meta = None  # (shape_key, tail_dev, n_slabs)
-        self._ptr_meta = None     # (ptr_key, full_meta_dev)
+        # per-param-group caches (a single slot would thrash when several
+        # groups alternate within one step)
+        self._static_meta = {}  # group idx -> (shape_key, tail, n_slabs)
+        self._ptr_meta = {}     # group idx -> (ptr_key, meta_dev)
 
-    def _meta(self, params: List[torch.Tensor], grads: List[torch.Tensor],
-              ms: List[torch.Tensor], vs: List[torch.Tensor], device):
+    def _meta(self, gi: int, params: List[torch.Tensor],
+              grads: List[torch.Tensor], ms: List[torch.Tensor],
+              vs: List[torch.Tensor], device):
         T = len(params)
         shape_key = (T, tuple(p.numel() for p in params))
-        if self._static_meta is None or self._static_meta[0] != shape_key:
+        if (gi not in self._static_meta
+                or self._static_meta[gi][0] != shape_key):
             numels = torch.tensor([p.numel() for p in params],
                                   dtype=torch.int64)
             st, si = [], []
@@ -55,15 +59,15 @@ class FusedAdamW(torch.optim.Optimizer):
                 torch.tensor(st, dtype=torch.int32).view(torch.uint8),
                 torch.tensor(si, dtype=torch.int32).view(torch.uint8),
             ])
-            self._static_meta = (shape_key, tail, len(st))
-            self._ptr_meta = None
-        _, tail, n_slabs = self._static_meta
+            self._static_meta[gi] = (shape_key, tail, len(st))
+            self._ptr_meta.pop(gi, None)
+        _, tail, n_slabs = self._static_meta[gi]
         # key over grads AND params: grads reallocate every step under
         # set_to_none=True, while params can move under a ZeRO flat-shard
         # rebuild with grads untouched (set_to_none=False)
         ptr_key = tuple(t.data_ptr() for t in grads) + \
             tuple(p.data_ptr() for p in params)
-        if self._ptr_meta is None or self._ptr_meta[0] != ptr_key:
+        if gi not in self._ptr_meta or self._ptr_meta[gi][0] != ptr_key:
             ptrs = torch.empty(4 * T, dtype=torch.int64)
             for i, (p, g, m, v) in enumerate(zip(params, grads, ms, vs)):
                 ptrs[i] = p.data_ptr()
@@ -72,8 +76,8 @@ class FusedAdamW(torch.optim.Optimizer):
                 ptrs[3 * T + i] = v.data_ptr()
             meta = torch.cat([ptrs.view(torch.uint8), tail]).to(
                 device, non_blocking=True)
-            self._ptr_meta = (ptr_key, meta)
-        return self._ptr_meta[1], n_slabs
+            self._ptr_meta[gi] = (ptr_key, meta)
+        return self._ptr_meta[gi][1], n_slabs
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -82,7 +86,7 @@ class FusedAdamW(torch.optim.Optimizer):
             with torch.enable_grad():
                 loss = closure()
         ext = get_extension()
-        for group in self.param_groups:
+        for gi, group in enumerate(self.param_groups):
             params, grads, ms, vs = [], [], [], []
             step_t = None
             for p in group["params"]:
@@ -121,7 +125,7 @@ class FusedAdamW(torch.optim.Optimizer):
                         for p, g, m in zip(params, grads, ms))
             )
             if use_kernel:
-                meta, n_slabs = self._meta(params, grads, ms, vs,
+                meta, n_slabs = self._meta(gi, params, grads, ms, vs,
                                            params[0].device)
                 ext.adamw_fused_step(params, grads, ms, vs, meta, n_slabs,
                                      lr, beta1, beta2, eps, wd, step_t)
