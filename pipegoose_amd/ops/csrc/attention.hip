@@ -724,6 +724,12 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
         const char* e = getenv("PG_ATTN_V2");
         return e ? atoi(e) : 1;
     }();
+    // PG_ATTN_W4=1: 4-wave workgroups (two independent WGs per CU instead
+    // of one barrier-locked 8-wave WG — A/B for phase overlap)
+    static const int force_w4 = [] {
+        const char* e = getenv("PG_ATTN_W4");
+        return e ? atoi(e) : 0;
+    }();
     if (use_v2 && S % 128 == 0) {
 #define LAUNCH_FWD2(DV, WV)                                                   \
     do {                                                                      \
@@ -742,9 +748,11 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
             o.stride(0), o.stride(1), o.stride(2));                           \
     } while (0)
         if (D == 64) {
-            if (S % 256 == 0) LAUNCH_FWD2(64, 8); else LAUNCH_FWD2(64, 4);
+            if (S % 256 == 0 && !force_w4) LAUNCH_FWD2(64, 8);
+            else LAUNCH_FWD2(64, 4);
         } else {
-            if (S % 256 == 0) LAUNCH_FWD2(128, 8); else LAUNCH_FWD2(128, 4);
+            if (S % 256 == 0 && !force_w4) LAUNCH_FWD2(128, 8);
+            else LAUNCH_FWD2(128, 4);
         }
 #undef LAUNCH_FWD2
         HIP_CHECK_LAUNCH();
@@ -1863,9 +1871,13 @@ void attn_bwd_into(torch::Tensor dout, torch::Tensor q,
         // D=64 fits the fused dk+dv kernel in 2 waves/SIMD; D=128 would
         // spill, so it runs as a dv-only and a dk-only pass (St recomputed,
         // +25% MFMA but no scratch traffic)
+        static const int force_w4b = [] {
+            const char* e = getenv("PG_ATTN_W4");
+            return e ? atoi(e) : 0;
+        }();
         if (D == 64) {
             LAUNCH_DELTA(64);
-            if (S % 256 == 0) {
+            if (S % 256 == 0 && !force_w4b) {
                 LAUNCH_DQ2(64, 8);
                 LAUNCH_BWD2(64, 8, true, true, -1);
             } else {
