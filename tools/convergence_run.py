@@ -10,6 +10,7 @@ distribution, NOT synthetic random tokens.
 
 Usage:
   python tools/convergence_run.py --steps 300            # 1-GPU 560M curve
+  python tools/convergence_run.py --steps 200 --moe 8    # MoE (ref run_ep.py)
   python tools/convergence_run.py --model bloom-tiny --steps 40 --device cpu
 
 Writes the loss curve as JSON lines to --out.
@@ -51,6 +52,9 @@ def main():
     ap.add_argument("--lr", type=float, default=3e-4)
     ap.add_argument("--device", default=None)
     ap.add_argument("--out", default="gpurun_out/convergence_curve.jsonl")
+    ap.add_argument("--moe", type=int, default=0,
+                    help="replace MLPs with E-expert Switch MoE "
+                         "(mirrors the reference's run_ep.py artifact)")
     ap.add_argument("--seed", type=int, default=1234)
     args = ap.parse_args()
 
@@ -66,7 +70,24 @@ def main():
 
     cfg = bloom_560m() if args.model == "bloom-560m" else bloom_tiny()
     torch.manual_seed(args.seed)
-    model = BloomForCausalLM(cfg, ctx).to(device=device, dtype=dtype)
+    model = BloomForCausalLM(cfg, ctx)
+    loss_wrap = None
+    if args.moe > 0:
+        from torch import nn
+        from pipegoose_amd.nn import ExpertParallel
+        from pipegoose_amd.nn.expert_parallel import (ExpertLoss,
+                                                      SwitchNoisePolicy,
+                                                      Top1Router)
+        h = cfg.hidden_size
+        proto = nn.Sequential(nn.Linear(h, 4 * h), nn.GELU(),
+                              nn.Linear(4 * h, h))
+        model = ExpertParallel(
+            model, args.moe, expert=proto,
+            router=Top1Router(SwitchNoisePolicy(), args.moe, h),
+            enable_tensor_parallel=False,
+            parallel_context=ctx).parallelize()
+        loss_wrap = ExpertLoss(lambda l: l)
+    model = model.to(device=device, dtype=dtype)
 
     corpus = load_corpus()
     if args.model == "bloom-tiny":
@@ -87,6 +108,8 @@ def main():
         ids = torch.stack([corpus[s:s + args.seq_len] for s in starts]).to(device)
         optim.zero_grad(set_to_none=True)
         loss = model(ids, labels=ids)
+        if loss_wrap is not None:
+            loss = loss_wrap(loss)  # + router aux/z losses
         loss.backward()
         optim.step()
         if step % 5 == 0 or step == args.steps - 1:
