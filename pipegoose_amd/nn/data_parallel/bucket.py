@@ -67,7 +67,11 @@ class BucketManager:
     ZeRO-1 needs since only the owner steps those params."""
 
     def __init__(self, parallel_context: ParallelContext,
-                 bucket_size_mb: int = BUCKET_SIZE_MB):
+                 bucket_size_mb: int = None):
+        if bucket_size_mb is None:
+            # RuntimeConfig (env PG_BUCKET_MB / yaml) overrides the constant
+            from pipegoose_amd.config import get_config
+            bucket_size_mb = get_config().bucket_size_mb or BUCKET_SIZE_MB
         self.parallel_context = parallel_context
         self.bucket_bytes = bucket_size_mb * 1024 * 1024
         self.buckets: Dict[Tuple[ParallelMode, torch.dtype, Optional[int]],
