@@ -115,3 +115,29 @@ def test_fused_adamw_gpu_bf16_vs_fp32_state_reference():
         st = o2.state[p2]
         assert torch.allclose(st["exp_avg"], m_ref, atol=1e-4)
         assert torch.allclose(st["exp_avg_sq"], v_ref, atol=1e-5)
+
+
+def test_fused_adamw_checkpoint_resume(tmp_path):
+    """save_training_state/load_training_state round-trips FusedAdamW state
+    (weights_only-safe: tensors + primitives only)."""
+    import os
+    from pipegoose_amd.nn.utils import (load_training_state,
+                                        save_training_state)
+    from pipegoose_amd.testing.utils import init_parallel_context
+
+    ctx = init_parallel_context(0, 1, 29930)
+    m1, m2 = _models(torch.float32, "cpu")
+    o1 = FusedAdamW(m1.parameters(), lr=1e-2)
+    _run(m1, o1, "cpu", torch.float32, steps=3)
+    save_training_state(o1, str(tmp_path), parallel_context=ctx, step=3)
+
+    m2.load_state_dict(m1.state_dict())  # resume = weights + optim state
+    o2 = FusedAdamW(m2.parameters(), lr=1e-2)
+    payload = load_training_state(o2, str(tmp_path), parallel_context=ctx)
+    assert payload["step"] == 3
+    # the resumed run must continue EXACTLY like the uninterrupted one
+    _run(m2, o2, "cpu", torch.float32, steps=2, seed=13)
+    _run(m1, o1, "cpu", torch.float32, steps=2, seed=13)
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.equal(p1, p2)
+    ctx.destroy()
