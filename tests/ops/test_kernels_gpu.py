@@ -709,3 +709,29 @@ def test_moe_grouped_layer_gpu(topk):
     ExpertContext.get_instance().pop_all_aux_loss()
     ExpertContext.get_instance().pop_all_z_loss()
     ctx.destroy()
+
+
+def test_generate_uses_graph_decoder_and_matches_eager():
+    """generate()'s serving fast path (hipGraph decode at greedy/tp1) must
+    be token-exact with the eager path it replaces."""
+    import os
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29889")
+    os.environ.setdefault("RANK", "0")
+    os.environ.setdefault("WORLD_SIZE", "1")
+    from pipegoose_amd import ParallelContext
+    from pipegoose_amd.models.bloom import BloomForCausalLM, bloom_tiny
+    from pipegoose_amd.models.generation import generate
+    ctx = ParallelContext.from_torch()
+    torch.manual_seed(23)
+    model = BloomForCausalLM(bloom_tiny(), ctx).to("cuda", torch.bfloat16)
+    ids = torch.randint(0, 256, (4, 32), device="cuda")
+    out_graph = generate(model, ids, max_new_tokens=16, parallel_context=ctx)
+    os.environ["PG_GRAPH_DECODE"] = "0"
+    try:
+        out_eager = generate(model, ids, max_new_tokens=16,
+                             parallel_context=ctx)
+    finally:
+        os.environ.pop("PG_GRAPH_DECODE", None)
+    assert torch.equal(out_graph, out_eager)
+    ctx.destroy()
