@@ -45,8 +45,17 @@ def generate(
             and hasattr(model, "new_graph_kv_cache")
             and os.environ.get("PG_GRAPH_DECODE", "1") == "1"):
         from pipegoose_amd.models.graph_decode import GraphDecoder
-        dec = GraphDecoder(model, batch_size=ids.size(0),
-                           max_len=ids.size(1) + max_new_tokens)
+        need = ids.size(1) + max_new_tokens
+        # reuse the captured graph across calls: a decoder is keyed by batch
+        # size and is valid for any request that fits its cache capacity
+        dec = getattr(model, "_pg_graph_decoder", None)
+        pptr = next(model.parameters()).data_ptr()
+        if dec is None or dec.ids.size(0) != ids.size(0) \
+                or dec.max_len < need \
+                or getattr(dec, "_param_ptr", None) != pptr:
+            dec = GraphDecoder(model, batch_size=ids.size(0), max_len=need)
+            dec._param_ptr = pptr  # graph bakes in weight addresses: a model
+            model._pg_graph_decoder = dec  # move/reload invalidates the capture
         new = dec.generate(ids, max_new_tokens)
         return torch.cat([ids, new], dim=1)
     finished = torch.zeros(ids.size(0), dtype=torch.bool, device=ids.device)

@@ -72,6 +72,22 @@ def main():
           f"decode {t_decode / N * 1e3:.2f} ms/step "
           f"= {B * N / t_decode:,.0f} tok/s")
 
+    # The public entry point (generate() auto-routes greedy tp=1 CUDA decode
+    # through the hipGraph decoder) — this is the number a user of
+    # model.generate() actually sees.
+    with torch.no_grad():
+        model.generate(ids, max_new_tokens=N)  # warmup (+ graph capture)
+        if use_gpu:
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        model.generate(ids, max_new_tokens=N)
+        if use_gpu:
+            torch.cuda.synchronize()
+        t_gen = time.perf_counter() - t0
+    print(f"  generate() [default path]: includes prefill; "
+          f"{(t_gen - t_prefill) / N * 1e3:.2f} ms/step "
+          f"= {B * N / (t_gen - t_prefill):,.0f} tok/s")
+
     if args.graph:
         from pipegoose_amd.models.graph_decode import GraphDecoder
         dec = GraphDecoder(model, batch_size=B, max_len=P + 2 * N + 8)
