@@ -329,6 +329,24 @@ template <int D> __device__ __forceinline__ int bwd_tr(int row, int col) {
     return D == 64 ? row * 40 + col : swz64(row, col);
 }
 
+// Staging-packet decode for the [4][16]-sub-tiled LDS images: one 16-B
+// single-row packet per lane with (col-half e, row&3) in the LOW packet
+// bits.  The [4][16] tiles are 128 B (bank-aligned), so a b128 store's
+// 8-lane group — which under the naive row-pair mapping varies only the
+// column packet — hits the same 8 of 32 banks 4 ways (store bank =
+// (addr/4)%32; tile stride ≡ 0).  With (e, rw) in the low 3 lane bits the
+// 8 addresses land on banks 8*rw + 4*e + [0..3] = all 32 exactly once,
+// while each 16-lane (row) subset of a load instruction still covers one
+// contiguous 256-B row slice, so global coalescing is unchanged.
+template <int D>
+__device__ __forceinline__ void pkt_rc(int p, int& row, int& col) {
+    constexpr int DLOG = (D / 16 == 8) ? 3 : 2;
+    const int rw = (p >> 1) & 3;
+    const int dd = (p >> 3) & (D / 16 - 1);
+    row = (p >> (3 + DLOG)) * 4 + rw;
+    col = dd * 16 + (p & 1) * 8;
+}
+
 // C/D layout of mfma_f32_32x32x16_bf16: col = lane&31, row = (r&3) +
 // 8*(r>>2) + 4*(lane>>5), r in [0,16).  A/B fragments: i/j = lane&31,
 // k = 8*(lane>>5) + elem (verified by mfma_probe32).
@@ -349,7 +367,7 @@ void attn_fwd_v2_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
     constexpr int DCH = D / 16;             // QK^T k-chunks (K=16 per mfma)
     constexpr int DSUB = D / 32;            // O row subtiles (O^T layout)
     constexpr int NPK = KVB * (D / 8);      // K tile 16-B packets
-    constexpr int NPV = (KVB / 2) * (D / 8);  // V tile row-pair packets
+    constexpr int NPV = KVB * (D / 8);      // V tile single-row packets
     constexpr int KPL = (NPK + NT - 1) / NT;
     constexpr int VPL = (NPV + NT - 1) / NT;
     // V image is [kv/4][d/16][kv%4][16]: each [4 kv][16 d] sub-tile is 128 B
@@ -421,7 +439,7 @@ void attn_fwd_v2_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
     nkv = min(max(nkv, 0), S / KVB);
 
     frag_ab kreg[KPL];
-    frag_ab vreg[VPL][2];
+    frag_ab vreg[VPL];
 
     auto stage_load = [&](int nb) {
         const int kvrow0 = nb * KVB;
@@ -439,12 +457,10 @@ void attn_fwd_v2_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
         for (int i = 0; i < VPL; ++i) {
             const int p = tid + i * NT;
             if (p < NPV) {
-                const int row = (p / (D / 8)) * 2;
-                const int col = (p % (D / 8)) * 8;
-                vreg[i][0] = *reinterpret_cast<const frag_ab*>(
+                int row, col;
+                pkt_rc<D>(p, row, col);
+                vreg[i] = *reinterpret_cast<const frag_ab*>(
                     vp + (int64_t)(kvrow0 + row) * vs + col);
-                vreg[i][1] = *reinterpret_cast<const frag_ab*>(
-                    vp + (int64_t)(kvrow0 + row + 1) * vs + col);
             }
         }
     };
@@ -463,16 +479,11 @@ void attn_fwd_v2_kernel(const bf16* __restrict__ q, const bf16* __restrict__ k,
         for (int i = 0; i < VPL; ++i) {
             const int p = tid + i * NT;
             if (p < NPV) {
-                const int row = (p / (D / 8)) * 2;
-                const int col = (p % (D / 8)) * 8;
-#pragma unroll
-                for (int rr = 0; rr < 2; ++rr) {
-                    const int r = row + rr;
-                    const int idx = (r >> 2) * VST + (col >> 4) * 64 +
-                                    (r & 3) * 16 + (col & 15);
-                    *reinterpret_cast<frag_ab*>(&vs_lds[buf][idx]) =
-                        vreg[i][rr];
-                }
+                int row, col;
+                pkt_rc<D>(p, row, col);
+                const int idx = (row >> 2) * VST + (col >> 4) * 64 +
+                                (row & 3) * 16 + (col & 15);
+                *reinterpret_cast<frag_ab*>(&vs_lds[buf][idx]) = vreg[i];
             }
         }
     };
@@ -1272,7 +1283,9 @@ void attn_bwd_dq_v2_kernel(const bf16* __restrict__ dout,
     constexpr int DCH = D / 16;
     constexpr int DSUB = D / 32;
     constexpr int VST = (D / 16) * 64;
-    constexpr int NP = (KVB / 2) * (D / 8);     // row-pair packets per tile
+    constexpr int NPK = KVB * (D / 8);          // K single-row packets
+    constexpr int KPL = (NPK + NT - 1) / NT;
+    constexpr int NP = (KVB / 2) * (D / 8);     // V row-pair packets
     constexpr int PPL = (NP + NT - 1) / NT;
 
     __shared__ __attribute__((aligned(16))) bf16 ks_lds[2][KVB * D];
@@ -1324,19 +1337,25 @@ void attn_bwd_dq_v2_kernel(const bf16* __restrict__ dout,
         : min(S, qblock * ROWS + ROWS - kv_off);
     const int nkv = max(0, (kv_end + KVB - 1) / KVB);
 
-    frag_ab kreg[PPL][2], vreg[PPL][2];
+    frag_ab kreg[KPL], vreg[PPL][2];
     auto stage_load = [&](int nb) {
         const int kvrow0 = nb * KVB;
+#pragma unroll
+        for (int i = 0; i < KPL; ++i) {
+            const int p = tid + i * NT;
+            if (p < NPK) {
+                int row, col;
+                pkt_rc<D>(p, row, col);
+                kreg[i] = *reinterpret_cast<const frag_ab*>(
+                    kp + (int64_t)(kvrow0 + row) * ks2 + col);
+            }
+        }
 #pragma unroll
         for (int i = 0; i < PPL; ++i) {
             const int p = tid + i * NT;
             if (p < NP) {
                 const int row = (p / (D / 8)) * 2;
                 const int col = (p % (D / 8)) * 8;
-                kreg[i][0] = *reinterpret_cast<const frag_ab*>(
-                    kp + (int64_t)(kvrow0 + row) * ks2 + col);
-                kreg[i][1] = *reinterpret_cast<const frag_ab*>(
-                    kp + (int64_t)(kvrow0 + row + 1) * ks2 + col);
                 vreg[i][0] = *reinterpret_cast<const frag_ab*>(
                     vp + (int64_t)(kvrow0 + row) * vs2 + col);
                 vreg[i][1] = *reinterpret_cast<const frag_ab*>(
@@ -1346,6 +1365,16 @@ void attn_bwd_dq_v2_kernel(const bf16* __restrict__ dout,
     };
     auto stage_write = [&](int buf) {
 #pragma unroll
+        for (int i = 0; i < KPL; ++i) {
+            const int p = tid + i * NT;
+            if (p < NPK) {
+                int row, col;
+                pkt_rc<D>(p, row, col);
+                *reinterpret_cast<frag_ab*>(
+                    &ks_lds[buf][sub_idx<D>(row, col)]) = kreg[i];
+            }
+        }
+#pragma unroll
         for (int i = 0; i < PPL; ++i) {
             const int p = tid + i * NT;
             if (p < NP) {
@@ -1353,8 +1382,6 @@ void attn_bwd_dq_v2_kernel(const bf16* __restrict__ dout,
                 const int col = (p % (D / 8)) * 8;
 #pragma unroll
                 for (int rr = 0; rr < 2; ++rr) {
-                    *reinterpret_cast<frag_ab*>(
-                        &ks_lds[buf][sub_idx<D>(row + rr, col)]) = kreg[i][rr];
                     *reinterpret_cast<frag_ab*>(
                         &v_lds[buf][swz128(row + rr, col)]) = vreg[i][rr];
                 }
@@ -1504,7 +1531,7 @@ void attn_bwd_dkdv_v2_kernel(const bf16* __restrict__ dout,
     constexpr int ROWS = WAVES * 32;         // kv rows per workgroup
     constexpr int DCH = D / 16;
     constexpr int DSUB = D / 32;
-    constexpr int NP = (QT / 2) * (D / 8);
+    constexpr int NP = QT * (D / 8);         // single-row staging packets
     constexpr int PPL = (NP + NT - 1) / NT;
 
     __shared__ __attribute__((aligned(16))) bf16 qs_lds[2][QT * D];
@@ -1561,22 +1588,18 @@ void attn_bwd_dkdv_v2_kernel(const bf16* __restrict__ dout,
         const bf16* dop = dout + b * gb + hq * gh;
         const float slope2 = slopes[hq] * LOG2E;
 
-        frag_ab qreg[PPL][2], doreg[PPL][2];
+        frag_ab qreg[PPL], doreg[PPL];
         auto stage_load = [&](int q0) {
 #pragma unroll
             for (int i = 0; i < PPL; ++i) {
                 const int p = tid + i * NT;
                 if (p < NP) {
-                    const int row = (p / (D / 8)) * 2;
-                    const int col = (p % (D / 8)) * 8;
-                    qreg[i][0] = *reinterpret_cast<const frag_ab*>(
+                    int row, col;
+                    pkt_rc<D>(p, row, col);
+                    qreg[i] = *reinterpret_cast<const frag_ab*>(
                         qp + (int64_t)(q0 + row) * qs2 + col);
-                    qreg[i][1] = *reinterpret_cast<const frag_ab*>(
-                        qp + (int64_t)(q0 + row + 1) * qs2 + col);
-                    doreg[i][0] = *reinterpret_cast<const frag_ab*>(
+                    doreg[i] = *reinterpret_cast<const frag_ab*>(
                         dop + (int64_t)(q0 + row) * gs + col);
-                    doreg[i][1] = *reinterpret_cast<const frag_ab*>(
-                        dop + (int64_t)(q0 + row + 1) * gs + col);
                 }
             }
         };
@@ -1585,17 +1608,12 @@ void attn_bwd_dkdv_v2_kernel(const bf16* __restrict__ dout,
             for (int i = 0; i < PPL; ++i) {
                 const int p = tid + i * NT;
                 if (p < NP) {
-                    const int row = (p / (D / 8)) * 2;
-                    const int col = (p % (D / 8)) * 8;
-#pragma unroll
-                    for (int rr = 0; rr < 2; ++rr) {
-                        *reinterpret_cast<frag_ab*>(
-                            &qs_lds[buf][sub_idx<D>(row + rr, col)]) =
-                            qreg[i][rr];
-                        *reinterpret_cast<frag_ab*>(
-                            &dos_lds[buf][sub_idx<D>(row + rr, col)]) =
-                            doreg[i][rr];
-                    }
+                    int row, col;
+                    pkt_rc<D>(p, row, col);
+                    *reinterpret_cast<frag_ab*>(
+                        &qs_lds[buf][sub_idx<D>(row, col)]) = qreg[i];
+                    *reinterpret_cast<frag_ab*>(
+                        &dos_lds[buf][sub_idx<D>(row, col)]) = doreg[i];
                 }
             }
             for (int i = tid; i < QT; i += NT) {
