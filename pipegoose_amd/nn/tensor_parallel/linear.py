@@ -23,6 +23,17 @@ from pipegoose_amd.nn.tensor_parallel._functional import (
 )
 
 
+def _local_linear(mod, input, bias):
+    """The local GEMM of a TP linear: bf16 hipBLASLt by default; fp8
+    (dynamic-scale e4m3, ops/fp8.py) when the layer carries the ``fp8``
+    flag set by ``convert_linear_to_fp8`` — collectives around it are
+    unchanged."""
+    if getattr(mod, "fp8", False):
+        from pipegoose_amd.ops.fp8 import fp8_linear
+        return fp8_linear(input, mod.weight, bias)
+    return TF.linear(input, mod.weight, bias)
+
+
 class ColumnParallelLinear(nn.Module):
     """Y = XW^T + b with W split along output dim: each rank computes a slice
     of the output features; optionally all-gathered along the last dim."""
@@ -58,7 +69,7 @@ class ColumnParallelLinear(nn.Module):
             input = all_gather_sequence(input, self.parallel_context, dim=1)
         else:
             input = broadcast_to_tensor_group(input, self.parallel_context)
-        output = TF.linear(input, self.weight, self.bias)
+        output = _local_linear(self, input, self.bias)
         if self.gather_output:
             output = gather_to_tensor_group(output, dim=-1,
                                             parallel_context=self.parallel_context)
@@ -96,11 +107,11 @@ class RowParallelLinear(nn.Module):
         tp = self.parallel_context.get_world_size(ParallelMode.TENSOR)
         if tp == 1:
             # no combine needed: let hipBLASLt fuse the bias epilogue
-            return TF.linear(input, self.weight, self.bias)
+            return _local_linear(self, input, self.bias)
         if input.size(-1) == self.in_features * tp:
             input = scatter_to_tensor_group(input, dim=-1,
                                             parallel_context=self.parallel_context)
-        output = TF.linear(input, self.weight)
+        output = _local_linear(self, input, None)
         if getattr(self, "sequence_parallel", False):
             output = reduce_scatter_sequence(output, self.parallel_context, dim=1)
             if self.bias is not None:

@@ -46,7 +46,11 @@ def fused_bias_gelu(column_linear, hidden: torch.Tensor) -> torch.Tensor:
         hidden = all_gather_sequence(hidden, column_linear.parallel_context, dim=1)
     else:
         hidden = broadcast_to_tensor_group(hidden, column_linear.parallel_context)
-    x = TF.linear(hidden, column_linear.weight)  # hipBLASLt GEMM, no bias
+    if getattr(column_linear, "fp8", False):
+        from pipegoose_amd.ops.fp8 import fp8_linear
+        x = fp8_linear(hidden, column_linear.weight)      # fp8 GEMM, no bias
+    else:
+        x = TF.linear(hidden, column_linear.weight)  # hipBLASLt GEMM, no bias
     if column_linear.bias is not None:
         return bias_gelu(x, column_linear.bias)
     return TF.gelu(x, approximate="tanh")
