@@ -54,6 +54,8 @@ std::vector<torch::Tensor> router_topk(torch::Tensor logits, int64_t k);
 torch::Tensor silu_mul_fwd(torch::Tensor gate, torch::Tensor up);
 std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dy, torch::Tensor gate,
                                         torch::Tensor up);
+std::vector<torch::Tensor> fp8_quant(torch::Tensor x, bool e5m2);
+torch::Tensor fp8_transpose(torch::Tensor q);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
@@ -91,6 +93,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "rotary embedding with in-kernel cos/sin (gfx950)");
     m.def("router_topk", &router_topk,
           "fused Switch router: softmax+topk+colsum+lse in one pass (gfx950)");
+    m.def("fp8_quant", &fp8_quant,
+          "dynamic per-tensor fp8 quantize: one amax pass + one cast pass");
+    m.def("fp8_transpose", &fp8_transpose,
+          "LDS-tiled byte transpose for fp8 GEMM operand layouts");
     m.def("silu_mul_fwd", &silu_mul_fwd, "fused SwiGLU silu(g)*u (gfx950)");
     m.def("silu_mul_bwd", &silu_mul_bwd, "fused SwiGLU backward (gfx950)");
 }

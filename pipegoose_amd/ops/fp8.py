@@ -36,12 +36,32 @@ def quantize(t: torch.Tensor, dtype: torch.dtype):
     """Per-tensor dynamic scale: returns (q, scale) with q ≈ t / scale.
 
     ``scale`` is the DEQUANT factor (torch._scaled_mm convention:
-    out = (scale_a * a8) @ (scale_b * b8))."""
+    out = (scale_a * a8) @ (scale_b * b8)).  On GPU the HIP kernel does
+    it in one amax pass + one cast pass (csrc/fp8_quant.hip) — the eager
+    route below is 4-5 fp32 passes and is kept as the CPU oracle."""
+    if t.is_cuda and t.dtype == torch.bfloat16:
+        from pipegoose_amd.ops import get_extension
+        ext = get_extension(required=False)
+        if ext is not None and hasattr(ext, "fp8_quant"):
+            q, sc = ext.fp8_quant(t.contiguous(),
+                                  dtype == torch.float8_e5m2)
+            return q, sc[0]
     max_val = E4M3_MAX if dtype == torch.float8_e4m3fn else E5M2_MAX
     amax = t.abs().amax().float().clamp(min=1e-12)
     scale = amax / max_val
     q = (t.float() / scale).clamp(-max_val, max_val).to(dtype)
     return q, scale
+
+
+def _transpose_q(q: torch.Tensor) -> torch.Tensor:
+    """Contiguous byte transpose of a 2D fp8 tensor (LDS-tiled kernel on
+    GPU; the eager fallback is a strided copy)."""
+    if q.is_cuda:
+        from pipegoose_amd.ops import get_extension
+        ext = get_extension(required=False)
+        if ext is not None and hasattr(ext, "fp8_transpose"):
+            return ext.fp8_transpose(q)
+    return q.t().contiguous()
 
 
 def dequantize(q: torch.Tensor, scale: torch.Tensor,
@@ -83,11 +103,11 @@ class _Fp8LinearFn(torch.autograd.Function):
         dy2 = dy.reshape(-1, dy.size(-1))
         dy8, sdy = quantize(dy2, torch.float8_e5m2)
         # dgrad: dy [M,N] @ W [N,K] — scaled_mm wants mat2 column-major
-        w8_cm = w8.t().contiguous().t()
+        w8_cm = _transpose_q(w8).t()
         dx = _scaled_mm(dy8, w8_cm, sdy, sw, dy.dtype)
         # wgrad: dy^T [N,M] @ x [M,K]
-        dy8_t = dy8.t().contiguous()
-        x8_cm = x8.t().contiguous().t()
+        dy8_t = _transpose_q(dy8)
+        x8_cm = _transpose_q(x8).t()
         dw = _scaled_mm(dy8_t, x8_cm, sdy, sx, dy.dtype)
         db = dy2.sum(0) if ctx.has_bias else None
         return dx.reshape(ctx.in_shape), dw, db

@@ -25,6 +25,7 @@ SRC = [
     "pipegoose_amd/ops/csrc/silu_mul.hip",
     "pipegoose_amd/ops/csrc/adamw.hip",
     "pipegoose_amd/ops/csrc/gemm.hip",
+    "pipegoose_amd/ops/csrc/fp8_quant.hip",
 ]
 
 setup(

@@ -78,6 +78,48 @@ def test_fp8_linear_matches_bf16_gpu():
 
 
 @pytest.mark.gpu
+def test_fp8_quant_kernel_matches_eager_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from pipegoose_amd.ops import get_extension
+    from pipegoose_amd.ops.fp8 import E4M3_MAX, E5M2_MAX
+    ext = get_extension(required=True)
+    torch.manual_seed(0)
+    for n, dtype, e5m2, maxv in [((4096, 1024), torch.float8_e4m3fn, False,
+                                  E4M3_MAX),
+                                 ((333,), torch.float8_e5m2, True, E5M2_MAX)]:
+        t = torch.randn(*n, device="cuda", dtype=torch.bfloat16) * 5.0
+        q, sc = ext.fp8_quant(t, e5m2)
+        assert q.dtype == dtype and q.shape == t.shape
+        # eager oracle (same math): amax, scale, clamp, RNE cast
+        amax = t.abs().amax().float().clamp(min=1e-12)
+        want_scale = amax / maxv
+        assert torch.allclose(sc[0], want_scale, rtol=1e-6)
+        want = (t.float() / want_scale).clamp(-maxv, maxv).to(dtype)
+        mism = (q.view(torch.uint8) != want.view(torch.uint8)).sum().item()
+        assert mism <= q.numel() * 1e-3, f"{mism} byte mismatches"
+        # dequantized roundtrip sanity
+        back = q.float() * sc[0]
+        rel = (back - t.float()).abs().mean() / t.float().abs().mean()
+        assert rel < (0.1 if e5m2 else 0.05)
+
+
+@pytest.mark.gpu
+def test_fp8_transpose_kernel_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from pipegoose_amd.ops import get_extension
+    ext = get_extension(required=True)
+    for (r, c) in [(128, 256), (65, 130), (4096, 1024)]:
+        t = torch.randint(0, 255, (r, c), device="cuda",
+                          dtype=torch.uint8).view(torch.float8_e4m3fn)
+        out = ext.fp8_transpose(t)
+        assert out.shape == (c, r)
+        assert torch.equal(out.view(torch.uint8),
+                           t.view(torch.uint8).t().contiguous())
+
+
+@pytest.mark.gpu
 def test_fp8_bloom_mlp_forward_gpu():
     if not torch.cuda.is_available():
         pytest.skip("no GPU")
