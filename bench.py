@@ -25,7 +25,8 @@ def parse_args():
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", type=str, default="bloom-560m",
-                   choices=["bloom-560m", "bloom-1b7", "bloom-7b1", "bloom-tiny"])
+                   choices=["bloom-560m", "bloom-1b7", "bloom-7b1",
+                            "bloom-tiny", "llama-7b", "llama-1b"])
     p.add_argument("--seq-len", type=int, default=2048)
     p.add_argument("--micro-batch", type=int, default=8,
                    help="per-DP-rank batch size (weak scaling)")
@@ -53,6 +54,11 @@ def parse_args():
                    help="replace MLPs with E-expert Switch MoE layers "
                         "(BASELINE config 4: bloom-1b7 with 8 experts; "
                         "all-to-all dispatch over the TP/EP group when tp>1)")
+    p.add_argument("--fp8-mlp", action="store_true",
+                   help="EXPERIMENTAL: run the MLP GEMMs in OCP fp8 "
+                        "(dynamic e4m3/e5m2 scaling, ops/fp8.py). Never "
+                        "the default: the reported dtype changes and the "
+                        "headline benchmark stays bf16.")
     p.add_argument("--device", type=str, default=None)
     return p.parse_args()
 
@@ -133,7 +139,17 @@ def main():
         t.start()
 
     cfg = {"bloom-560m": bloom_560m, "bloom-1b7": bloom_1b7,
-           "bloom-7b1": bloom_7b1, "bloom-tiny": bloom_tiny}[args.model]()
+           "bloom-7b1": bloom_7b1, "bloom-tiny": bloom_tiny,
+           "llama-7b": None, "llama-1b": None}[args.model]
+    is_llama = cfg is None
+    if is_llama:
+        from pipegoose_amd.models.llama import (LlamaForCausalLM, llama_1b,
+                                                llama_7b)
+        cfg = (llama_7b if args.model == "llama-7b" else llama_1b)()
+        assert pp == 1 and args.moe == 0, \
+            "llama bench covers tp/dp/cp (pp/moe are the bloom configs)"
+    else:
+        cfg = cfg()
 
     if args.sp:
         cfg.sequence_parallel = True
@@ -141,7 +157,7 @@ def main():
         cfg.context_parallel = True
         assert args.seq_len % cp == 0
     torch.manual_seed(1234)
-    model = BloomForCausalLM(cfg, ctx)
+    model = LlamaForCausalLM(cfg, ctx) if is_llama else BloomForCausalLM(cfg, ctx)
     moe_loss_wrap = None
     if args.moe > 0:
         from torch import nn
@@ -166,6 +182,13 @@ def main():
             schedule="interleaved" if v > 1 else "1f1b", virtual_stages=v,
             loss_fn=make_causal_lm_loss(ctx)).parallelize()
     model = model.to(device=device, dtype=dtype)
+    if args.fp8_mlp:
+        from pipegoose_amd.ops.fp8 import convert_linear_to_fp8
+        n_fp8 = convert_linear_to_fp8(
+            model, names=["dense_h_to_4h", "dense_4h_to_h", "gate_proj",
+                          "up_proj", "down_proj"])
+        if rank == 0:
+            print(f"[bench] fp8 MLP: {n_fp8} linears converted", flush=True)
     if dp > 1:
         model = DataParallel(model, ctx).parallelize()
     if cp > 1:
@@ -264,6 +287,7 @@ def main():
     tokens_per_sec = tokens_per_step * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1e3
 
+    dtype_name = "bf16+fp8-mlp(experimental)" if args.fp8_mlp else "bf16"
     if rank == 0:
         print(json.dumps({
             "metric": "tokens/sec",
@@ -276,7 +300,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if use_gpu else "fp32",
+            "dtype": dtype_name if use_gpu else "fp32",
             "data": "synthetic",
             "config": {
                 "model": args.model,
