@@ -56,6 +56,9 @@ def main():
                     help="replace MLPs with E-expert Switch MoE "
                          "(mirrors the reference's run_ep.py artifact)")
     ap.add_argument("--seed", type=int, default=1234)
+    ap.add_argument("--fp8-mlp", action="store_true",
+                    help="run the MLP GEMMs in OCP fp8 (convergence check "
+                         "for the experimental fp8 path)")
     args = ap.parse_args()
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
@@ -88,6 +91,11 @@ def main():
             parallel_context=ctx).parallelize()
         loss_wrap = ExpertLoss(lambda l: l)
     model = model.to(device=device, dtype=dtype)
+    if args.fp8_mlp:
+        from pipegoose_amd.ops.fp8 import convert_linear_to_fp8
+        n8 = convert_linear_to_fp8(model, names=["dense_h_to_4h",
+                                                 "dense_4h_to_h"])
+        print(f"fp8 MLP: {n8} linears converted", flush=True)
 
     corpus = load_corpus()
     if args.model == "bloom-tiny":
