@@ -59,6 +59,9 @@ def parse_args():
                         "(dynamic e4m3/e5m2 scaling, ops/fp8.py). Never "
                         "the default: the reported dtype changes and the "
                         "headline benchmark stays bf16.")
+    p.add_argument("--fp8-attn", action="store_true",
+                   help="EXPERIMENTAL: also run the attention projection "
+                        "GEMMs (qkv / output dense) in fp8")
     p.add_argument("--device", type=str, default=None)
     return p.parse_args()
 
@@ -182,13 +185,18 @@ def main():
             schedule="interleaved" if v > 1 else "1f1b", virtual_stages=v,
             loss_fn=make_causal_lm_loss(ctx)).parallelize()
     model = model.to(device=device, dtype=dtype)
-    if args.fp8_mlp:
+    if args.fp8_mlp or args.fp8_attn:
         from pipegoose_amd.ops.fp8 import convert_linear_to_fp8
-        n_fp8 = convert_linear_to_fp8(
-            model, names=["dense_h_to_4h", "dense_4h_to_h", "gate_proj",
-                          "up_proj", "down_proj"])
+        names = []
+        if args.fp8_mlp:
+            names += ["dense_h_to_4h", "dense_4h_to_h", "gate_proj",
+                      "up_proj", "down_proj"]
+        if args.fp8_attn:
+            names += ["query_key_value", ".dense", "q_proj", "k_proj",
+                      "v_proj", "o_proj"]
+        n_fp8 = convert_linear_to_fp8(model, names=names)
         if rank == 0:
-            print(f"[bench] fp8 MLP: {n_fp8} linears converted", flush=True)
+            print(f"[bench] fp8: {n_fp8} linears converted", flush=True)
     if dp > 1:
         model = DataParallel(model, ctx).parallelize()
     if cp > 1:
@@ -287,7 +295,11 @@ def main():
     tokens_per_sec = tokens_per_step * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1e3
 
-    dtype_name = "bf16+fp8-mlp(experimental)" if args.fp8_mlp else "bf16"
+    dtype_name = "bf16"
+    if args.fp8_mlp or args.fp8_attn:
+        tags = (["mlp"] if args.fp8_mlp else []) + \
+               (["attn"] if args.fp8_attn else [])
+        dtype_name = f"bf16+fp8-{'+'.join(tags)}(experimental)"
     if rank == 0:
         print(json.dumps({
             "metric": "tokens/sec",
