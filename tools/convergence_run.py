@@ -59,6 +59,8 @@ def main():
     ap.add_argument("--fp8-mlp", action="store_true",
                     help="run the MLP GEMMs in OCP fp8 (convergence check "
                          "for the experimental fp8 path)")
+    ap.add_argument("--fp8-attn", action="store_true",
+                    help="also run the attention projections in fp8")
     args = ap.parse_args()
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
@@ -91,11 +93,12 @@ def main():
             parallel_context=ctx).parallelize()
         loss_wrap = ExpertLoss(lambda l: l)
     model = model.to(device=device, dtype=dtype)
-    if args.fp8_mlp:
+    if args.fp8_mlp or args.fp8_attn:
         from pipegoose_amd.ops.fp8 import convert_linear_to_fp8
-        n8 = convert_linear_to_fp8(model, names=["dense_h_to_4h",
-                                                 "dense_4h_to_h"])
-        print(f"fp8 MLP: {n8} linears converted", flush=True)
+        names = (["dense_h_to_4h", "dense_4h_to_h"] if args.fp8_mlp else []) \
+            + (["query_key_value", ".dense"] if args.fp8_attn else [])
+        n8 = convert_linear_to_fp8(model, names=names)
+        print(f"fp8: {n8} linears converted", flush=True)
 
     corpus = load_corpus()
     if args.model == "bloom-tiny":
