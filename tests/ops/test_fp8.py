@@ -77,6 +77,34 @@ def test_fp8_linear_matches_bf16_gpu():
     assert rel(lin.bias.grad, gb_ref) < 0.02   # bias grad stays bf16-exact
 
 
+def test_tp_layer_fp8_flag_cpu_fallback():
+    """convert_linear_to_fp8 flags Column/RowParallelLinear; off-GPU the
+    flagged forward must fall back to the exact bf16/fp32 path."""
+    import os
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29514")
+    from pipegoose_amd.testing.utils import init_parallel_context
+    from pipegoose_amd.nn.tensor_parallel.linear import (ColumnParallelLinear,
+                                                         RowParallelLinear)
+    ctx = init_parallel_context(0, 1, 29514)
+    try:
+        torch.manual_seed(0)
+        col = ColumnParallelLinear(32, 64, parallel_context=ctx)
+        row = RowParallelLinear(64, 32, parallel_context=ctx)
+        for m in (col, row):
+            torch.nn.init.normal_(m.weight)
+            torch.nn.init.normal_(m.bias)
+        x = torch.randn(4, 32)
+        want = row(col(x))
+        holder = nn.ModuleDict({"c": col, "r": row})
+        assert convert_linear_to_fp8(holder) == 2
+        assert col.fp8 and row.fp8
+        assert torch.equal(row(col(x)), want)
+        assert revert_fp8(holder) == 2
+    finally:
+        ctx.destroy()
+
+
 @pytest.mark.gpu
 def test_fp8_quant_kernel_matches_eager_gpu():
     if not torch.cuda.is_available():
