@@ -55,6 +55,7 @@ torch::Tensor silu_mul_fwd(torch::Tensor gate, torch::Tensor up);
 std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dy, torch::Tensor gate,
                                         torch::Tensor up);
 std::vector<torch::Tensor> fp8_quant(torch::Tensor x, bool e5m2);
+std::vector<torch::Tensor> fp8_quant_dual(torch::Tensor x, bool e5m2);
 torch::Tensor fp8_transpose(torch::Tensor q);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -95,6 +96,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused Switch router: softmax+topk+colsum+lse in one pass (gfx950)");
     m.def("fp8_quant", &fp8_quant,
           "dynamic per-tensor fp8 quantize: one amax pass + one cast pass");
+    m.def("fp8_quant_dual", &fp8_quant_dual,
+          "fp8 quantize emitting row-major AND transposed images in one "
+          "read (kills the separate backward-layout transpose pass)");
     m.def("fp8_transpose", &fp8_transpose,
           "LDS-tiled byte transpose for fp8 GEMM operand layouts");
     m.def("silu_mul_fwd", &silu_mul_fwd, "fused SwiGLU silu(g)*u (gfx950)");

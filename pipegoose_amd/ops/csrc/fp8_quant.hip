@@ -97,6 +97,67 @@ __global__ void cast_kernel(const bf16* __restrict__ x,
     }
 }
 
+// Fused cast + dual-layout write: one bf16 read produces BOTH the
+// row-major and the transposed fp8 image (64x64 LDS tile for the
+// transposed side).  Eliminates the separate byte-transpose pass the
+// backward GEMM layouts otherwise need (measured 5.8% of the fp8-7b1
+// step).  Requires C % 4 == 0 (uchar4 row-major stores).
+template <bool E5M2>
+__global__ void cast_dual_kernel(const bf16* __restrict__ in,
+                                 uint8_t* __restrict__ q,
+                                 uint8_t* __restrict__ qt,
+                                 const float* __restrict__ scale,
+                                 int R, int C) {
+    __shared__ uint8_t tile[64][65];
+    const float inv = scale[1];
+    const float maxv = E5M2 ? E5M2_MAX : E4M3_MAX;
+    const int tr = blockIdx.y * 64;
+    const int tc = blockIdx.x * 64;
+    const int t = threadIdx.x;            // 256 threads
+    const int lr = t / 16, lc4 = (t % 16) * 4;
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+        const int r = tr + lr + s * 16;
+        const int c = tc + lc4;
+        if (r < R && c < C) {
+            union { uint8_t b[4]; uchar4 u; } o;
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                float f = to_float(in[(int64_t)r * C + c + j]);
+                f = fminf(fmaxf(f * inv, -maxv), maxv);
+                if (E5M2) {
+                    __hip_fp8_e5m2 cv(f);
+                    o.b[j] = cv.__x;
+                } else {
+                    __hip_fp8_e4m3 cv(f);
+                    o.b[j] = cv.__x;
+                }
+                tile[lr + s * 16][lc4 + j] = o.b[j];
+            }
+            *reinterpret_cast<uchar4*>(q + (int64_t)r * C + c) = o.u;
+        }
+    }
+    __syncthreads();
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+        const int c = tc + lr + s * 16;   // output row = input column
+        if (c < C) {
+            const int r = tr + lc4;
+            if (r + 3 < R) {
+                uchar4 v;
+                v.x = tile[lc4 + 0][lr + s * 16];
+                v.y = tile[lc4 + 1][lr + s * 16];
+                v.z = tile[lc4 + 2][lr + s * 16];
+                v.w = tile[lc4 + 3][lr + s * 16];
+                *reinterpret_cast<uchar4*>(qt + (int64_t)c * R + r) = v;
+            } else {
+                for (int j = 0; j < 4 && r + j < R; ++j)
+                    qt[(int64_t)c * R + r + j] = tile[lc4 + j][lr + s * 16];
+            }
+        }
+    }
+}
+
 // [R, C] bytes -> [C, R]: 64x64 LDS tiles, uchar4 global loads AND stores
 // (both sides coalesced; the torch fallback is a strided byte copy).
 __global__ void transpose_u8_kernel(const uint8_t* __restrict__ in,
@@ -181,6 +242,44 @@ std::vector<torch::Tensor> fp8_quant(torch::Tensor x, bool e5m2) {
                            dim3(threads), 0, stream, xp, qp,
                            scale.data_ptr<float>(), n);
     return {q, scale};
+}
+
+// Returns {q [R,C], qT [C,R], scale[2]} from one bf16 read.
+std::vector<torch::Tensor> fp8_quant_dual(torch::Tensor x, bool e5m2) {
+    TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 &&
+                x.dim() == 2 && x.is_contiguous() && x.size(1) % 4 == 0,
+                "fp8_quant_dual: contiguous 2D bf16, cols % 4 == 0");
+    const int R = (int)x.size(0), C = (int)x.size(1);
+    auto opts8 = x.options().dtype(e5m2 ? torch::kFloat8_e5m2
+                                        : torch::kFloat8_e4m3fn);
+    auto q = torch::empty({R, C}, opts8);
+    auto qt = torch::empty({C, R}, opts8);
+    auto scale = torch::empty({2}, x.options().dtype(torch::kFloat));
+    const int64_t n = x.numel();
+    const int threads = 256;
+    const int blocks = (int)std::min<int64_t>((n / 8 + threads - 1) / threads,
+                                              2048);
+    auto partials = torch::empty({std::max(blocks, 1)},
+                                 x.options().dtype(torch::kFloat));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    const float maxv = e5m2 ? E5M2_MAX : E4M3_MAX;
+    const bf16* xp = reinterpret_cast<const bf16*>(x.data_ptr());
+    hipLaunchKernelGGL(amax_partial_kernel, dim3(std::max(blocks, 1)),
+                       dim3(threads), 0, stream, xp,
+                       partials.data_ptr<float>(), n);
+    hipLaunchKernelGGL(finalize_scale_kernel, dim3(1), dim3(threads), 0,
+                       stream, partials.data_ptr<float>(),
+                       std::max(blocks, 1), maxv, scale.data_ptr<float>());
+    dim3 grid((C + 63) / 64, (R + 63) / 64);
+    uint8_t* qp = reinterpret_cast<uint8_t*>(q.data_ptr());
+    uint8_t* qtp = reinterpret_cast<uint8_t*>(qt.data_ptr());
+    if (e5m2)
+        hipLaunchKernelGGL((cast_dual_kernel<true>), grid, dim3(256), 0,
+                           stream, xp, qp, qtp, scale.data_ptr<float>(), R, C);
+    else
+        hipLaunchKernelGGL((cast_dual_kernel<false>), grid, dim3(256), 0,
+                           stream, xp, qp, qtp, scale.data_ptr<float>(), R, C);
+    return {q, qt, scale};
 }
 
 torch::Tensor fp8_transpose(torch::Tensor q) {

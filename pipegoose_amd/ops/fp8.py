@@ -64,6 +64,21 @@ def _transpose_q(q: torch.Tensor) -> torch.Tensor:
     return q.t().contiguous()
 
 
+def _quant_dual(t: torch.Tensor, dtype: torch.dtype):
+    """quantize() that also returns the transposed fp8 image: one fused
+    kernel on GPU, quantize+transpose fallback elsewhere."""
+    if t.is_cuda and t.dtype == torch.bfloat16 and t.dim() == 2 \
+            and t.size(1) % 4 == 0:
+        from pipegoose_amd.ops import get_extension
+        ext = get_extension(required=False)
+        if ext is not None and hasattr(ext, "fp8_quant_dual"):
+            q, qt, sc = ext.fp8_quant_dual(t.contiguous(),
+                                           dtype == torch.float8_e5m2)
+            return q, qt, sc[0]
+    q, s = quantize(t, dtype)
+    return q, _transpose_q(q), s
+
+
 def dequantize(q: torch.Tensor, scale: torch.Tensor,
                dtype: torch.dtype = torch.float32) -> torch.Tensor:
     return q.to(torch.float32).mul(scale).to(dtype)
@@ -87,28 +102,29 @@ class _Fp8LinearFn(torch.autograd.Function):
     def forward(ctx, x, weight, bias):
         shp = x.shape
         x2 = x.reshape(-1, shp[-1])
-        x8, sx = quantize(x2, torch.float8_e4m3fn)
-        w8, sw = quantize(weight, torch.float8_e4m3fn)
+        # dual quantize: the transposed images the backward GEMM layouts
+        # need are produced in the same read (csrc cast_dual_kernel) —
+        # no separate byte-transpose pass at backward time
+        x8, x8t, sx = _quant_dual(x2, torch.float8_e4m3fn)
+        w8, w8t, sw = _quant_dual(weight, torch.float8_e4m3fn)
         y = _scaled_mm(x8, w8.t(), sx, sw, x.dtype)
         if bias is not None:
             y = y + bias
-        ctx.save_for_backward(x8, sx, w8, sw)
+        ctx.save_for_backward(x8t, sx, w8t, sw)
         ctx.has_bias = bias is not None
         ctx.in_shape = shp
         return y.reshape(*shp[:-1], weight.size(0))
 
     @staticmethod
     def backward(ctx, dy):
-        x8, sx, w8, sw = ctx.saved_tensors
+        x8t, sx, w8t, sw = ctx.saved_tensors
         dy2 = dy.reshape(-1, dy.size(-1))
-        dy8, sdy = quantize(dy2, torch.float8_e5m2)
-        # dgrad: dy [M,N] @ W [N,K] — scaled_mm wants mat2 column-major
-        w8_cm = _transpose_q(w8).t()
-        dx = _scaled_mm(dy8, w8_cm, sdy, sw, dy.dtype)
-        # wgrad: dy^T [N,M] @ x [M,K]
-        dy8_t = _transpose_q(dy8)
-        x8_cm = _transpose_q(x8).t()
-        dw = _scaled_mm(dy8_t, x8_cm, sdy, sx, dy.dtype)
+        dy8, dy8t, sdy = _quant_dual(dy2, torch.float8_e5m2)
+        # dgrad: dy [M,N] @ W [N,K] — scaled_mm wants mat2 column-major,
+        # i.e. the [K,N] row-major transposed image viewed back
+        dx = _scaled_mm(dy8, w8t.t(), sdy, sw, dy.dtype)
+        # wgrad: dy^T [N,M] @ x [M,K] (col-major via the [K,M] image)
+        dw = _scaled_mm(dy8t, x8t.t(), sdy, sx, dy.dtype)
         db = dy2.sum(0) if ctx.has_bias else None
         return dx.reshape(ctx.in_shape), dw, db
 

@@ -133,6 +133,23 @@ def test_fp8_quant_kernel_matches_eager_gpu():
 
 
 @pytest.mark.gpu
+def test_fp8_quant_dual_matches_separate_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from pipegoose_amd.ops import get_extension
+    ext = get_extension(required=True)
+    torch.manual_seed(3)
+    for (r, c), e5m2 in [((512, 256), False), ((48, 132), True)]:
+        t = torch.randn(r, c, device="cuda", dtype=torch.bfloat16) * 2.0
+        q, qt, sc = ext.fp8_quant_dual(t, e5m2)
+        q_ref, sc_ref = ext.fp8_quant(t, e5m2)
+        assert torch.allclose(sc, sc_ref)
+        assert torch.equal(q.view(torch.uint8), q_ref.view(torch.uint8))
+        assert torch.equal(qt.view(torch.uint8),
+                           q_ref.view(torch.uint8).t().contiguous())
+
+
+@pytest.mark.gpu
 def test_fp8_transpose_kernel_gpu():
     if not torch.cuda.is_available():
         pytest.skip("no GPU")
