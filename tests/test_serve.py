@@ -32,7 +32,8 @@ def test_engine_batches_and_matches_direct():
     try:
         model = _tiny_model(ctx)
         engine = BatchingEngine(model, parallel_context=ctx, max_batch=4,
-                                max_wait_ms=50.0)
+                                max_wait_ms=300.0)  # generous: slow CI boxes
+                                                    # must still coalesce
         torch.manual_seed(1)
         prompts = [torch.randint(0, 100, (8,)) for _ in range(3)]
         odd = torch.randint(0, 100, (5,))          # different length
