@@ -185,6 +185,9 @@ def main():
             schedule="interleaved" if v > 1 else "1f1b", virtual_stages=v,
             loss_fn=make_causal_lm_loss(ctx)).parallelize()
     model = model.to(device=device, dtype=dtype)
+    # flags OR the PG_FP8_* env knobs (config.py RuntimeConfig)
+    args.fp8_mlp = args.fp8_mlp or os.environ.get("PG_FP8_MLP") == "1"
+    args.fp8_attn = args.fp8_attn or os.environ.get("PG_FP8_ATTN") == "1"
     if args.fp8_mlp or args.fp8_attn:
         from pipegoose_amd.ops.fp8 import convert_linear_to_fp8
         names = []

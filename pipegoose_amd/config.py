@@ -29,6 +29,8 @@ _ENV_MAP = {
     "moe_grouped": "PG_MOE_GROUPED",
     "hand_gemm": "PG_HAND_GEMM",
     "bgelu_rows": "PG_BGELU_ROWS",
+    "fp8_mlp": "PG_FP8_MLP",
+    "fp8_attn": "PG_FP8_ATTN",
     "optimizer": "PG_OPT",
     "bench_step_timeout_s": "PG_BENCH_STEP_TIMEOUT_S",
     "bench_deadline_s": "PG_BENCH_DEADLINE_S",
@@ -48,6 +50,9 @@ class RuntimeConfig:
     hand_gemm: int = 0                # 1 forces the hand MFMA GEMM
     bgelu_rows: int = 1024            # bias-gelu bwd grid rows
     disable_ext: int = 0              # 1 disables the HIP extension
+    # precision (EXPERIMENTAL: headline benchmarks stay bf16)
+    fp8_mlp: int = 0                  # 1 = MLP GEMMs in OCP fp8 (ops/fp8.py)
+    fp8_attn: int = 0                 # 1 = attention projections in fp8 too
     # training-loop
     optimizer: str = "hip"            # hip | fused | foreach (bench.py)
     bench_step_timeout_s: float = 300.0
